@@ -1,0 +1,123 @@
+"""Flagship benchmark: FM k=16 training on synthetic Criteo-shaped sparse
+data (BASELINE.json config #2: "FM k=16 on synthetic Criteo-1TB-shaped
+sparse, 1xMI355X, FTRL fused update"), scaled to N GPUs (weak scaling: fixed
+per-GPU batch; N>1 shards the feature table by feature-hash across ranks and
+exchanges embeddings/grads with RCCL all-to-all over xGMI).
+
+Contract (driver): `python bench.py --gpus N --steps K --warmup W`; for N>1
+launched via torch.distributed.run with one rank per GPU. Rank 0 prints one
+JSON line with the whole-job examples/sec.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--batch", type=int, default=65536,
+                    help="per-GPU minibatch (rows)")
+    ap.add_argument("--k", type=int, default=16)
+    ap.add_argument("--features", type=int, default=1 << 24)
+    ap.add_argument("--optimizer", default="ftrl")
+    args = ap.parse_args()
+
+    from lightctr_amd.data.synthetic import SyntheticCriteo
+    from lightctr_amd.models.fm import FMHyper, FMModel
+
+    rank = int(os.environ.get("RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    have_gpu = torch.cuda.is_available()
+    local_rank = int(os.environ.get("LOCAL_RANK", 0))
+    device = f"cuda:{local_rank}" if have_gpu else "cpu"
+    if have_gpu:
+        torch.cuda.set_device(local_rank)
+
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+
+        dist = dist_mod
+        dist.init_process_group(backend="nccl" if have_gpu else "gloo")
+
+    hyper = FMHyper(num_features=args.features, k=args.k,
+                    optimizer=args.optimizer, seed=1234)
+    if world > 1:
+        from lightctr_amd.parallel.sharded_fm import ShardedFMModel
+
+        model = ShardedFMModel(hyper, device=device)
+    else:
+        model = FMModel(hyper, device=device)
+
+    gen = SyntheticCriteo(num_features=args.features, seed=1234 + rank,
+                          device=device)
+    # pre-generate a rotating pool of batches so the timed region is pure
+    # training (synthetic data; generation is not the benchmark subject)
+    pool = [gen.batch(args.batch) for _ in range(4)]
+
+    def step(i):
+        row_ptr, fields, fids, vals, labels = pool[i % len(pool)]
+        return model.train_step(row_ptr, fids, vals, labels)
+
+    for i in range(args.warmup):
+        step(i)
+    if dist:
+        dist.barrier()
+    if have_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        step(i)
+    if dist:
+        dist.barrier()
+    if have_gpu:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    # MAX over ranks
+    if dist:
+        t = torch.tensor([elapsed], device=device if have_gpu else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    n_gpus = world if world > 1 else (1 if have_gpu else 0) or 1
+    total_examples = args.batch * args.steps * world
+    value = total_examples / elapsed
+    if rank == 0:
+        out = {
+            "metric": "examples/sec (whole node), FM training on synthetic Criteo-shaped sparse",
+            "value": value,
+            "unit": "examples/sec",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": f"FM k={args.k}, {args.optimizer} fused update",
+                "global_batch": args.batch * world,
+                "num_features": args.features,
+                "num_fields": 39,
+                "parallelism": f"hash-sharded table, all-to-all, dp{world}"
+                if world > 1 else "single-gpu",
+            },
+        }
+        print(json.dumps(out))
+    if dist:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,46 @@
+"""Loader for the in-tree HIP extension (lightctr_amd.ops._hip_ops).
+
+The extension is built in-tree by ``python setup.py build_ext --inplace``
+(driven by __graft_entry__.build()) so the .so ships with the repo snapshot
+to GPU boxes. On a machine with a GPU, ops FAIL LOUDLY if the extension is
+missing — no silent eager fallback on the compute path.
+"""
+
+from __future__ import annotations
+
+import torch  # noqa: F401  -- loads libc10/libtorch before the extension
+
+_hip_ops = None
+_import_error: Exception | None = None
+
+try:
+    from . import _hip_ops as _ext  # type: ignore
+
+    _hip_ops = _ext
+except ImportError as e:  # extension not built (CPU-only dev box is fine)
+    _import_error = e
+
+
+def has_hip_ops() -> bool:
+    return _hip_ops is not None
+
+
+def require_hip_ops():
+    if _hip_ops is None:
+        raise ImportError(
+            "lightctr_amd HIP extension (_hip_ops) is not built but a GPU "
+            "op was requested. Build it in-tree with "
+            "`PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace`. "
+            f"Original import error: {_import_error!r}"
+        )
+    return _hip_ops
+
+
+class _LazyOps:
+    """Attribute access proxy: hip_ops.fm_forward(...) resolves at call time."""
+
+    def __getattr__(self, name):
+        return getattr(require_hip_ops(), name)
+
+
+hip_ops = _LazyOps()

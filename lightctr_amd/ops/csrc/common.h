@@ -1,0 +1,40 @@
+// Common device helpers for lightctr_amd CDNA4 (gfx950) kernels.
+// Hand-written HIP for MI355X: 64-wide wavefronts, LDS, MFMA.
+#pragma once
+#include <hip/hip_runtime.h>
+
+#define LCTR_WAVE 64
+
+// Full-wave sum reduction (64 lanes).
+__device__ __forceinline__ float wave_reduce_sum(float v) {
+#pragma unroll
+  for (int s = 1; s < LCTR_WAVE; s <<= 1) v += __shfl_xor(v, s);
+  return v;
+}
+
+// Reduce across stride groups: lanes {l, l+stride, l+2*stride, ...} end up
+// holding the sum over their group (used to combine per-feature-group
+// partials that live at lane = g*K + k for fixed k).
+template <int START>
+__device__ __forceinline__ float group_reduce_sum(float v) {
+#pragma unroll
+  for (int s = START; s < LCTR_WAVE; s <<= 1) v += __shfl_xor(v, s);
+  return v;
+}
+
+__device__ __forceinline__ float sigmoidf_clamped(float z) {
+  // Reference semantics: sigmoid clamped to +-16 pre-activation
+  // (reference util/activations.h Sigmoid).
+  z = fminf(16.f, fmaxf(-16.f, z));
+  return 1.f / (1.f + __expf(-z));
+}
+
+#define LCTR_CHECK_HIP(expr)                                        \
+  do {                                                              \
+    hipError_t _e = (expr);                                         \
+    if (_e != hipSuccess) {                                         \
+      fprintf(stderr, "HIP error %s at %s:%d\n",                    \
+              hipGetErrorString(_e), __FILE__, __LINE__);           \
+      abort();                                                      \
+    }                                                               \
+  } while (0)
