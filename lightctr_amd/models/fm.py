@@ -71,6 +71,7 @@ class FMModel:
         self.uniq = torch.zeros(cap, dtype=torch.int32, device=self.device)
         self.count = torch.zeros(1, dtype=torch.int32, device=self.device)
         self._use_hip = self.device.type == "cuda"
+        self.backward_mode = "sorted"  # "sorted" (default) | "atomic"
         if self._use_hip:
             require_hip_ops()  # fail loudly if extension missing on GPU
 
@@ -97,8 +98,18 @@ class FMModel:
             ops = require_hip_ops()
             pred, sumVX = ops.fm_forward(row_ptr, fids, vals, self.W, self.V)
             loss, dpred = ops.logloss_grad(pred, labels, scale)
-            ops.fm_backward(row_ptr, fids, vals, self.V, sumVX, dpred,
-                            self.gradW, self.gradV, self.touched)
+            if self.backward_mode == "sorted":
+                # contention-free backward: emit per-entry grads, radix-sort
+                # by fid, segment-reduce into the slabs (see
+                # profiles/r01_fm_atomic_backward.txt for why)
+                gw, gv = ops.fm_backward_emit(row_ptr, fids, vals, self.V,
+                                              sumVX, dpred)
+                sorted_fids, perm = torch.sort(fids)
+                ops.fm_sorted_apply(sorted_fids, perm, gw, gv, self.gradW,
+                                    self.gradV, self.touched)
+            else:
+                ops.fm_backward(row_ptr, fids, vals, self.V, sumVX, dpred,
+                                self.gradW, self.gradV, self.touched)
             self.count.zero_()
             ops.bitmap_compact(self.touched, self.uniq, self.count)
             if self.h.optimizer == "ftrl":

@@ -29,6 +29,20 @@ __device__ __forceinline__ float sigmoidf_clamped(float z) {
   return 1.f / (1.f + __expf(-z));
 }
 
+// Runtime -> compile-time K dispatch (K must divide the 64-lane wave).
+#define DISPATCH_K(KVAL, ...)                                        \
+  switch (KVAL) {                                                    \
+    case 4: { constexpr int KC = 4; __VA_ARGS__; break; }            \
+    case 8: { constexpr int KC = 8; __VA_ARGS__; break; }            \
+    case 16: { constexpr int KC = 16; __VA_ARGS__; break; }          \
+    case 32: { constexpr int KC = 32; __VA_ARGS__; break; }          \
+    case 64: { constexpr int KC = 64; __VA_ARGS__; break; }          \
+    default:                                                         \
+      fprintf(stderr, "lightctr_amd: unsupported K=%d\n", KVAL);     \
+      abort();                                                       \
+  }
+#define DISPATCH_FFM_K DISPATCH_K
+
 #define LCTR_CHECK_HIP(expr)                                        \
   do {                                                              \
     hipError_t _e = (expr);                                         \

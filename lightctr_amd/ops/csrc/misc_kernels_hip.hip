@@ -1,0 +1,104 @@
+#include "hip/hip_runtime.h"
+// Generic fused sparse optimizers (runtime latent width D) + small utility
+// kernels for MI355X/gfx950. Used by FFM (D = nfields*K) and any model with
+// per-feature parameter blocks; semantics follow the reference updaters
+// (/root/reference/LightCTR/util/gradientUpdater.h: Adagrad :128-154,
+// FTRL-proximal :235-278).
+#include "common.h"
+
+namespace lightctr {
+
+// declared in fm_kernels.hip
+__device__ __forceinline__ void ftrl_update_g(float* w, float* z, float* n,
+                                              float g, float alpha, float beta,
+                                              float l1, float l2) {
+  const float g2 = g * g;
+  const float nold = *n;
+  const float sigma = (sqrtf(nold + g2) - sqrtf(nold)) / alpha;
+  const float znew = *z + g - sigma * (*w);
+  const float nnew = nold + g2;
+  *z = znew;
+  *n = nnew;
+  if (fabsf(znew) <= l1) {
+    *w = 0.f;
+  } else {
+    *w = -(znew - copysignf(l1, znew)) / ((beta + sqrtf(nnew)) / alpha + l2);
+  }
+}
+
+// One wavefront per unique feature; lanes stride the D latent params
+// (coalesced 64-wide), lane 0 also updates the linear weight W.
+__global__ void sparse_adagrad_apply_g_kernel(
+    const int* __restrict__ uniq, const int* __restrict__ count,
+    float* __restrict__ W, float* __restrict__ V, float* __restrict__ nW,
+    float* __restrict__ nV, float* __restrict__ gradW,
+    float* __restrict__ gradV, float lr, float eps, float l2, int capacity,
+    int D) {
+  const int i = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  const int lane = threadIdx.x & (LCTR_WAVE - 1);
+  if (i >= capacity || i >= *count) return;
+  const int fid = uniq[i];
+  const size_t base = (size_t)fid * D;
+  for (int d = lane; d < D; d += LCTR_WAVE) {
+    const float g = gradV[base + d] + l2 * V[base + d];
+    const float acc = nV[base + d] + g * g;
+    nV[base + d] = acc;
+    V[base + d] -= lr * g * __frsqrt_rn(acc + eps);
+    gradV[base + d] = 0.f;
+  }
+  if (lane == 0 && W != nullptr) {
+    const float gw = gradW[fid] + l2 * W[fid];
+    const float a = nW[fid] + gw * gw;
+    nW[fid] = a;
+    W[fid] -= lr * gw * __frsqrt_rn(a + eps);
+    gradW[fid] = 0.f;
+  }
+}
+
+__global__ void sparse_ftrl_apply_g_kernel(
+    const int* __restrict__ uniq, const int* __restrict__ count,
+    float* __restrict__ W, float* __restrict__ V, float* __restrict__ zW,
+    float* __restrict__ nW, float* __restrict__ zV, float* __restrict__ nV,
+    float* __restrict__ gradW, float* __restrict__ gradV, float alpha,
+    float beta, float l1, float l2, int capacity, int D) {
+  const int i = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  const int lane = threadIdx.x & (LCTR_WAVE - 1);
+  if (i >= capacity || i >= *count) return;
+  const int fid = uniq[i];
+  const size_t base = (size_t)fid * D;
+  for (int d = lane; d < D; d += LCTR_WAVE) {
+    ftrl_update_g(&V[base + d], &zV[base + d], &nV[base + d], gradV[base + d],
+                  alpha, beta, l1, l2);
+    gradV[base + d] = 0.f;
+  }
+  if (lane == 0 && W != nullptr) {
+    ftrl_update_g(&W[fid], &zW[fid], &nW[fid], gradW[fid], alpha, beta, l1,
+                  l2);
+    gradW[fid] = 0.f;
+  }
+}
+
+void sparse_adagrad_apply_launch(const int* uniq, const int* count, float* W,
+                                 float* V, float* nW, float* nV, float* gradW,
+                                 float* gradV, float lr, float eps, float l2,
+                                 int capacity, int D, hipStream_t stream) {
+  dim3 block(256);
+  dim3 grid((capacity + 3) / 4);
+  hipLaunchKernelGGL(sparse_adagrad_apply_g_kernel, grid, block, 0, stream,
+                     uniq, count, W, V, nW, nV, gradW, gradV, lr, eps, l2,
+                     capacity, D);
+}
+
+void sparse_ftrl_apply_launch(const int* uniq, const int* count, float* W,
+                              float* V, float* zW, float* nW, float* zV,
+                              float* nV, float* gradW, float* gradV,
+                              float alpha, float beta, float l1, float l2,
+                              int capacity, int D, hipStream_t stream) {
+  dim3 block(256);
+  dim3 grid((capacity + 3) / 4);
+  hipLaunchKernelGGL(sparse_ftrl_apply_g_kernel, grid, block, 0, stream, uniq,
+                     count, W, V, zW, nW, zV, nV, gradW, gradV, alpha, beta,
+                     l1, l2, capacity, D);
+}
+
+}  // namespace lightctr
