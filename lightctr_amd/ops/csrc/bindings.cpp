@@ -217,6 +217,188 @@ void fm_ftrl_apply(at::Tensor uniq, at::Tensor count, at::Tensor W,
       (float)l2, (int)uniq.numel(), K, cur_stream());
 }
 
+// ---- dense NN ops ----
+
+at::Tensor gemm_bf16(at::Tensor A, at::Tensor Bst,
+                     c10::optional<at::Tensor> bias, int64_t M, int64_t N,
+                     int64_t K, int64_t transA, int64_t transB, int64_t act,
+                     bool emit_bf16) {
+  CHK(A.is_cuda() && A.scalar_type() == at::kBFloat16, "A must be bf16 GPU");
+  CHK(Bst.is_cuda() && Bst.scalar_type() == at::kBFloat16, "Bst bf16");
+  auto C = at::empty({M, N}, A.options().dtype(at::kFloat));
+  at::Tensor Cbf;
+  void* cbf_ptr = nullptr;
+  if (emit_bf16) {
+    Cbf = at::empty({M, N}, A.options());
+    cbf_ptr = Cbf.data_ptr();
+  }
+  const float* bias_ptr =
+      bias.has_value() ? bias->data_ptr<float>() : nullptr;
+  lightctr::gemm_bf16_launch(A.data_ptr(), Bst.data_ptr(), bias_ptr,
+                             C.data_ptr<float>(), cbf_ptr, (int)M, (int)N,
+                             (int)K, (int)transA, (int)transB, (int)act,
+                             cur_stream());
+  return C;
+}
+
+std::vector<at::Tensor> gemm_bf16_full(at::Tensor A, at::Tensor Bst,
+                                       c10::optional<at::Tensor> bias,
+                                       int64_t M, int64_t N, int64_t K,
+                                       int64_t transA, int64_t transB,
+                                       int64_t act) {
+  CHK(A.is_cuda() && A.scalar_type() == at::kBFloat16, "A must be bf16 GPU");
+  CHK(Bst.is_cuda() && Bst.scalar_type() == at::kBFloat16, "Bst bf16");
+  auto C = at::empty({M, N}, A.options().dtype(at::kFloat));
+  auto Cbf = at::empty({M, N}, A.options());
+  const float* bias_ptr =
+      bias.has_value() ? bias->data_ptr<float>() : nullptr;
+  lightctr::gemm_bf16_launch(A.data_ptr(), Bst.data_ptr(), bias_ptr,
+                             C.data_ptr<float>(), Cbf.data_ptr(), (int)M,
+                             (int)N, (int)K, (int)transA, (int)transB,
+                             (int)act, cur_stream());
+  return {C, Cbf};
+}
+
+std::vector<at::Tensor> act_backward(at::Tensor dY, at::Tensor Y,
+                                     int64_t act) {
+  check_cuda_f32(dY, "dY");
+  auto dZ = at::empty_like(dY);
+  auto dZbf = at::empty_like(dY, dY.options().dtype(at::kBFloat16));
+  lightctr::act_backward_launch(dY.data_ptr<float>(), Y.data_ptr<float>(),
+                                dZ.data_ptr<float>(), dZbf.data_ptr(),
+                                dY.numel(), (int)act, cur_stream());
+  return {dZ, dZbf};
+}
+
+at::Tensor colsum(at::Tensor dZ) {
+  check_cuda_f32(dZ, "dZ");
+  const int M = (int)dZ.size(0), N = (int)dZ.size(1);
+  auto db = at::empty({N}, dZ.options());
+  lightctr::colsum_launch(dZ.data_ptr<float>(), db.data_ptr<float>(), M, N,
+                          cur_stream());
+  return db;
+}
+
+at::Tensor to_bf16(at::Tensor x) {
+  check_cuda_f32(x, "x");
+  auto y = at::empty_like(x, x.options().dtype(at::kBFloat16));
+  lightctr::to_bf16_launch(x.data_ptr<float>(), y.data_ptr(), x.numel(),
+                           cur_stream());
+  return y;
+}
+
+void dense_adam(at::Tensor W, at::Tensor grad, at::Tensor m, at::Tensor v,
+                c10::optional<at::Tensor> Wbf, c10::optional<at::Tensor> Wtbf,
+                double lr, double beta1, double beta2, double eps,
+                int64_t step, double l2) {
+  check_cuda_f32(W, "W");
+  const int rows = (int)W.size(0);
+  const int cols = (int)(W.numel() / W.size(0));
+  const float bc1 = 1.f - powf((float)beta1, (float)step);
+  const float bc2 = 1.f - powf((float)beta2, (float)step);
+  lightctr::dense_adam_launch(
+      W.data_ptr<float>(), grad.data_ptr<float>(), m.data_ptr<float>(),
+      v.data_ptr<float>(), Wbf.has_value() ? Wbf->data_ptr() : nullptr,
+      Wtbf.has_value() ? Wtbf->data_ptr() : nullptr, rows, cols, (float)lr,
+      (float)beta1, (float)beta2, (float)eps, bc1, bc2, (float)l2,
+      cur_stream());
+}
+
+void dense_adagrad(at::Tensor W, at::Tensor grad, at::Tensor n,
+                   c10::optional<at::Tensor> Wbf,
+                   c10::optional<at::Tensor> Wtbf, double lr, double eps,
+                   double l2) {
+  check_cuda_f32(W, "W");
+  const int rows = (int)W.size(0);
+  const int cols = (int)(W.numel() / W.size(0));
+  lightctr::dense_adagrad_launch(
+      W.data_ptr<float>(), grad.data_ptr<float>(), n.data_ptr<float>(),
+      Wbf.has_value() ? Wbf->data_ptr() : nullptr,
+      Wtbf.has_value() ? Wtbf->data_ptr() : nullptr, rows, cols, (float)lr,
+      (float)eps, (float)l2, cur_stream());
+}
+
+// ---- embedding / NFM ----
+
+at::Tensor embed_gather(at::Tensor row_ptr, at::Tensor fids, at::Tensor vals,
+                        at::Tensor E, int64_t nf) {
+  check_cuda_i32(row_ptr, "row_ptr");
+  check_cuda_f32(E, "E");
+  const int B = (int)row_ptr.numel() - 1;
+  const int K = (int)E.size(1);
+  auto out = at::empty({B, nf * K}, E.options().dtype(at::kBFloat16));
+  lightctr::embed_gather_launch(row_ptr.data_ptr<int>(), fids.data_ptr<int>(),
+                                vals.data_ptr<float>(), E.data_ptr<float>(),
+                                out.data_ptr(), (int)nf, B, K, cur_stream());
+  return out;
+}
+
+std::vector<at::Tensor> embed_backward_emit(at::Tensor row_ptr,
+                                            at::Tensor vals, at::Tensor dOut,
+                                            at::Tensor dwide, int64_t nf,
+                                            int64_t K) {
+  check_cuda_i32(row_ptr, "row_ptr");
+  check_cuda_f32(dOut, "dOut");
+  const int B = (int)row_ptr.numel() - 1;
+  const auto nnz = vals.numel();
+  auto gv = at::empty({nnz, K}, dOut.options());
+  auto gw = at::empty({nnz}, dOut.options());
+  lightctr::embed_backward_emit_launch(
+      row_ptr.data_ptr<int>(), vals.data_ptr<float>(), dOut.data_ptr<float>(),
+      dwide.data_ptr<float>(), gv.data_ptr<float>(), gw.data_ptr<float>(),
+      (int)nf, B, (int)K, cur_stream());
+  return {gw, gv};
+}
+
+at::Tensor wide_forward(at::Tensor row_ptr, at::Tensor fids, at::Tensor vals,
+                        at::Tensor W) {
+  check_cuda_i32(row_ptr, "row_ptr");
+  check_cuda_f32(W, "W");
+  const int B = (int)row_ptr.numel() - 1;
+  auto wide = at::empty({B}, W.options());
+  lightctr::wide_forward_launch(row_ptr.data_ptr<int>(), fids.data_ptr<int>(),
+                                vals.data_ptr<float>(), W.data_ptr<float>(),
+                                wide.data_ptr<float>(), B, cur_stream());
+  return wide;
+}
+
+std::vector<at::Tensor> nfm_forward(at::Tensor row_ptr, at::Tensor fids,
+                                    at::Tensor vals, at::Tensor W,
+                                    at::Tensor V) {
+  check_cuda_i32(row_ptr, "row_ptr");
+  check_cuda_f32(V, "V");
+  const int B = (int)row_ptr.numel() - 1;
+  const int K = (int)V.size(1);
+  auto wide = at::empty({B}, W.options());
+  auto sumVX = at::empty({B, K}, V.options());
+  auto vec = at::empty({B, K}, V.options());
+  auto vec_bf = at::empty({B, K}, V.options().dtype(at::kBFloat16));
+  lightctr::nfm_forward_launch(row_ptr.data_ptr<int>(), fids.data_ptr<int>(),
+                               vals.data_ptr<float>(), W.data_ptr<float>(),
+                               V.data_ptr<float>(), wide.data_ptr<float>(),
+                               sumVX.data_ptr<float>(), vec.data_ptr<float>(),
+                               vec_bf.data_ptr(), B, K, cur_stream());
+  return {wide, sumVX, vec, vec_bf};
+}
+
+std::vector<at::Tensor> nfm_backward_emit(at::Tensor row_ptr, at::Tensor fids,
+                                          at::Tensor vals, at::Tensor V,
+                                          at::Tensor sumVX, at::Tensor dvec,
+                                          at::Tensor dwide) {
+  check_cuda_i32(row_ptr, "row_ptr");
+  const int B = (int)row_ptr.numel() - 1;
+  const int K = (int)V.size(1);
+  const auto nnz = fids.numel();
+  auto gw = at::empty({nnz}, V.options());
+  auto gv = at::empty({nnz, K}, V.options());
+  lightctr::nfm_backward_emit_launch(
+      row_ptr.data_ptr<int>(), fids.data_ptr<int>(), vals.data_ptr<float>(),
+      V.data_ptr<float>(), sumVX.data_ptr<float>(), dvec.data_ptr<float>(),
+      dwide.data_ptr<float>(), gw.data_ptr<float>(), gv.data_ptr<float>(), B,
+      K, cur_stream());
+  return {gw, gv};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -234,6 +416,23 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "generic sparse fused Adagrad (runtime D)");
   m.def("sparse_ftrl_apply", &sparse_ftrl_apply,
         "generic sparse fused FTRL (runtime D)");
+  m.def("gemm_bf16", &gemm_bf16, "MFMA bf16 GEMM (C fp32), fused bias+act");
+  m.def("gemm_bf16_full", &gemm_bf16_full,
+        "MFMA bf16 GEMM returning (C fp32, C bf16)");
+  m.def("act_backward", &act_backward, "dZ = dY*act'(Y), + bf16 mirror");
+  m.def("colsum", &colsum, "bias gradient column sum");
+  m.def("to_bf16", &to_bf16, "f32 -> bf16 convert");
+  m.def("dense_adam", &dense_adam, "dense Adam + bf16 mirror refresh");
+  m.def("dense_adagrad", &dense_adagrad,
+        "dense Adagrad + bf16 mirror refresh");
+  m.def("embed_gather", &embed_gather, "concat embedding gather -> bf16");
+  m.def("embed_backward_emit", &embed_backward_emit,
+        "per-entry embedding grads (gw, gv) for sorted apply");
+  m.def("wide_forward", &wide_forward, "LR wide forward sum");
+  m.def("nfm_forward", &nfm_forward,
+        "NFM bi-interaction fwd (wide, sumVX, vec, vec_bf16)");
+  m.def("nfm_backward_emit", &nfm_backward_emit,
+        "NFM per-entry grads for sorted apply");
   m.def("bitmap_compact", &bitmap_compact, "touched bitmap -> fid list");
   m.def("fm_adagrad_apply", &fm_adagrad_apply, "sparse fused Adagrad");
   m.def("fm_ftrl_apply", &fm_ftrl_apply, "sparse fused FTRL-proximal");

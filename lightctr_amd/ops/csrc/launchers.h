@@ -62,4 +62,44 @@ void fm_ftrl_apply_launch(const int* uniq, const int* count, float* W,
                           float l1, float l2, int capacity, int K,
                           ihipStream_t* stream);
 
+// --- gemm_kernels.hip ---
+void gemm_bf16_launch(const void* A, const void* Bst, const float* bias,
+                      float* C, void* Cbf, int M, int N, int K, int transA,
+                      int transB, int act, ihipStream_t* stream);
+
+// --- nn_kernels.hip ---
+void act_backward_launch(const float* dY, const float* Y, float* dZ,
+                         void* dZbf, long n, int act, ihipStream_t* stream);
+void colsum_launch(const float* dZ, float* db, int M, int N,
+                   ihipStream_t* stream);
+void to_bf16_launch(const float* x, void* y, long n, ihipStream_t* stream);
+void dense_adam_launch(float* W, const float* grad, float* m_t, float* v_t,
+                       void* Wbf, void* Wtbf, int rows, int cols, float lr,
+                       float beta1, float beta2, float eps, float bc1,
+                       float bc2, float l2, ihipStream_t* stream);
+void dense_adagrad_launch(float* W, const float* grad, float* n_t, void* Wbf,
+                          void* Wtbf, int rows, int cols, float lr, float eps,
+                          float l2, ihipStream_t* stream);
+
+// --- embed_kernels.hip ---
+void embed_gather_launch(const int* row_ptr, const int* fids,
+                         const float* vals, const float* E, void* out_bf,
+                         int nf, int B, int K, ihipStream_t* stream);
+void embed_backward_emit_launch(const int* row_ptr, const float* vals,
+                                const float* dOut, const float* dwide,
+                                float* gv, float* gw, int nf, int B, int K,
+                                ihipStream_t* stream);
+void wide_forward_launch(const int* row_ptr, const int* fids,
+                         const float* vals, const float* W, float* wide,
+                         int B, ihipStream_t* stream);
+void nfm_forward_launch(const int* row_ptr, const int* fids, const float* vals,
+                        const float* W, const float* V, float* wide,
+                        float* sumVX, float* vec, void* vec_bf, int B, int K,
+                        ihipStream_t* stream);
+void nfm_backward_emit_launch(const int* row_ptr, const int* fids,
+                              const float* vals, const float* V,
+                              const float* sumVX, const float* dvec,
+                              const float* dwide, float* gw, float* gv, int B,
+                              int K, ihipStream_t* stream);
+
 }  // namespace lightctr

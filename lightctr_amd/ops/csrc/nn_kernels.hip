@@ -1,0 +1,155 @@
+// Dense-NN support kernels for MI355X/gfx950: activation backward, bias
+// gradient (column sum), dense fused optimizers that keep fp32 master
+// weights and maintain the bf16 + bf16-transposed GEMM operand mirrors.
+//
+// Together with gemm_kernels.hip these replace the reference's
+// Fully_Conn_Layer forward/backward + per-layer updaters
+// (/root/reference/LightCTR/train/layer/fullyconnLayer.h:95-237,
+// util/momentumUpdater.h Adam :113-215) in CDNA4 form.
+#include "common.h"
+
+namespace lightctr {
+
+// dZ = dY * act'(Y) where Y is the post-activation output; also emits a
+// bf16 mirror of dZ (GEMM operand for dgrad/wgrad). act: 0 none, 1 relu,
+// 2 sigmoid.
+__global__ void act_backward_kernel(const float* __restrict__ dY,
+                                    const float* __restrict__ Y,
+                                    float* __restrict__ dZ,
+                                    __bf16* __restrict__ dZbf, long n,
+                                    int act) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  float g = dY[i];
+  if (act == 1) {
+    g = Y[i] > 0.f ? g : 0.f;
+  } else if (act == 2) {
+    const float y = Y[i];
+    g *= y * (1.f - y);
+  }
+  dZ[i] = g;
+  if (dZbf) dZbf[i] = (__bf16)g;
+}
+
+// db[n] = sum_m dZ[m,n]. One wave per column, grid-stride over rows.
+__global__ void colsum_kernel(const float* __restrict__ dZ,
+                              float* __restrict__ db, int M, int N) {
+  const int col = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  const int lane = threadIdx.x & 63;
+  if (col >= N) return;
+  float s = 0.f;
+  for (int m = lane; m < M; m += 64) s += dZ[(size_t)m * N + col];
+  s = wave_reduce_sum(s);
+  if (lane == 0) db[col] = s;
+}
+
+// f32 -> bf16 convert (vectorized x4)
+__global__ void to_bf16_kernel(const float* __restrict__ x,
+                               __bf16* __restrict__ y, long n) {
+  const long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  if (i + 3 < n) {
+    const float4 v = *(const float4*)&x[i];
+    y[i] = (__bf16)v.x;
+    y[i + 1] = (__bf16)v.y;
+    y[i + 2] = (__bf16)v.z;
+    y[i + 3] = (__bf16)v.w;
+  } else {
+    for (long j = i; j < n; ++j) y[j] = (__bf16)x[j];
+  }
+}
+
+// Dense Adam on fp32 master W[out,in]; refreshes bf16 mirror Wbf[out,in]
+// and transposed mirror Wtbf[in,out] (both GEMM operands).
+__global__ void dense_adam_kernel(float* __restrict__ W,
+                                  const float* __restrict__ grad,
+                                  float* __restrict__ m_t,
+                                  float* __restrict__ v_t,
+                                  __bf16* __restrict__ Wbf,
+                                  __bf16* __restrict__ Wtbf, int rows,
+                                  int cols, float lr, float beta1, float beta2,
+                                  float eps, float bc1, float bc2, float l2) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long n = (long)rows * cols;
+  if (i >= n) return;
+  const float g = grad[i] + l2 * W[i];
+  const float m = beta1 * m_t[i] + (1.f - beta1) * g;
+  const float v = beta2 * v_t[i] + (1.f - beta2) * g * g;
+  m_t[i] = m;
+  v_t[i] = v;
+  const float w = W[i] - lr * (m / bc1) / (sqrtf(v / bc2) + eps);
+  W[i] = w;
+  if (Wbf) Wbf[i] = (__bf16)w;
+  if (Wtbf) {
+    const long r = i / cols, c = i % cols;
+    Wtbf[c * rows + r] = (__bf16)w;
+  }
+}
+
+// Dense Adagrad (reference default updater), same mirror maintenance.
+__global__ void dense_adagrad_kernel(float* __restrict__ W,
+                                     const float* __restrict__ grad,
+                                     float* __restrict__ n_t,
+                                     __bf16* __restrict__ Wbf,
+                                     __bf16* __restrict__ Wtbf, int rows,
+                                     int cols, float lr, float eps, float l2) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long n = (long)rows * cols;
+  if (i >= n) return;
+  const float g = grad[i] + l2 * W[i];
+  const float acc = n_t[i] + g * g;
+  n_t[i] = acc;
+  const float w = W[i] - lr * g * __frsqrt_rn(acc + eps);
+  W[i] = w;
+  if (Wbf) Wbf[i] = (__bf16)w;
+  if (Wtbf) {
+    const long r = i / cols, c = i % cols;
+    Wtbf[c * rows + r] = (__bf16)w;
+  }
+}
+
+void act_backward_launch(const float* dY, const float* Y, float* dZ,
+                         void* dZbf, long n, int act, hipStream_t stream) {
+  dim3 block(256);
+  dim3 grid((unsigned)((n + 255) / 256));
+  hipLaunchKernelGGL(act_backward_kernel, grid, block, 0, stream, dY, Y, dZ,
+                     (__bf16*)dZbf, n, act);
+}
+
+void colsum_launch(const float* dZ, float* db, int M, int N,
+                   hipStream_t stream) {
+  dim3 block(256);
+  dim3 grid((N + 3) / 4);
+  hipLaunchKernelGGL(colsum_kernel, grid, block, 0, stream, dZ, db, M, N);
+}
+
+void to_bf16_launch(const float* x, void* y, long n, hipStream_t stream) {
+  dim3 block(256);
+  dim3 grid((unsigned)((n / 4 + 256) / 256) + 1);
+  hipLaunchKernelGGL(to_bf16_kernel, grid, block, 0, stream, x, (__bf16*)y,
+                     n);
+}
+
+void dense_adam_launch(float* W, const float* grad, float* m_t, float* v_t,
+                       void* Wbf, void* Wtbf, int rows, int cols, float lr,
+                       float beta1, float beta2, float eps, float bc1,
+                       float bc2, float l2, hipStream_t stream) {
+  const long n = (long)rows * cols;
+  dim3 block(256);
+  dim3 grid((unsigned)((n + 255) / 256));
+  hipLaunchKernelGGL(dense_adam_kernel, grid, block, 0, stream, W, grad, m_t,
+                     v_t, (__bf16*)Wbf, (__bf16*)Wtbf, rows, cols, lr, beta1,
+                     beta2, eps, bc1, bc2, l2);
+}
+
+void dense_adagrad_launch(float* W, const float* grad, float* n_t, void* Wbf,
+                          void* Wtbf, int rows, int cols, float lr, float eps,
+                          float l2, hipStream_t stream) {
+  const long n = (long)rows * cols;
+  dim3 block(256);
+  dim3 grid((unsigned)((n + 255) / 256));
+  hipLaunchKernelGGL(dense_adagrad_kernel, grid, block, 0, stream, W, grad,
+                     n_t, (__bf16*)Wbf, (__bf16*)Wtbf, rows, cols, lr, eps,
+                     l2);
+}
+
+}  // namespace lightctr

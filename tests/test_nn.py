@@ -1,0 +1,212 @@
+"""Dense engine tests: MFMA GEMM parity vs torch, MLP training, NFM / W&D."""
+
+import pytest
+import torch
+
+from lightctr_amd.data import LibffmDataset
+from lightctr_amd.data.synthetic import SyntheticCriteo
+from lightctr_amd.models.mlp import MLP, DenseLayer
+from lightctr_amd.models.nfm import NFMHyper, NFMModel, NFMTrainer
+from lightctr_amd.models.wide_deep import (WideDeepHyper, WideDeepModel,
+                                           WideDeepTrainer)
+
+
+def test_mlp_cpu_learns_xor_like():
+    g = torch.Generator().manual_seed(0)
+    X = torch.randn(512, 8, generator=g)
+    y = ((X[:, 0] * X[:, 1]) > 0).float().unsqueeze(1)
+    mlp = MLP([8, 32, 1], optimizer="adam", lr=1e-2, seed=1, device="cpu")
+    for _ in range(300):
+        out = mlp.forward(X)
+        p = torch.sigmoid(out)
+        dloss = (p - y) / 512
+        mlp.backward(dloss)
+        mlp.apply_grads()
+    acc = ((torch.sigmoid(mlp.forward(X, train=False)) > 0.5) == (y > 0.5))
+    assert acc.float().mean() > 0.9
+
+
+def test_dense_layer_backward_matches_autograd():
+    g = torch.Generator().manual_seed(3)
+    x = torch.randn(32, 16, generator=g)
+    layer = DenseLayer(16, 8, act="relu", optimizer="adam", device="cpu",
+                       seed=5)
+    W = layer.W.clone().requires_grad_(True)
+    b = layer.b.clone().requires_grad_(True)
+    y_ref = torch.relu(x @ W.t() + b)
+    dy = torch.randn(32, 8, generator=g)
+    y_ref.backward(dy)
+    y, _ = layer.forward(x)
+    dx = layer.backward(dy)
+    assert torch.allclose(y, y_ref.detach(), atol=1e-6)
+    assert torch.allclose(layer._dW, W.grad, atol=1e-5)
+    assert torch.allclose(layer._db, b.grad, atol=1e-5)
+    xg = x.clone().requires_grad_(True)
+    torch.relu(xg @ W.detach().t() + b.detach()).backward(dy)
+    assert torch.allclose(dx, xg.grad, atol=1e-5)
+
+
+def test_nfm_cpu_convergence():
+    gen = SyntheticCriteo(num_features=1 << 13, seed=31)
+    row_ptr, fields, fids, vals, labels = gen.batch(1024)
+    ds = LibffmDataset(row_ptr, fields, fids, vals, labels)
+    h = NFMHyper(num_features=1 << 13, k=8, hidden=(32,))
+    tr = NFMTrainer(ds, h, device="cpu", batch_size=256, epochs=3)
+    m0 = tr.evaluate()
+    tr.train(log=None)
+    m1 = tr.evaluate()
+    assert m1["logloss"] < m0["logloss"]
+    assert m1["auc"] > 0.55
+
+
+def test_widedeep_cpu_convergence():
+    gen = SyntheticCriteo(num_features=1 << 13, seed=37)
+    row_ptr, fields, fids, vals, labels = gen.batch(1024)
+    ds = LibffmDataset(row_ptr, fields, fids, vals, labels)
+    h = WideDeepHyper(num_features=1 << 13, num_fields=39, k=8,
+                      hidden=(64, 32))
+    tr = WideDeepTrainer(ds, h, device="cpu", batch_size=256, epochs=3)
+    m0 = tr.evaluate()
+    tr.train(log=None)
+    m1 = tr.evaluate()
+    assert m1["logloss"] < m0["logloss"]
+    assert m1["auc"] > 0.55
+
+
+# ---------------- GPU ----------------
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("shape", [(128, 128, 64), (256, 384, 156),
+                                   (1000, 250, 100), (64, 1, 256),
+                                   (513, 129, 65)])
+def test_gemm_bf16_parity(shape):
+    """MFMA GEMM vs torch bf16 matmul (fp32 accumulate) incl. odd tails.
+
+    Asymmetric random operands (transpose-detecting, guide §5.4 rule 16)."""
+    from lightctr_amd.ops import hip_ops
+
+    M, N, K = shape
+    g = torch.Generator().manual_seed(M + N + K)
+    A = (torch.randn(M, K, generator=g) * 0.5).to(torch.bfloat16).cuda()
+    Bst = (torch.randn(N, K, generator=g) * 0.5).to(torch.bfloat16).cuda()
+    bias = torch.randn(N, generator=g).cuda()
+    C = hip_ops.gemm_bf16(A, Bst, bias, M, N, K, 0, 0, 0, False)
+    ref = (A.float() @ Bst.float().t()) + bias
+    assert torch.allclose(C, ref, atol=2e-2 * K ** 0.5, rtol=1e-2), \
+        (C - ref).abs().max()
+
+
+@pytest.mark.gpu
+def test_gemm_bf16_transA_parity():
+    from lightctr_amd.ops import hip_ops
+
+    M, N, K = 200, 96, 320  # A stored [K, M]
+    g = torch.Generator().manual_seed(9)
+    Ast = (torch.randn(K, M, generator=g) * 0.5).to(torch.bfloat16).cuda()
+    Bst = (torch.randn(N, K, generator=g) * 0.5).to(torch.bfloat16).cuda()
+    C = hip_ops.gemm_bf16(Ast, Bst, None, M, N, K, 1, 0, 0, False)
+    ref = Ast.float().t() @ Bst.float().t()
+    assert torch.allclose(C, ref, atol=2e-2 * K ** 0.5, rtol=1e-2)
+
+
+@pytest.mark.gpu
+def test_gemm_bf16_transB_parity():
+    from lightctr_amd.ops import hip_ops
+
+    M, N, K = 160, 200, 192  # B stored [K, N] (wgrad operand layout)
+    g = torch.Generator().manual_seed(10)
+    A = (torch.randn(M, K, generator=g) * 0.5).to(torch.bfloat16).cuda()
+    B = (torch.randn(K, N, generator=g) * 0.5).to(torch.bfloat16).cuda()
+    C = hip_ops.gemm_bf16(A, B, None, M, N, K, 0, 1, 0, False)
+    ref = A.float() @ B.float()
+    assert torch.allclose(C, ref, atol=2e-2 * K ** 0.5, rtol=1e-2)
+
+
+@pytest.mark.gpu
+def test_gemm_relu_bias_epilogue():
+    from lightctr_amd.ops import hip_ops
+
+    M, N, K = 256, 64, 128
+    g = torch.Generator().manual_seed(11)
+    A = torch.randn(M, K, generator=g).to(torch.bfloat16).cuda()
+    Bst = torch.randn(N, K, generator=g).to(torch.bfloat16).cuda()
+    bias = torch.randn(N, generator=g).cuda()
+    C, Cbf = hip_ops.gemm_bf16_full(A, Bst, bias, M, N, K, 0, 0, 1)
+    ref = torch.relu(A.float() @ Bst.float().t() + bias)
+    assert torch.allclose(C, ref, atol=0.2, rtol=1e-2)
+    assert torch.allclose(Cbf.float(), C, atol=0.1, rtol=1e-2)
+
+
+@pytest.mark.gpu
+def test_mlp_gpu_matches_cpu():
+    """One fwd/bwd/step of the bf16 MFMA MLP vs the fp32 CPU oracle."""
+    g = torch.Generator().manual_seed(4)
+    x = torch.randn(256, 64, generator=g)
+    dy = torch.randn(256, 1, generator=g) * 0.1
+    cpu = MLP([64, 32, 1], optimizer="adam", lr=1e-3, seed=7, device="cpu")
+    gpu = MLP([64, 32, 1], optimizer="adam", lr=1e-3, seed=7, device="cuda:0")
+    for lc, lg in zip(cpu.layers, gpu.layers):
+        lg.W.copy_(lc.W)
+        lg.b.copy_(lc.b)
+        lg.Wbf.copy_(lg.W.to(torch.bfloat16))
+        lg.Wtbf.copy_(lg.W.t().contiguous().to(torch.bfloat16))
+    y_cpu = cpu.forward(x)
+    y_gpu = gpu.forward(x.cuda().to(torch.bfloat16))
+    assert torch.allclose(y_gpu.cpu(), y_cpu, atol=0.08, rtol=0.05), \
+        (y_gpu.cpu() - y_cpu).abs().max()
+    dx_cpu = cpu.backward(dy)
+    dx_gpu = gpu.backward(dy.cuda())
+    assert torch.allclose(dx_gpu.cpu(), dx_cpu, atol=0.02, rtol=0.05), \
+        (dx_gpu.cpu() - dx_cpu).abs().max()
+    cpu.apply_grads()
+    gpu.apply_grads()
+    for lc, lg in zip(cpu.layers, gpu.layers):
+        assert torch.allclose(lg.W.cpu(), lc.W, atol=5e-3), \
+            (lg.W.cpu() - lc.W).abs().max()
+
+
+@pytest.mark.gpu
+def test_nfm_gpu_convergence():
+    gen = SyntheticCriteo(num_features=1 << 15, seed=41, device="cuda:0")
+    h = NFMHyper(num_features=1 << 15, k=16, hidden=(64,))
+    model = NFMModel(h, device="cuda:0")
+    losses = []
+    for _ in range(25):
+        row_ptr, fields, fids, vals, labels = gen.batch(4096)
+        loss = model.train_step(row_ptr, fids, vals, labels)
+        losses.append(float(loss.mean()))
+    assert losses[-1] < losses[0] * 0.99, losses[:3] + losses[-3:]
+
+
+@pytest.mark.gpu
+def test_widedeep_gpu_convergence():
+    gen = SyntheticCriteo(num_features=1 << 15, seed=43, device="cuda:0")
+    h = WideDeepHyper(num_features=1 << 15, num_fields=39, k=16,
+                      hidden=(256, 128))
+    model = WideDeepModel(h, device="cuda:0")
+    losses = []
+    for _ in range(25):
+        row_ptr, fields, fids, vals, labels = gen.batch(4096)
+        loss = model.train_step(row_ptr, fids, vals, labels)
+        losses.append(float(loss.mean()))
+    assert losses[-1] < losses[0] * 0.99, losses[:3] + losses[-3:]
+
+
+@pytest.mark.gpu
+def test_nfm_gpu_forward_parity():
+    """GPU fused bi-interaction vs CPU reference pieces."""
+    from lightctr_amd.ops import hip_ops
+
+    gen = SyntheticCriteo(num_features=1 << 12, seed=47, device="cuda:0")
+    row_ptr, fields, fids, vals, labels = gen.batch(64)
+    F, K = 1 << 12, 8
+    g = torch.Generator().manual_seed(5)
+    W = torch.randn(F, generator=g).cuda()
+    V = (torch.randn(F, K, generator=g) * 0.1).cuda()
+    wide, sumVX, vec, vec_bf = hip_ops.nfm_forward(row_ptr, fids, vals, W, V)
+    from lightctr_amd.ops import fm_ref
+
+    pred_ref, sumVX_ref = fm_ref.fm_forward_ref(row_ptr, fids, vals, W, V)
+    assert torch.allclose(sumVX, sumVX_ref, atol=1e-4)
+    # wide + vec.sum() must equal the FM prediction
+    assert torch.allclose(wide + vec.sum(dim=1), pred_ref, atol=1e-3)
