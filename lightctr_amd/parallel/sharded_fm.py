@@ -1,0 +1,207 @@
+"""Hash-sharded FM across N GPUs — the MI355X-native replacement of the
+reference's Parameter-Server + ConsistentHash DHT
+(/root/reference/LightCTR/distribut/{paramserver.h,consistent_hash.h,
+push.h,pull.h}): instead of DHT-over-ZeroMQ, the feature table lives
+sharded by `fid % world` across the HBM of all ranks, and each train step
+exchanges deduplicated (fid -> embedding row) requests and gradients with
+two RCCL all-to-alls over xGMI (the "batched parameter request" the
+reference README describes, without per-row round trips).
+
+Every rank is simultaneously worker AND parameter shard (the reference's
+4 PS + 4 worker split maps onto 8 ranks each playing both roles — same
+aggregate layout, no idle GPUs). Synchronous flavor: gradient exchange
+completes before the owner applies its fused optimizer, so training is
+exactly minibatch SGD/FTRL on the union batch (the async SSP/DCASGD
+semantics of the reference live in lightctr_amd.parallel.ps).
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.distributed as dist
+
+from ..models.fm import FMHyper
+from ..ops import fm_ref
+from ..ops._extension import require_hip_ops
+
+
+class ShardedFMModel:
+    def __init__(self, hyper: FMHyper, device: str = "cpu", group=None):
+        self.h = hyper
+        self.device = torch.device(device)
+        self.group = group
+        self.world = dist.get_world_size(group)
+        self.rank = dist.get_rank(group)
+        F, K = hyper.num_features, hyper.k
+        self.F_local = (F + self.world - 1) // self.world
+        g = torch.Generator().manual_seed(hyper.seed + 17 * self.rank)
+        dev = self.device
+        # shard tensors: local row i holds global feature i*world + rank
+        self.W = torch.zeros(self.F_local, device=dev)
+        self.V = (torch.randn(self.F_local, K, generator=g)
+                  * hyper.init_sigma).to(dev)
+        self.gradW = torch.zeros_like(self.W)
+        self.gradV = torch.zeros_like(self.V)
+        self.nW = torch.zeros_like(self.W)
+        self.nV = torch.zeros_like(self.V)
+        if hyper.optimizer == "ftrl":
+            self.zW = torch.zeros_like(self.W)
+            self.zV = torch.zeros_like(self.V)
+        nwords = (self.F_local + 63) // 64
+        self.touched = torch.zeros(nwords, dtype=torch.int64, device=dev)
+        self.uniq = torch.zeros(self.F_local, dtype=torch.int32, device=dev)
+        self.count = torch.zeros(1, dtype=torch.int32, device=dev)
+        self._use_hip = dev.type == "cuda"
+        if self._use_hip:
+            require_hip_ops()
+
+    # ---- parameter exchange ----
+    def _exchange(self, send: torch.Tensor, send_counts, recv_counts):
+        """all_to_all_single with known splits; send/recv along dim 0."""
+        out_shape = (sum(recv_counts),) + tuple(send.shape[1:])
+        out = torch.empty(out_shape, dtype=send.dtype, device=send.device)
+        dist.all_to_all_single(out, send.contiguous(),
+                               output_split_sizes=recv_counts,
+                               input_split_sizes=send_counts,
+                               group=self.group)
+        return out
+
+    def train_step(self, row_ptr, fids, vals, labels) -> torch.Tensor:
+        """One synchronous sharded step on this rank's local batch."""
+        h = self.h
+        B = row_ptr.numel() - 1
+        world = self.world
+        # global mean over the union batch (all ranks' batches)
+        scale = 1.0 / (B * world)
+
+        # 1. dedup local batch; express batch in local mini-table ids
+        uniq, inverse = torch.unique(fids, return_inverse=True)
+        U = uniq.numel()
+        owner = (uniq.long() % world)
+        order = torch.argsort(owner, stable=True)
+        uniq_o = uniq[order]
+        send_counts = torch.bincount(owner, minlength=world).cpu().tolist()
+        cnt_t = torch.tensor(send_counts, dtype=torch.int64)
+        recv_cnt_t = torch.empty(world, dtype=torch.int64)
+        dist.all_to_all_single(recv_cnt_t, cnt_t, group=self.group)
+        recv_counts = recv_cnt_t.tolist()
+
+        # 2. exchange requested fids; owners gather their shard rows
+        req = self._exchange(uniq_o, send_counts, recv_counts)
+        lidx = (req.long() // world)
+        Wv = self.W[lidx]
+        Vv = self.V[lidx]
+
+        # 3. send values back (reverse splits) and un-permute to uniq order
+        Wl_o = self._exchange(Wv, recv_counts, send_counts)
+        Vl_o = self._exchange(Vv, recv_counts, send_counts)
+        Wl = torch.empty_like(Wl_o)
+        Vl = torch.empty_like(Vl_o)
+        Wl[order] = Wl_o
+        Vl[order] = Vl_o
+
+        # 4. local fused compute on the mini-table
+        fids_local = inverse.to(torch.int32)
+        if self._use_hip:
+            ops = require_hip_ops()
+            pred, sumVX = ops.fm_forward(row_ptr, fids_local, vals, Wl, Vl)
+            loss, dpred = ops.logloss_grad(pred, labels, scale)
+            gw, gv = ops.fm_backward_emit(row_ptr, fids_local, vals, Vl,
+                                          sumVX, dpred)
+            sorted_l, perm = torch.sort(fids_local)
+            gWl = torch.zeros(U, device=self.device)
+            gVl = torch.zeros(U, h.k, device=self.device)
+            scratch_bitmap = torch.zeros((U + 63) // 64, dtype=torch.int64,
+                                         device=self.device)
+            ops.fm_sorted_apply(sorted_l, perm, gw, gv, gWl, gVl,
+                                scratch_bitmap)
+        else:
+            pred, sumVX = fm_ref.fm_forward_ref(row_ptr, fids_local, vals,
+                                                Wl, Vl)
+            loss, dpred = fm_ref.logloss_grad_ref(pred, labels, scale)
+            gWl, gVl = fm_ref.fm_backward_ref(row_ptr, fids_local, vals, Vl,
+                                              sumVX, dpred)
+
+        # 5. route gradients to owners (same splits as the request)
+        gW_recv = self._exchange(gWl[order], send_counts, recv_counts)
+        gV_recv = self._exchange(gVl[order], send_counts, recv_counts)
+
+        # 6. owner accumulates (features may arrive from several ranks) and
+        #    applies the fused sparse optimizer on its shard
+        lidx32 = lidx.to(torch.int32)
+        if self._use_hip:
+            ops = require_hip_ops()
+            sorted_own, perm_own = torch.sort(lidx32)
+            ops.fm_sorted_apply(sorted_own, perm_own, gW_recv.contiguous(),
+                                gV_recv.contiguous(), self.gradW, self.gradV,
+                                self.touched)
+            self.count.zero_()
+            ops.bitmap_compact(self.touched, self.uniq, self.count)
+            if h.optimizer == "ftrl":
+                ops.fm_ftrl_apply(self.uniq, self.count, self.W, self.V,
+                                  self.zW, self.nW, self.zV, self.nV,
+                                  self.gradW, self.gradV, h.ftrl_alpha,
+                                  h.ftrl_beta, h.ftrl_l1, h.ftrl_l2)
+            else:
+                ops.fm_adagrad_apply(self.uniq, self.count, self.W, self.V,
+                                     self.nW, self.nV, self.gradW, self.gradV,
+                                     h.lr, h.eps, h.l2)
+        else:
+            self.gradW.index_add_(0, lidx, gW_recv)
+            self.gradV.index_add_(0, lidx, gV_recv)
+            own_uniq = torch.unique(lidx).int()
+            if h.optimizer == "ftrl":
+                fm_ref.ftrl_apply_ref(own_uniq, self.W, self.V, self.zW,
+                                      self.nW, self.zV, self.nV, self.gradW,
+                                      self.gradV, h.ftrl_alpha, h.ftrl_beta,
+                                      h.ftrl_l1, h.ftrl_l2)
+            else:
+                fm_ref.adagrad_apply_ref(own_uniq, self.W, self.V, self.nW,
+                                         self.nV, self.gradW, self.gradV,
+                                         h.lr, h.eps, h.l2)
+        return loss
+
+    # ---- inference (pull-only) ----
+    def predict_proba(self, row_ptr, fids, vals):
+        world = self.world
+        uniq, inverse = torch.unique(fids, return_inverse=True)
+        owner = (uniq.long() % world)
+        order = torch.argsort(owner, stable=True)
+        uniq_o = uniq[order]
+        send_counts = torch.bincount(owner, minlength=world).cpu().tolist()
+        cnt_t = torch.tensor(send_counts, dtype=torch.int64)
+        recv_cnt_t = torch.empty(world, dtype=torch.int64)
+        dist.all_to_all_single(recv_cnt_t, cnt_t, group=self.group)
+        recv_counts = recv_cnt_t.tolist()
+        req = self._exchange(uniq_o, send_counts, recv_counts)
+        lidx = req.long() // world
+        Wl_o = self._exchange(self.W[lidx], recv_counts, send_counts)
+        Vl_o = self._exchange(self.V[lidx], recv_counts, send_counts)
+        Wl = torch.empty_like(Wl_o)
+        Vl = torch.empty_like(Vl_o)
+        Wl[order] = Wl_o
+        Vl[order] = Vl_o
+        fids_local = inverse.to(torch.int32)
+        if self._use_hip:
+            ops = require_hip_ops()
+            pred, _ = ops.fm_forward(row_ptr, fids_local, vals, Wl, Vl)
+        else:
+            pred, _ = fm_ref.fm_forward_ref(row_ptr, fids_local, vals, Wl, Vl)
+        return torch.sigmoid(torch.clamp(pred, -16, 16))
+
+    # ---- sharded checkpoint (one file per rank shard) ----
+    def save(self, path_prefix: str) -> None:
+        d = {"W": self.W, "V": self.V, "nW": self.nW, "nV": self.nV,
+             "rank": self.rank, "world": self.world, "hyper": self.h.__dict__}
+        if self.h.optimizer == "ftrl":
+            d["zW"], d["zV"] = self.zW, self.zV
+        torch.save(d, f"{path_prefix}.shard{self.rank}of{self.world}.pt")
+
+    def load(self, path_prefix: str) -> None:
+        d = torch.load(f"{path_prefix}.shard{self.rank}of{self.world}.pt",
+                       map_location=self.device, weights_only=True)
+        assert d["world"] == self.world, "shard layout mismatch"
+        self.W.copy_(d["W"]); self.V.copy_(d["V"])
+        self.nW.copy_(d["nW"]); self.nV.copy_(d["nV"])
+        if self.h.optimizer == "ftrl" and "zW" in d:
+            self.zW.copy_(d["zW"]); self.zV.copy_(d["zV"])
