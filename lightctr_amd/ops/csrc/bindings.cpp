@@ -399,6 +399,51 @@ std::vector<at::Tensor> nfm_backward_emit(at::Tensor row_ptr, at::Tensor fids,
   return {gw, gv};
 }
 
+// ---- codecs ----
+
+at::Tensor quantile_encode(at::Tensor x, at::Tensor table) {
+  check_cuda_f32(x, "x");
+  check_cuda_f32(table, "table");
+  auto code = at::empty(x.sizes(), x.options().dtype(at::kByte));
+  lightctr::quantile_encode_launch(x.data_ptr<float>(),
+                                   code.data_ptr<unsigned char>(),
+                                   table.data_ptr<float>(),
+                                   (int)table.numel(), x.numel(),
+                                   cur_stream());
+  return code;
+}
+
+at::Tensor quantile_decode(at::Tensor code, at::Tensor table) {
+  check_cuda_f32(table, "table");
+  auto x = at::empty(code.sizes(), table.options());
+  lightctr::quantile_decode_launch(code.data_ptr<unsigned char>(),
+                                   x.data_ptr<float>(),
+                                   table.data_ptr<float>(),
+                                   (int)table.numel(), code.numel(),
+                                   cur_stream());
+  return x;
+}
+
+at::Tensor lowbit_encode(at::Tensor x, double thresh, int64_t bits) {
+  check_cuda_f32(x, "x");
+  const long n = x.numel();
+  const long nw = (n * bits + 31) / 32;
+  auto words = at::empty({nw}, x.options().dtype(at::kInt));
+  lightctr::lowbit_encode_launch(x.data_ptr<float>(),
+                                 (unsigned int*)words.data_ptr(),
+                                 (float)thresh, (int)bits, n, cur_stream());
+  return words;
+}
+
+at::Tensor lowbit_decode(at::Tensor words, double lo, double hi, int64_t bits,
+                         int64_t n) {
+  auto x = at::empty({n}, words.options().dtype(at::kFloat));
+  lightctr::lowbit_decode_launch((const unsigned int*)words.data_ptr(),
+                                 x.data_ptr<float>(), (float)lo, (float)hi,
+                                 (int)bits, n, cur_stream());
+  return x;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -429,6 +474,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("embed_backward_emit", &embed_backward_emit,
         "per-entry embedding grads (gw, gv) for sorted apply");
   m.def("wide_forward", &wide_forward, "LR wide forward sum");
+  m.def("quantile_encode", &quantile_encode, "int8 quantile encode");
+  m.def("quantile_decode", &quantile_decode, "int8 quantile decode");
+  m.def("lowbit_encode", &lowbit_encode, "1/2-bit sign quantize");
+  m.def("lowbit_decode", &lowbit_decode, "1/2-bit dequantize");
   m.def("nfm_forward", &nfm_forward,
         "NFM bi-interaction fwd (wide, sumVX, vec, vec_bf16)");
   m.def("nfm_backward_emit", &nfm_backward_emit,

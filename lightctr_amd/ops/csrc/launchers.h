@@ -102,4 +102,16 @@ void nfm_backward_emit_launch(const int* row_ptr, const int* fids,
                               const float* dwide, float* gw, float* gv, int B,
                               int K, ihipStream_t* stream);
 
+// --- codec_kernels.hip ---
+void quantile_encode_launch(const float* x, unsigned char* code,
+                            const float* table, int levels, long n,
+                            ihipStream_t* stream);
+void quantile_decode_launch(const unsigned char* code, float* x,
+                            const float* table, int levels, long n,
+                            ihipStream_t* stream);
+void lowbit_encode_launch(const float* x, unsigned int* words, float thresh,
+                          int bits, long n, ihipStream_t* stream);
+void lowbit_decode_launch(const unsigned int* words, float* x, float lo,
+                          float hi, int bits, long n, ihipStream_t* stream);
+
 }  // namespace lightctr

@@ -1,5 +1,7 @@
 from .sharded_fm import ShardedFMModel
 from .ring import RingDataParallel, broadcast_params, allreduce_gradients
+from .ps import PSConfig, PSShard, PSWorker, ps_train_fm, setup_pair_groups
 
 __all__ = ["ShardedFMModel", "RingDataParallel", "broadcast_params",
-           "allreduce_gradients"]
+           "allreduce_gradients", "PSConfig", "PSShard", "PSWorker",
+           "ps_train_fm", "setup_pair_groups"]
