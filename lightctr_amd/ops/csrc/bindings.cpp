@@ -444,6 +444,22 @@ at::Tensor lowbit_decode(at::Tensor words, double lo, double hi, int64_t bits,
   return x;
 }
 
+at::Tensor gbm_hist(at::Tensor bins, at::Tensor grad, at::Tensor hess,
+                    at::Tensor node_of_row, int64_t n_nodes) {
+  CHK(bins.is_cuda() && bins.scalar_type() == at::kByte, "bins u8");
+  check_cuda_f32(grad, "grad");
+  check_cuda_f32(hess, "hess");
+  check_cuda_i32(node_of_row, "node_of_row");
+  const int N = (int)bins.size(0), D = (int)bins.size(1);
+  auto hist = at::zeros({n_nodes, D, 256, 2}, grad.options());
+  lightctr::gbm_hist_launch(bins.data_ptr<unsigned char>(),
+                            grad.data_ptr<float>(), hess.data_ptr<float>(),
+                            node_of_row.data_ptr<int>(),
+                            hist.data_ptr<float>(), N, D, (int)n_nodes,
+                            cur_stream());
+  return hist;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -478,6 +494,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("quantile_decode", &quantile_decode, "int8 quantile decode");
   m.def("lowbit_encode", &lowbit_encode, "1/2-bit sign quantize");
   m.def("lowbit_decode", &lowbit_decode, "1/2-bit dequantize");
+  m.def("gbm_hist", &gbm_hist, "GBM per-node (grad,hess) histograms");
   m.def("nfm_forward", &nfm_forward,
         "NFM bi-interaction fwd (wide, sumVX, vec, vec_bf16)");
   m.def("nfm_backward_emit", &nfm_backward_emit,

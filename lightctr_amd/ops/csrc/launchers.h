@@ -114,4 +114,9 @@ void lowbit_encode_launch(const float* x, unsigned int* words, float thresh,
 void lowbit_decode_launch(const unsigned int* words, float* x, float lo,
                           float hi, int bits, long n, ihipStream_t* stream);
 
+// --- gbm_kernels.hip ---
+void gbm_hist_launch(const unsigned char* bins, const float* grad,
+                     const float* hess, const int* node_of_row, float* hist,
+                     int N, int D, int n_nodes, ihipStream_t* stream);
+
 }  // namespace lightctr

@@ -17,15 +17,10 @@ from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E40
 CSRC = os.path.join(os.path.dirname(os.path.abspath(__file__)),
                     "lightctr_amd", "ops", "csrc")
 
-sources = [
-    os.path.join(CSRC, "bindings.cpp"),
-    os.path.join(CSRC, "fm_kernels.hip"),
-]
-for extra in ["ffm_kernels.hip", "nn_kernels.hip", "gemm_kernels.hip",
-              "embed_kernels.hip", "codec_kernels.hip", "misc_kernels.hip"]:
-    p = os.path.join(CSRC, extra)
-    if os.path.exists(p):
-        sources.append(p)
+import glob
+
+sources = [os.path.join(CSRC, "bindings.cpp")] + sorted(
+    glob.glob(os.path.join(CSRC, "*.hip")))
 
 ext = CUDAExtension(
     name="lightctr_amd.ops._hip_ops",
