@@ -1,0 +1,201 @@
+"""CNN — conv/pool/adapter/FC chain (LeNet-style), MI355X-native.
+
+Capability parity with the reference CNN stack
+(/root/reference/LightCTR/train/layer/convLayer.h conv fwd/bwd via
+Matrix::convolution/deconvolution, poolingLayer.h max-pool with argmax
+unpooling, adapterLayer.h flatten bridge, train/train_cnn_algo.h:37-63 the
+Conv->Pool->Conv->Conv->Adapter->FC->FC LeNet variant). Redesigned for
+CDNA4: convolution = im2col + the hand-written MFMA bf16 GEMM (the conv
+becomes a DenseLayer over unfolded patches, so it inherits the fused
+bias+act epilogue, fp32 master weights and bf16 mirrors); pooling keeps
+argmax indices for the unpool backward. (The reference's LeNet 6x16 sparse
+connection mask is subsumed by full connectivity.)
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+
+import torch
+import torch.nn.functional as F
+
+from .mlp import MLP, DenseLayer
+
+
+class Conv2DLayer:
+    def __init__(self, in_c, out_c, kernel, stride=1, padding=0, act="relu",
+                 optimizer="adam", lr=1e-3, seed=0, device="cpu"):
+        self.in_c, self.out_c = in_c, out_c
+        self.k, self.stride, self.padding = kernel, stride, padding
+        self.fc = DenseLayer(in_c * kernel * kernel, out_c, act=act,
+                             optimizer=optimizer, lr=lr, seed=seed,
+                             device=device)
+        self.device = torch.device(device)
+        self._gpu = self.device.type == "cuda"
+
+    def _prep(self, x):
+        if self._gpu:
+            from ..ops._extension import require_hip_ops
+
+            return require_hip_ops().to_bf16(x.contiguous())
+        return x
+
+    def forward(self, x, train=True):
+        """x: [B, C, H, W] fp32. Returns [B, out_c, OH, OW] fp32."""
+        B, C, H, W = x.shape
+        col = F.unfold(x, self.k, stride=self.stride,
+                       padding=self.padding)  # [B, C*k*k, L]
+        L = col.shape[2]
+        OH = (H + 2 * self.padding - self.k) // self.stride + 1
+        OW = L // OH
+        cols = col.transpose(1, 2).reshape(B * L, -1).contiguous()
+        y, _ = self.fc.forward(self._prep(cols), train=train)
+        self._shape = (B, C, H, W, L, OH, OW)
+        return y.view(B, L, self.out_c).permute(0, 2, 1) \
+            .reshape(B, self.out_c, OH, OW).contiguous()
+
+    def backward(self, dy):
+        B, C, H, W, L, OH, OW = self._shape
+        dyf = dy.reshape(B, self.out_c, L).permute(0, 2, 1) \
+            .reshape(B * L, self.out_c).contiguous()
+        dcol = self.fc.backward(dyf)  # [B*L, C*k*k]
+        dcol = dcol.view(B, L, -1).transpose(1, 2)
+        dx = F.fold(dcol, (H, W), self.k, stride=self.stride,
+                    padding=self.padding)
+        return dx
+
+    def apply_grads(self):
+        self.fc.apply_grads()
+
+    def state_dict(self):
+        return self.fc.state_dict()
+
+    def load_state_dict(self, d):
+        self.fc.load_state_dict(d)
+
+
+class MaxPool2DLayer:
+    """Max pool with argmax mask for the unpooling backward
+    (reference poolingLayer.h)."""
+
+    def __init__(self, size=2):
+        self.size = size
+
+    def forward(self, x, train=True):
+        y, idx = F.max_pool2d(x, self.size, return_indices=True)
+        self._idx = idx
+        self._in_shape = x.shape
+        return y
+
+    def backward(self, dy):
+        return F.max_unpool2d(dy, self._idx, self.size,
+                              output_size=self._in_shape[2:])
+
+    def apply_grads(self):
+        pass
+
+    def state_dict(self):
+        return {}
+
+    def load_state_dict(self, d):
+        pass
+
+
+class FlattenLayer:
+    """Adapter: feature maps <-> flat vector (reference adapterLayer.h)."""
+
+    def forward(self, x, train=True):
+        self._shape = x.shape
+        return x.reshape(x.shape[0], -1).contiguous()
+
+    def backward(self, dy):
+        return dy.reshape(self._shape)
+
+    def apply_grads(self):
+        pass
+
+    def state_dict(self):
+        return {}
+
+    def load_state_dict(self, d):
+        pass
+
+
+@dataclass
+class CNNHyper:
+    in_shape: tuple = (1, 28, 28)
+    n_classes: int = 10
+    lr: float = 1e-3
+    seed: int = 1234
+
+
+class CNNModel:
+    """LeNet variant of train_cnn_algo.h:37-63."""
+
+    def __init__(self, hyper: CNNHyper, device: str = "cpu"):
+        self.h = hyper
+        self.device = torch.device(device)
+        C, H, W = hyper.in_shape
+        lr, s = hyper.lr, hyper.seed
+        self.layers = [
+            Conv2DLayer(C, 6, 5, stride=2, padding=2, lr=lr, seed=s,
+                        device=device),
+            MaxPool2DLayer(2),
+            Conv2DLayer(6, 16, 3, padding=1, lr=lr, seed=s + 1,
+                        device=device),
+            Conv2DLayer(16, 20, 3, padding=1, lr=lr, seed=s + 2,
+                        device=device),
+            FlattenLayer(),
+        ]
+        # probe flatten dim
+        with torch.no_grad():
+            x = torch.zeros(1, C, H, W, device=self.device)
+            for la in self.layers:
+                x = la.forward(x, train=False)
+        flat = x.shape[1]
+        self.head = MLP([flat, 64, hyper.n_classes],
+                        acts=["relu", "none"], optimizer="adam", lr=lr,
+                        seed=s + 3, device=device)
+        self._gpu = self.device.type == "cuda"
+
+    def _prep(self, x):
+        if self._gpu:
+            from ..ops._extension import require_hip_ops
+
+            return require_hip_ops().to_bf16(x.contiguous())
+        return x
+
+    def forward(self, x, train=True):
+        for la in self.layers:
+            x = la.forward(x, train=train)
+        return self.head.forward(self._prep(x), train=train)  # logits
+
+    def train_step(self, x, y) -> float:
+        B = x.shape[0]
+        logits = self.forward(x, train=True)
+        p = torch.softmax(logits, dim=1)
+        yk = torch.nn.functional.one_hot(y.long(),
+                                         self.h.n_classes).float()
+        loss = float(torch.nn.functional.cross_entropy(logits, y.long()))
+        dlogits = (p - yk) / B
+        dx = self.head.backward(dlogits.contiguous())
+        for la in reversed(self.layers):
+            dx = la.backward(dx)
+        self.head.apply_grads()
+        for la in self.layers:
+            la.apply_grads()
+        return loss
+
+    def predict_proba(self, x):
+        return torch.softmax(self.forward(x, train=False), dim=1)
+
+    def save(self, path):
+        torch.save({"layers": [la.state_dict() for la in self.layers],
+                    "head": self.head.state_dict(),
+                    "hyper": self.h.__dict__}, path)
+
+    def load(self, path):
+        d = torch.load(path, map_location=self.device, weights_only=True)
+        for la, sd in zip(self.layers, d["layers"]):
+            la.load_state_dict(sd)
+        self.head.load_state_dict(d["head"])
