@@ -1,0 +1,143 @@
+"""Aux subsystem tests: ANN index, PCA, stats, ensembling, watchdog,
+persistent buffer, shm hashtable, batch predictor."""
+
+import time
+
+import pytest
+import torch
+
+from lightctr_amd.predict.ann_index import ANNIndex
+from lightctr_amd.predict.predictor import BatchPredictor
+from lightctr_amd.utils.ensembling import AdaBoost, VotingEnsemble
+from lightctr_amd.utils.pca import SangerPCA, remove_top_pc
+from lightctr_amd.utils.persistent import PersistentBuffer, ShmHashTable
+from lightctr_amd.utils.stats import (normal_cdf, normal_cdf_inv,
+                                      shuffle_pick_k, z_test)
+from lightctr_amd.utils.watchdog import Watchdog
+
+
+def test_ann_index_recall():
+    g = torch.Generator().manual_seed(0)
+    X = torch.randn(2000, 16, generator=g)
+    idx = ANNIndex(X, n_trees=10, leaf_size=32, seed=1)
+    hits = 0
+    for i in range(20):
+        q = X[i * 7] + torch.randn(16, generator=g) * 0.01
+        approx, _ = idx.query(q, k=10, search_k=400)
+        exact, _ = idx.query_exact(q, k=10)
+        hits += len(set(approx.tolist()) & set(exact.tolist()))
+    assert hits / 200 > 0.6, hits / 200
+
+
+def test_sanger_pca_finds_top_direction():
+    g = torch.Generator().manual_seed(1)
+    basis = torch.tensor([[3.0, 0.0, 0.0], [0.0, 1.0, 0.0]])
+    X = torch.randn(2000, 2, generator=g) @ basis
+    pca = SangerPCA(3, 1, lr=0.05, seed=2)
+    pca.fit(X, iters=100)
+    c = pca.components()[0]
+    assert abs(float(c[0])) > 0.97  # dominant axis = x
+    Xr = remove_top_pc(X, 1)
+    assert float(Xr[:, 0].var()) < float(X[:, 0].var()) * 0.1
+
+
+def test_stats_helpers():
+    assert abs(normal_cdf(0.0) - 0.5) < 1e-9
+    assert abs(float(normal_cdf_inv(0.975)) - 1.95996) < 1e-3
+    z, p = z_test(1.0, 0.0, 1.0, 1.0, 100, 100)
+    assert p < 1e-10
+    picks = shuffle_pick_k(100, 10, seed=3)
+    assert picks.unique().numel() == 10
+
+
+def test_voting_and_adaboost():
+    g = torch.Generator().manual_seed(2)
+    X = torch.randn(500, 2, generator=g)
+    y = (X[:, 0] + 0.5 * X[:, 1] > 0).float()
+
+    def make_stump(X, y, w):
+        # best threshold stump on feature 0 or 1 under weights
+        best = None
+        for f in range(2):
+            for thr in torch.linspace(-1, 1, 21):
+                for sign in (1, -1):
+                    pred = ((X[:, f] * sign) > thr * sign).float()
+                    err = float((w * (pred != y).float()).sum())
+                    if best is None or err < best[0]:
+                        best = (err, f, float(thr), sign)
+        _, f, thr, sign = best
+        return lambda X: ((X[:, f] * sign) > thr * sign).float()
+
+    ada = AdaBoost(make_stump, n_rounds=8).fit(X, y)
+    acc = (ada.predict(X) == y).float().mean()
+    assert acc > 0.9, float(acc)
+
+    vote = VotingEnsemble([lambda X: torch.full((X.shape[0],), 0.8),
+                           lambda X: torch.full((X.shape[0],), 0.4)],
+                          weights=[1.0, 1.0])
+    assert torch.allclose(vote.predict_proba(X),
+                          torch.full((500,), 0.6))
+
+
+def test_watchdog_detects_stale_rank():
+    events = []
+    wd = Watchdog(2, soft_s=0.1, dead_s=0.25, period_s=0.05,
+                  on_soft=lambda r: events.append(("soft", r)),
+                  on_dead=lambda r: events.append(("dead", r))).start()
+    for _ in range(8):
+        wd.heartbeat(0)
+        time.sleep(0.05)
+    wd.stop()
+    states = wd.snapshot()
+    assert states[0] == "alive"
+    assert states[1] == "dead"
+    assert ("dead", 1) in events
+
+
+def test_persistent_buffer_roundtrip(tmp_path):
+    p = str(tmp_path / "buf.bin")
+    b = PersistentBuffer(p, capacity=1024)
+    b.write(b"hello ")
+    b.write(b"world")
+    b.flush()
+    b.close()
+    b2 = PersistentBuffer(p, capacity=1024)
+    assert b2.read_all() == b"hello world"
+    b2.clear()
+    assert b2.size == 0
+    b2.close()
+
+
+def test_shm_hashtable(tmp_path):
+    p = str(tmp_path / "shm.bin")
+    t = ShmHashTable(p, slots_per_table=64, n_tables=4, value_dim=2)
+    import numpy as np
+
+    for k in range(50):
+        assert t.put(k * 977 + 3, np.array([k, k * 0.5], dtype=np.float32))
+    for k in range(50):
+        v = t.get(k * 977 + 3)
+        assert v is not None and abs(v[0] - k) < 1e-6
+    assert t.get(123456789) is None
+    t.close()
+    # reopen persists
+    t2 = ShmHashTable(p, slots_per_table=64, n_tables=4, value_dim=2)
+    v = t2.get(3)
+    assert v is not None and v[0] == 0
+    t2.close()
+
+
+def test_batch_predictor_report():
+    from lightctr_amd.data import LibffmDataset
+    from lightctr_amd.data.synthetic import SyntheticCriteo
+    from lightctr_amd.models.fm import FMHyper, FMModel
+
+    gen = SyntheticCriteo(num_features=1 << 12, seed=9)
+    row_ptr, fields, fids, vals, labels = gen.batch(512)
+    ds = LibffmDataset(row_ptr, fields, fids, vals, labels)
+    model = FMModel(FMHyper(num_features=1 << 12, k=8), device="cpu")
+    bp = BatchPredictor(model, batch_size=128)
+    pred = bp.predict_csr(ds)
+    rep = bp.report(pred, ds.labels)
+    for key in ("auc", "logloss", "accuracy", "precision", "recall", "f1"):
+        assert key in rep
