@@ -76,13 +76,13 @@ class FFMModel:
 
     def _apply_optimizer_hip(self, ops):
         if self.h.optimizer == "ftrl":
-            ops.sparse_ftrl_apply(self.uniq, self.count, self.W, self.V,
+            ops.sparse_ftrl_apply(live, self.count, self.W, self.V,
                                   self.zW, self.nW, self.zV, self.nV,
                                   self.gradW, self.gradV, self.h.ftrl_alpha,
                                   self.h.ftrl_beta, self.h.ftrl_l1,
                                   self.h.ftrl_l2)
         else:
-            ops.sparse_adagrad_apply(self.uniq, self.count, self.W, self.V,
+            ops.sparse_adagrad_apply(live, self.count, self.W, self.V,
                                      self.nW, self.nV, self.gradW, self.gradV,
                                      self.h.lr, self.h.eps, self.h.l2)
 
@@ -97,6 +97,7 @@ class FFMModel:
                              self.gradW, self.gradV, self.touched)
             self.count.zero_()
             ops.bitmap_compact(self.touched, self.uniq, self.count)
+            live = self.uniq[: min(self.uniq.numel(), fids.numel())]
             self._apply_optimizer_hip(ops)
             return loss
         pred = ffm_ref.ffm_forward_ref(row_ptr, fields, fids, vals, self.W,

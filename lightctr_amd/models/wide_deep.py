@@ -125,14 +125,15 @@ class WideDeepModel:
                                 self.gradE, self.touched)
             self.count.zero_()
             ops.bitmap_compact(self.touched, self.uniq, self.count)
+            live = self.uniq[: min(self.uniq.numel(), fids.numel())]
             if self.h.optimizer == "ftrl":
-                ops.fm_ftrl_apply(self.uniq, self.count, self.W, self.E,
+                ops.fm_ftrl_apply(live, self.count, self.W, self.E,
                                   self.zW, self.nW, self.zE, self.nE,
                                   self.gradW, self.gradE, self.h.ftrl_alpha,
                                   self.h.ftrl_beta, self.h.ftrl_l1,
                                   self.h.ftrl_l2)
             else:
-                ops.fm_adagrad_apply(self.uniq, self.count, self.W, self.E,
+                ops.fm_adagrad_apply(live, self.count, self.W, self.E,
                                      self.nW, self.nE, self.gradW, self.gradE,
                                      self.h.lr, self.h.eps, self.h.l2)
             self.mlp.apply_grads()
