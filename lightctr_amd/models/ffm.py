@@ -74,7 +74,7 @@ class FFMModel:
         pred = self.forward(row_ptr, fields, fids, vals)
         return torch.sigmoid(torch.clamp(pred, -16, 16))
 
-    def _apply_optimizer_hip(self, ops):
+    def _apply_optimizer_hip(self, ops, live):
         if self.h.optimizer == "ftrl":
             ops.sparse_ftrl_apply(live, self.count, self.W, self.V,
                                   self.zW, self.nW, self.zV, self.nV,
@@ -98,7 +98,7 @@ class FFMModel:
             self.count.zero_()
             ops.bitmap_compact(self.touched, self.uniq, self.count)
             live = self.uniq[: min(self.uniq.numel(), fids.numel())]
-            self._apply_optimizer_hip(ops)
+            self._apply_optimizer_hip(ops, live)
             return loss
         pred = ffm_ref.ffm_forward_ref(row_ptr, fields, fids, vals, self.W,
                                        self.V)
