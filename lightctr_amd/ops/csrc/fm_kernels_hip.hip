@@ -170,26 +170,47 @@ __global__ void fm_sorted_apply_kernel(
   const int base = wave * chunk;
   if (base >= nnz) return;
   const int end = min(base + chunk, nnz);
+  // each K-lane subgroup owns a CONTIGUOUS sub-chunk (runs stay contiguous
+  // -> one flush per run) and walks it 4 entries at a time with the gathers
+  // issued up-front (4 independent 64B loads in flight per subgroup: the
+  // random gv gather is latency-bound, unrolling feeds the memory system)
+  const int sub = (chunk + G - 1) / G;
+  const int sb = min(base + g * sub, end);
+  const int se = min(sb + sub, end);
 
   int cur_fid = -1;
   float acc = 0.f, accw = 0.f;
-  for (int e = base + g; e < end; e += G) {
-    const int fid = sorted_fids[e];
-    if (fid != cur_fid) {
-      if (cur_fid >= 0) {
-        atomicAdd(&gradV[(size_t)cur_fid * K + k], acc);
-        if (k == 0) atomicAdd(&gradW[cur_fid], accw);
-      }
-      cur_fid = fid;
-      acc = 0.f;
-      accw = 0.f;
-      if (k == 0 && (e == 0 || sorted_fids[e - 1] != fid)) {
-        atomicOr(&touched[fid >> 6], 1ull << (fid & 63));
+  for (int e = sb; e < se; e += 4) {
+    const int nvalid = min(4, se - e);
+    float v[4], vw[4];
+    int f[4];
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      if (u < nvalid) {
+        f[u] = sorted_fids[e + u];
+        const long p = perm[e + u];
+        v[u] = gv[(size_t)p * K + k];
+        vw[u] = (k == 0) ? gw[p] : 0.f;
       }
     }
-    const long p = perm[e];
-    acc += gv[(size_t)p * K + k];
-    if (k == 0) accw += gw[p];
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      if (u >= nvalid) break;
+      if (f[u] != cur_fid) {
+        if (cur_fid >= 0) {
+          atomicAdd(&gradV[(size_t)cur_fid * K + k], acc);
+          if (k == 0) atomicAdd(&gradW[cur_fid], accw);
+        }
+        cur_fid = f[u];
+        acc = 0.f;
+        accw = 0.f;
+        if (k == 0 && (e + u == 0 || sorted_fids[e + u - 1] != f[u])) {
+          atomicOr(&touched[f[u] >> 6], 1ull << (f[u] & 63));
+        }
+      }
+      acc += v[u];
+      accw += vw[u];
+    }
   }
   if (cur_fid >= 0) {
     atomicAdd(&gradV[(size_t)cur_fid * K + k], acc);

@@ -5,19 +5,21 @@
 // GEMM — redesigned for the CDNA4 matrix cores:
 //   * v_mfma_f32_16x16x32_bf16 per-wave tiles, fp32 accumulation
 //   * 128x128 block tile, 4 waves (2x2), 64x64 per wave = 4x4 fragments
-//   * K-tile 64 staged through LDS, +8-element row pad (16 B) so the
-//     ds_read_b128 fragment reads are bank-conflict-free (lanes read 16
-//     different rows at the same k-range; pad makes row stride 144 B,
-//     gcd(36,64)=4 -> 16 distinct banks per 16-lane group)
-//   * B is ALWAYS consumed as Bst[N,K] ("transB" torch-Linear layout), so
-//     both operand fragments are contiguous-k 16 B LDS reads; the model
-//     keeps both W[out,in] and W^T[in,out] bf16 copies (HBM3E is abundant)
-//     so forward AND dgrad hit this fast path; wgrad uses TRANSA=1
+//   * K-tile 64 staged with __builtin_amdgcn_global_load_lds width-16
+//     (direct HBM->LDS DMA; measured +69% over plain-load staging on this
+//     structure per the CDNA4 guide ladder) when the tile is full and the
+//     row stride is 16B-aligned; scalar staging fallback for tails and
+//     transposed-layout operands
+//   * LDS image is LINEAR (glds writes lane-linear) with an XOR swizzle
+//     folded into the SOURCE address and the fragment READ address
+//     (elem_k ^= (row&7)<<3) so the 16-lane ds_read_b128 groups spread
+//     over 8 bank slots instead of hitting 2 (the T2 bank-conflict fix,
+//     both-sides-or-neither rule)
+//   * B is ALWAYS consumed as Bst[N,K] ("transB" torch-Linear layout); the
+//     model keeps both W[out,in] and W^T[in,out] bf16 copies so forward
+//     AND dgrad hit the fast path; wgrad uses TRANSA=1 (+TRANSB=1)
 //   * fused epilogue: optional bias add + activation (relu/sigmoid) +
 //     optional bf16 mirror of C for the next layer's input.
-//
-// Correctness-first structure (the "step-3" ladder shape of the CDNA4
-// guide); deliberate headroom: global_load_lds staging, 8-phase schedule.
 #include "common.h"
 
 namespace lightctr {
@@ -28,7 +30,6 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 #define GEMM_BM 128
 #define GEMM_BN 128
 #define GEMM_BK 64
-#define GEMM_PAD 8  // bf16 elements = 16 B
 
 __device__ __forceinline__ float act_apply(float v, int act) {
   if (act == 1) return fmaxf(v, 0.f);
@@ -36,112 +37,118 @@ __device__ __forceinline__ float act_apply(float v, int act) {
   return v;
 }
 
-// A logical [M,K]: TRANSA=0 -> stored row-major [M,K]; TRANSA=1 -> stored
-// [K,M] (i.e. logical A[m,k] = Aptr[k*M + m]; the wgrad path where A = dY^T).
-// B logical [K,N]: TRANSB=0 -> stored [N,K] (torch-Linear weight layout, the
-// fast contiguous-k path used by forward and dgrad); TRANSB=1 -> stored
-// [K,N] row-major (wgrad's activation operand).
+// element-index swizzle: flip k-elem bits 3..5 by row bits 0..2 (16B units)
+__device__ __forceinline__ int swz(int row, int ke) {
+  return ke ^ ((row & 7) << 3);
+}
+
+// Stage a [ROWS x 64] bf16 tile (global row-major, row stride `stride`
+// elems) into LDS `dst` (linear [ROWS][64] + swizzle) with glds. Full tiles
+// only; whole 256-thread block participates; ROWS = 128.
+__device__ __forceinline__ void stage_glds(const __bf16* __restrict__ gbase,
+                                           long stride, __bf16* dst) {
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+#pragma unroll
+  for (int it = 0; it < 4; ++it) {
+    const int seg = wave * 4 + it;  // 8 rows per segment
+    const int row = seg * 8 + (lane >> 3);
+    const int ke = swz(row, (lane & 7) * 8);
+    const __bf16* src = gbase + (long)row * stride + ke;
+    // LDS dest is wave-uniform base; HW writes lane*16 bytes per lane,
+    // which is exactly this mapping's (row, ke) order (lane-linear image)
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)src,
+        (__attribute__((address_space(3))) void*)(dst + seg * 512), 16, 0,
+        0);
+  }
+}
+
+// Scalar staging for tails / transposed layouts; writes the same swizzled
+// image. rows/cols guarded against (max_row, K).
+template <int TRANS>
+__device__ __forceinline__ void stage_scalar(const __bf16* __restrict__ g,
+                                             int base_row, int k0, int K,
+                                             int max_row, long stride,
+                                             __bf16* dst) {
+  const int tid = threadIdx.x;
+  if (TRANS == 0) {
+    // row-major [row, k]: thread t covers row=t/2, k-half=(t%2)*32
+    const int r = tid >> 1;
+    const int kk = (tid & 1) * 32;
+    const int grow = base_row + r;
+    for (int u = 0; u < 32; ++u) {
+      const int ke = kk + u;
+      __bf16 v = (__bf16)0.f;
+      if (grow < max_row && k0 + ke < K)
+        v = g[(long)grow * stride + k0 + ke];
+      dst[r * 64 + swz(r, ke)] = v;
+    }
+  } else {
+    // stored [K, rows]: thread t covers k=t/8 (+32), rowpart=(t%8)*16
+    const int kk = tid >> 3;
+    const int rp = (tid & 7) * 16;
+    for (int kr = 0; kr < 2; ++kr) {
+      const int ke = kk + kr * 32;
+      const int gk = k0 + ke;
+      for (int e = 0; e < 16; ++e) {
+        const int r = rp + e;
+        const int grow = base_row + r;
+        __bf16 v = (__bf16)0.f;
+        if (gk < K && grow < max_row) v = g[(long)gk * stride + grow];
+        dst[r * 64 + swz(r, ke)] = v;
+      }
+    }
+  }
+}
+
 template <int TRANSA, int TRANSB>
 __global__ __launch_bounds__(256) void gemm_bf16_kernel(
     const __bf16* __restrict__ A, const __bf16* __restrict__ Bst,
     const float* __restrict__ bias, float* __restrict__ C,
     __bf16* __restrict__ Cbf, int M, int N, int K, int act) {
-  __shared__ __bf16 As[GEMM_BM][GEMM_BK + GEMM_PAD];
-  __shared__ __bf16 Bs[GEMM_BN][GEMM_BK + GEMM_PAD];
+  // single __shared__ object (A tile then B tile) — guide §5 trap 4(a)
+  __shared__ __bf16 smem[2 * GEMM_BM * GEMM_BK];
+  __bf16* As = smem;
+  __bf16* Bs = smem + GEMM_BM * GEMM_BK;
 
-  const int tid = threadIdx.x;
-  const int lane = tid & 63;
-  const int wave = tid >> 6;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
   const int wm = wave >> 1, wn = wave & 1;
   const int M0 = blockIdx.y * GEMM_BM;
   const int N0 = blockIdx.x * GEMM_BN;
 
+  // glds eligibility (full tile + 16B-aligned row stride + 16B base)
+  const bool a_glds = (TRANSA == 0) && (M0 + GEMM_BM <= M) && (K % 8 == 0);
+  const bool b_glds = (TRANSB == 0) && (N0 + GEMM_BN <= N) && (K % 8 == 0);
+
   f32x4 acc[4][4] = {};
 
   for (int k0 = 0; k0 < K; k0 += GEMM_BK) {
-    // ---- stage A tile ----
-    if (TRANSA == 0) {
-      // thread t: row = t/2, k-half = (t%2)*32, 32 bf16 = 64 B contiguous
-      const int m = tid >> 1;
-      const int kk = (tid & 1) * 32;
-      const int gm = M0 + m;
-#pragma unroll
-      for (int u = 0; u < 32; u += 8) {
-        bf16x8 v = {};
-        if (gm < M && k0 + kk + u + 7 < K) {
-          v = *(const bf16x8*)&A[(size_t)gm * K + k0 + kk + u];
-        } else if (gm < M) {
-          for (int e = 0; e < 8; ++e)
-            if (k0 + kk + u + e < K)
-              ((__bf16*)&v)[e] = A[(size_t)gm * K + k0 + kk + u + e];
-        }
-        *(bf16x8*)&As[m][kk + u] = v;
-      }
+    const bool k_full = (k0 + GEMM_BK <= K);
+    if (a_glds && k_full) {
+      stage_glds(A + (long)M0 * K + k0, K, As);
     } else {
-      // stored [K,M]: thread t: k = t/8, m-part = (t%8)*16; contiguous in m,
-      // scatter-transposed into As
-      const int kk = tid >> 3;
-      const int mp = (tid & 7) * 16;
-      const int gk = k0 + kk;
-      // two k rows per thread (BK=64, 32 k-rows covered per 256 threads pass)
-#pragma unroll
-      for (int kr = 0; kr < 2; ++kr) {
-        const int kcur = kk + kr * 32;
-        const int gkc = k0 + kcur;
-        for (int e = 0; e < 16; ++e) {
-          const int gm = M0 + mp + e;
-          __bf16 v = (__bf16)0.f;
-          if (gkc < K && gm < M) v = A[(size_t)gkc * M + gm];
-          As[mp + e][kcur] = v;
-        }
-      }
-      (void)gk;
+      stage_scalar<TRANSA>(A, M0, k0, K, M, TRANSA ? M : K, As);
     }
-    // ---- stage B tile ----
-    if (TRANSB == 0) {
-      // stored [N,K]: same contiguous-k pattern as the TRANSA=0 A load
-      const int n = tid >> 1;
-      const int kk = (tid & 1) * 32;
-      const int gn = N0 + n;
-#pragma unroll
-      for (int u = 0; u < 32; u += 8) {
-        bf16x8 v = {};
-        if (gn < N && k0 + kk + u + 7 < K) {
-          v = *(const bf16x8*)&Bst[(size_t)gn * K + k0 + kk + u];
-        } else if (gn < N) {
-          for (int e = 0; e < 8; ++e)
-            if (k0 + kk + u + e < K)
-              ((__bf16*)&v)[e] = Bst[(size_t)gn * K + k0 + kk + u + e];
-        }
-        *(bf16x8*)&Bs[n][kk + u] = v;
-      }
+    if (b_glds && k_full) {
+      stage_glds(Bst + (long)N0 * K + k0, K, Bs);
     } else {
-      // stored [K,N] row-major: coalesced along n, scatter-transpose to Bs
-      const int kk = tid >> 3;
-      const int np = (tid & 7) * 16;
-#pragma unroll
-      for (int kr = 0; kr < 2; ++kr) {
-        const int kcur = kk + kr * 32;
-        const int gkc = k0 + kcur;
-        for (int e = 0; e < 16; ++e) {
-          const int gn = N0 + np + e;
-          __bf16 v = (__bf16)0.f;
-          if (gkc < K && gn < N) v = Bst[(size_t)gkc * N + gn];
-          Bs[np + e][kcur] = v;
-        }
-      }
+      stage_scalar<TRANSB>(Bst, N0, k0, K, N, TRANSB ? N : K, Bs);
     }
     __syncthreads();
 
-    // ---- MFMA over the K tile (2 chunks of 32) ----
 #pragma unroll
     for (int kc = 0; kc < GEMM_BK; kc += 32) {
       const int ks = kc + (lane >> 4) * 8;  // this lane's 8-deep k slice
       bf16x8 a[4], b[4];
 #pragma unroll
       for (int f = 0; f < 4; ++f) {
-        a[f] = *(const bf16x8*)&As[wm * 64 + f * 16 + (lane & 15)][ks];
-        b[f] = *(const bf16x8*)&Bs[wn * 64 + f * 16 + (lane & 15)][ks];
+        const int ra = wm * 64 + f * 16 + (lane & 15);
+        const int rb = wn * 64 + f * 16 + (lane & 15);
+        a[f] = *(const bf16x8*)&As[ra * 64 + swz(ra, ks)];
+        b[f] = *(const bf16x8*)&Bs[rb * 64 + swz(rb, ks)];
       }
 #pragma unroll
       for (int i = 0; i < 4; ++i)
