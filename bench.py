@@ -4,6 +4,10 @@ sparse, 1xMI355X, FTRL fused update"), scaled to N GPUs (weak scaling: fixed
 per-GPU batch; N>1 shards the feature table by feature-hash across ranks and
 exchanges embeddings/grads with RCCL all-to-all over xGMI).
 
+Other BASELINE configs are selectable with --model ffm|nfm|widedeep.
+Single-GPU steps are hipGraph-captured (one graph per pooled batch) unless
+--no-graph; capture failures fall back to eager launches.
+
 Contract (driver): `python bench.py --gpus N --steps K --warmup W`; for N>1
 launched via torch.distributed.run with one rank per GPU. Rank 0 prints one
 JSON line with the whole-job examples/sec.
@@ -19,6 +23,43 @@ import time
 import torch
 
 
+def build_model(args, device, world):
+    from lightctr_amd.models.fm import FMHyper, FMModel
+
+    if args.model == "fm":
+        hyper = FMHyper(num_features=args.features, k=args.k,
+                        optimizer=args.optimizer, seed=1234)
+        if world > 1:
+            from lightctr_amd.parallel.sharded_fm import ShardedFMModel
+
+            return ShardedFMModel(hyper, device=device), "sparse"
+        return FMModel(hyper, device=device), "sparse"
+    if args.model == "ffm":
+        from lightctr_amd.models.ffm import FFMHyper, FFMModel
+
+        assert world == 1, "ffm bench is single-GPU in this revision"
+        return FFMModel(FFMHyper(num_features=args.features, num_fields=39,
+                                 k=min(args.k, 8), optimizer=args.optimizer,
+                                 seed=1234), device=device), "fieldaware"
+    if args.model == "nfm":
+        from lightctr_amd.models.nfm import NFMHyper, NFMModel
+
+        assert world == 1
+        return NFMModel(NFMHyper(num_features=args.features, k=args.k,
+                                 hidden=(64,), seed=1234),
+                        device=device), "sparse"
+    if args.model == "widedeep":
+        from lightctr_amd.models.wide_deep import (WideDeepHyper,
+                                                   WideDeepModel)
+
+        assert world == 1
+        return WideDeepModel(
+            WideDeepHyper(num_features=args.features, num_fields=39,
+                          k=args.k, hidden=(256, 128), seed=1234),
+            device=device), "sparse"
+    raise SystemExit(f"unknown model {args.model}")
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -29,10 +70,13 @@ def main():
     ap.add_argument("--k", type=int, default=16)
     ap.add_argument("--features", type=int, default=1 << 24)
     ap.add_argument("--optimizer", default="ftrl")
+    ap.add_argument("--model", default="fm",
+                    choices=["fm", "ffm", "nfm", "widedeep"])
+    ap.add_argument("--no-graph", action="store_true",
+                    help="disable hipGraph capture of the step")
     args = ap.parse_args()
 
     from lightctr_amd.data.synthetic import SyntheticCriteo
-    from lightctr_amd.models.fm import FMHyper, FMModel
 
     rank = int(os.environ.get("RANK", 0))
     world = int(os.environ.get("WORLD_SIZE", 1))
@@ -49,14 +93,7 @@ def main():
         dist = dist_mod
         dist.init_process_group(backend="nccl" if have_gpu else "gloo")
 
-    hyper = FMHyper(num_features=args.features, k=args.k,
-                    optimizer=args.optimizer, seed=1234)
-    if world > 1:
-        from lightctr_amd.parallel.sharded_fm import ShardedFMModel
-
-        model = ShardedFMModel(hyper, device=device)
-    else:
-        model = FMModel(hyper, device=device)
+    model, flavor = build_model(args, device, world)
 
     gen = SyntheticCriteo(num_features=args.features, seed=1234 + rank,
                           device=device)
@@ -64,9 +101,37 @@ def main():
     # training (synthetic data; generation is not the benchmark subject)
     pool = [gen.batch(args.batch) for _ in range(4)]
 
-    def step(i):
+    def eager_step(i):
         row_ptr, fields, fids, vals, labels = pool[i % len(pool)]
+        if flavor == "fieldaware":
+            return model.train_step(row_ptr, fields, fids, vals, labels)
         return model.train_step(row_ptr, fids, vals, labels)
+
+    step = eager_step
+    use_graph = (have_gpu and world == 1 and not args.no_graph)
+    if use_graph:
+        try:
+            # warm up allocator + kernels, then capture one graph per
+            # pooled batch (shapes static per batch slot)
+            for i in range(len(pool)):
+                eager_step(i)
+            torch.cuda.synchronize()
+            graphs = []
+            for i in range(len(pool)):
+                gr = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(gr):
+                    eager_step(i)
+                graphs.append(gr)
+            torch.cuda.synchronize()
+
+            def graph_step(i):
+                graphs[i % len(graphs)].replay()
+
+            step = graph_step
+        except Exception as e:  # capture unsupported -> eager fallback
+            print(f"# hipGraph capture failed ({type(e).__name__}: {e}); "
+                  "falling back to eager", flush=True)
+            step = eager_step
 
     for i in range(args.warmup):
         step(i)
@@ -92,8 +157,14 @@ def main():
     total_examples = args.batch * args.steps * world
     value = total_examples / elapsed
     if rank == 0:
+        mdesc = {"fm": f"FM k={args.k}, {args.optimizer} fused update",
+                 "ffm": f"FFM k={min(args.k, 8)}, 39 fields",
+                 "nfm": f"NFM k={args.k} + MLP(64)",
+                 "widedeep": f"Wide&Deep k={args.k} + MLP(256,128)"}
         out = {
-            "metric": "examples/sec (whole node), FM training on synthetic Criteo-shaped sparse",
+            "metric": "examples/sec (whole node), "
+                      f"{args.model.upper()} training on synthetic "
+                      "Criteo-shaped sparse",
             "value": value,
             "unit": "examples/sec",
             "n_gpus": n_gpus,
@@ -103,15 +174,16 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "fp32",
+            "dtype": "bf16" if args.model in ("nfm", "widedeep") else "fp32",
             "data": "synthetic",
             "config": {
-                "model": f"FM k={args.k}, {args.optimizer} fused update",
+                "model": mdesc[args.model],
                 "global_batch": args.batch * world,
                 "num_features": args.features,
                 "num_fields": 39,
                 "parallelism": f"hash-sharded table, all-to-all, dp{world}"
-                if world > 1 else "single-gpu",
+                if world > 1 else "single-gpu"
+                + (", hipGraph" if step is not eager_step else ""),
             },
         }
         print(json.dumps(out))
