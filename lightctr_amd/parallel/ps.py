@@ -186,6 +186,9 @@ class PSShard:
         else:
             gW = self._recv((n,), torch.float32, w, group)
             gV = self._recv((n, K), torch.float32, w, group)
+        # wire validity check (reference paramserver.h:172,244)
+        if not bool(torch.isfinite(gW).all() and torch.isfinite(gV).all()):
+            return  # drop corrupt push rather than poison the table
         # stale-push drop (reference paramserver.h:199-208)
         if epoch + cfg.staleness < int(self.worker_epoch.max()):
             return
@@ -280,6 +283,17 @@ class PSWorker:
     def push(self, fids_uniq: torch.Tensor, gW: torch.Tensor,
              gV: torch.Tensor, epoch: int):
         cfg = self.cfg
+        # "preferred value" gradient filter (reference push.h:60-71 +
+        # distributed_algo_abst.h:76-79): drop features whose grads are all
+        # ~0 (nothing to send) or contain non-finite / >15-magnitude values
+        mag = torch.maximum(gW.abs(), gV.abs().amax(dim=1))
+        ok = torch.isfinite(gW) & torch.isfinite(gV).all(dim=1) \
+            & (mag > 1e-12) & (mag < 15.0)
+        if not bool(ok.all()):
+            fids_uniq, gW, gV = fids_uniq[ok], gW[ok], gV[ok]
+        from ..utils.trace import debug_log
+
+        debug_log(f"push: {fids_uniq.numel()} keys epoch={epoch}")
         owner = _owner(fids_uniq, cfg.ps_shards)
         for ps in range(cfg.ps_shards):
             sel = owner == ps

@@ -80,11 +80,12 @@ class ShardedFMModel:
         owner = (uniq.long() % world)
         order = torch.argsort(owner, stable=True)
         uniq_o = uniq[order]
-        send_counts = torch.bincount(owner, minlength=world).cpu().tolist()
-        cnt_t = torch.tensor(send_counts, dtype=torch.int64)
-        recv_cnt_t = torch.empty(world, dtype=torch.int64)
+        cnt_t = torch.bincount(owner, minlength=world).to(self.device)
+        send_counts = cnt_t.cpu().tolist()
+        recv_cnt_t = torch.empty(world, dtype=cnt_t.dtype,
+                                 device=self.device)
         dist.all_to_all_single(recv_cnt_t, cnt_t, group=self.group)
-        recv_counts = recv_cnt_t.tolist()
+        recv_counts = recv_cnt_t.cpu().tolist()
 
         # 2. exchange requested fids; owners gather their shard rows
         req = self._exchange(uniq_o, send_counts, recv_counts)
@@ -168,11 +169,12 @@ class ShardedFMModel:
         owner = (uniq.long() % world)
         order = torch.argsort(owner, stable=True)
         uniq_o = uniq[order]
-        send_counts = torch.bincount(owner, minlength=world).cpu().tolist()
-        cnt_t = torch.tensor(send_counts, dtype=torch.int64)
-        recv_cnt_t = torch.empty(world, dtype=torch.int64)
+        cnt_t = torch.bincount(owner, minlength=world).to(self.device)
+        send_counts = cnt_t.cpu().tolist()
+        recv_cnt_t = torch.empty(world, dtype=cnt_t.dtype,
+                                 device=self.device)
         dist.all_to_all_single(recv_cnt_t, cnt_t, group=self.group)
-        recv_counts = recv_cnt_t.tolist()
+        recv_counts = recv_cnt_t.cpu().tolist()
         req = self._exchange(uniq_o, send_counts, recv_counts)
         lidx = req.long() // world
         Wl_o = self._exchange(self.W[lidx], recv_counts, send_counts)
