@@ -130,3 +130,15 @@ def test_precision_recall():
     assert p == pytest.approx(2 / 3)
     assert r == pytest.approx(1.0)
     assert f1 == pytest.approx(0.8)
+
+
+def test_dense_csv_loader(tmp_path):
+    from lightctr_amd.data import load_dense_csv
+
+    p = tmp_path / "dense.csv"
+    p.write_text("3,0,128,255,0\n7,255,0,0,64,\n")
+    X, y = load_dense_csv(str(p))
+    assert X.shape == (2, 4)
+    assert y.tolist() == [3, 7]
+    assert abs(float(X[0, 2]) - 1.0) < 1e-6
+    assert abs(float(X[1, 3]) - 64 / 255) < 1e-6
