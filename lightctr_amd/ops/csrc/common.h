@@ -2,6 +2,7 @@
 // Hand-written HIP for MI355X: 64-wide wavefronts, LDS, MFMA.
 #pragma once
 #include <hip/hip_runtime.h>
+#include <cstdio>
 
 #define LCTR_WAVE 64
 

@@ -103,11 +103,17 @@ __device__ __forceinline__ void stage_scalar(const __bf16* __restrict__ g,
   }
 }
 
+// SPLITK: gridDim.z > 1 partitions the K loop across z-blocks, each
+// accumulating its partial tile into C with fp32 atomics (C pre-zeroed by
+// the launcher). Used for the wgrad shapes (tiny M,N and K = batch) where
+// a single tile otherwise serializes hundreds of K-steps on a handful of
+// CUs. bias/act/bf16-emit are not supported with split-K (wgrad needs
+// neither).
 template <int TRANSA, int TRANSB>
 __global__ __launch_bounds__(256) void gemm_bf16_kernel(
     const __bf16* __restrict__ A, const __bf16* __restrict__ Bst,
     const float* __restrict__ bias, float* __restrict__ C,
-    __bf16* __restrict__ Cbf, int M, int N, int K, int act) {
+    __bf16* __restrict__ Cbf, int M, int N, int K, int act, int kchunk) {
   // single __shared__ object (A tile then B tile) — guide §5 trap 4(a)
   __shared__ __bf16 smem[2 * GEMM_BM * GEMM_BK];
   __bf16* As = smem;
@@ -125,8 +131,10 @@ __global__ __launch_bounds__(256) void gemm_bf16_kernel(
 
   f32x4 acc[4][4] = {};
 
-  for (int k0 = 0; k0 < K; k0 += GEMM_BK) {
-    const bool k_full = (k0 + GEMM_BK <= K);
+  const int kbeg = (kchunk > 0) ? blockIdx.z * kchunk : 0;
+  const int kend = (kchunk > 0) ? min(K, kbeg + kchunk) : K;
+  for (int k0 = kbeg; k0 < kend; k0 += GEMM_BK) {
+    const bool k_full = (k0 + GEMM_BK <= kend) && (k0 + GEMM_BK <= K);
     if (a_glds && k_full) {
       stage_glds(A + (long)M0 * K + k0, K, As);
     } else {
@@ -173,9 +181,13 @@ __global__ __launch_bounds__(256) void gemm_bf16_kernel(
       for (int r = 0; r < 4; ++r) {
         const int row = M0 + wm * 64 + i * 16 + (lane >> 4) * 4 + r;
         if (row >= M) continue;
-        const float v = act_apply(acc[i][j][r] + bv, act);
-        C[(size_t)row * N + col] = v;
-        if (Cbf) Cbf[(size_t)row * N + col] = (__bf16)v;
+        if (kchunk > 0) {
+          atomicAdd(&C[(size_t)row * N + col], acc[i][j][r]);
+        } else {
+          const float v = act_apply(acc[i][j][r] + bv, act);
+          C[(size_t)row * N + col] = v;
+          if (Cbf) Cbf[(size_t)row * N + col] = (__bf16)v;
+        }
       }
     }
   }
@@ -186,10 +198,22 @@ void gemm_bf16_launch(const void* A, const void* Bst, const float* bias,
                       int transB, int act, hipStream_t stream) {
   dim3 block(256);
   dim3 grid((N + GEMM_BN - 1) / GEMM_BN, (M + GEMM_BM - 1) / GEMM_BM);
+  // split-K when the output grid is far too small to fill 256 CUs and K is
+  // deep (the wgrad regime): target >= 512 z*xy blocks
+  int kchunk = 0;
+  const int xy = (int)(grid.x * grid.y);
+  if (xy < 64 && K >= 4096 && Cbf == nullptr && bias == nullptr &&
+      act == 0) {
+    int zw = min(64, max(2, 512 / xy));
+    kchunk = ((K + zw - 1) / zw + GEMM_BK - 1) / GEMM_BK * GEMM_BK;
+    grid.z = (K + kchunk - 1) / kchunk;
+    LCTR_CHECK_HIP(hipMemsetAsync(C, 0, (size_t)M * N * sizeof(float),
+                                  stream));
+  }
 #define LAUNCH_GEMM(TA, TB)                                                 \
   hipLaunchKernelGGL((gemm_bf16_kernel<TA, TB>), grid, block, 0, stream,    \
                      (const __bf16*)A, (const __bf16*)Bst, bias, C,         \
-                     (__bf16*)Cbf, M, N, K, act)
+                     (__bf16*)Cbf, M, N, K, act, kchunk)
   if (transA == 0 && transB == 0) LAUNCH_GEMM(0, 0);
   else if (transA == 1 && transB == 0) LAUNCH_GEMM(1, 0);
   else if (transA == 0 && transB == 1) LAUNCH_GEMM(0, 1);
