@@ -60,6 +60,7 @@ class FFMModel:
         self.uniq = torch.zeros(cap, dtype=torch.int32, device=self.device)
         self.count = torch.zeros(1, dtype=torch.int32, device=self.device)
         self._use_hip = self.device.type == "cuda"
+        self.backward_mode = "sorted"  # sorted (default) | atomic
         if self._use_hip:
             require_hip_ops()
 
@@ -93,8 +94,16 @@ class FFMModel:
             ops = require_hip_ops()
             pred = ops.ffm_forward(row_ptr, fields, fids, vals, self.W, self.V)
             loss, dpred = ops.logloss_grad(pred, labels, scale)
-            ops.ffm_backward(row_ptr, fields, fids, vals, self.V, dpred,
-                             self.gradW, self.gradV, self.touched)
+            if self.backward_mode == "sorted":
+                sorted_fids, perm = torch.sort(fids)
+                row_of_entry = ops.row_index(row_ptr, fids.numel())
+                ops.ffm_sorted_backward(sorted_fids, perm, row_of_entry,
+                                        row_ptr, fields, fids, vals, self.V,
+                                        dpred, self.gradW, self.gradV,
+                                        self.touched)
+            else:
+                ops.ffm_backward(row_ptr, fields, fids, vals, self.V, dpred,
+                                 self.gradW, self.gradV, self.touched)
             self.count.zero_()
             ops.bitmap_compact(self.touched, self.uniq, self.count)
             live = self.uniq[: min(self.uniq.numel(), fids.numel())]

@@ -146,6 +146,33 @@ void ffm_backward(at::Tensor row_ptr, at::Tensor fields, at::Tensor fids,
       (unsigned long long*)touched.data_ptr(), nfields, B, K, cur_stream());
 }
 
+void ffm_sorted_backward(at::Tensor sorted_fids, at::Tensor perm,
+                         at::Tensor row_of_entry, at::Tensor row_ptr,
+                         at::Tensor fields, at::Tensor fids, at::Tensor vals,
+                         at::Tensor V, at::Tensor dpred, at::Tensor gradW,
+                         at::Tensor gradV, at::Tensor touched) {
+  check_cuda_i32(sorted_fids, "sorted_fids");
+  CHK(perm.scalar_type() == at::kLong, "perm must be int64");
+  check_cuda_i32(row_of_entry, "row_of_entry");
+  const int nfields = (int)V.size(1);
+  const int K = (int)V.size(2);
+  lightctr::ffm_sorted_backward_launch(
+      sorted_fids.data_ptr<int>(), perm.data_ptr<long>(),
+      row_of_entry.data_ptr<int>(), row_ptr.data_ptr<int>(),
+      fields.data_ptr<int>(), fids.data_ptr<int>(), vals.data_ptr<float>(),
+      V.data_ptr<float>(), dpred.data_ptr<float>(), gradW.data_ptr<float>(),
+      gradV.data_ptr<float>(), (unsigned long long*)touched.data_ptr(),
+      nfields, (int)sorted_fids.numel(), K, cur_stream());
+}
+
+at::Tensor row_index(at::Tensor row_ptr, int64_t nnz) {
+  check_cuda_i32(row_ptr, "row_ptr");
+  auto out = at::empty({nnz}, row_ptr.options());
+  lightctr::row_index_launch(row_ptr.data_ptr<int>(), out.data_ptr<int>(),
+                             (int)row_ptr.numel() - 1, cur_stream());
+  return out;
+}
+
 // ---- generic sparse optimizers (D = latent block width, runtime) ----
 
 void sparse_adagrad_apply(at::Tensor uniq, at::Tensor count, at::Tensor W,
@@ -473,6 +500,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "FM backward phase 2: segment-reduce sorted grads into slabs");
   m.def("ffm_forward", &ffm_forward, "FFM fused pairwise forward");
   m.def("ffm_backward", &ffm_backward, "FFM fused pairwise backward scatter");
+  m.def("ffm_sorted_backward", &ffm_sorted_backward,
+        "FFM sorted segment-reduce backward (LDS block accumulate)");
+  m.def("row_index", &row_index, "entry -> row index from row_ptr");
   m.def("sparse_adagrad_apply", &sparse_adagrad_apply,
         "generic sparse fused Adagrad (runtime D)");
   m.def("sparse_ftrl_apply", &sparse_ftrl_apply,

@@ -110,6 +110,120 @@ __global__ void ffm_backward_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Sorted FFM backward: entries sorted by fid; each 64-lane wave walks a
+// contiguous chunk, accumulating the CURRENT feature's full [nfields, K]
+// gradient block in LDS (recomputed from the entry's row: for each other
+// feature j, block[Fj,:] += d*x_i*x_j*V[fid_j, F_i, :]) and flushing with
+// one atomicAdd sweep per run. Removes the hot-feature global-atomic
+// serialization of the naive scatter (the FM lesson, applied field-aware).
+// LDS atomics absorb duplicate fields within a row.
+// ---------------------------------------------------------------------------
+template <int K>
+__global__ void ffm_sorted_backward_kernel(
+    const int* __restrict__ sorted_fids, const long* __restrict__ perm,
+    const int* __restrict__ row_of_entry, const int* __restrict__ row_ptr,
+    const int* __restrict__ fields, const int* __restrict__ fids,
+    const float* __restrict__ vals, const float* __restrict__ V,
+    const float* __restrict__ dpred, float* __restrict__ gradW,
+    float* __restrict__ gradV, unsigned long long* __restrict__ touched,
+    int nfields, int nnz, int chunk) {
+  extern __shared__ float lds_acc[];  // [waves_per_block][nfields*K]
+  constexpr int G = LCTR_WAVE / K;
+  const int lane = threadIdx.x & 63;
+  const int wave_in_blk = threadIdx.x >> 6;
+  const int wave = blockIdx.x * (blockDim.x >> 6) + wave_in_blk;
+  const int jg = lane / K;
+  const int k = lane % K;
+  const int base = wave * chunk;
+  if (base >= nnz) return;
+  const int end = min(base + chunk, nnz);
+  float* acc = &lds_acc[wave_in_blk * nfields * K];
+  const int blk = nfields * K;
+  for (int i = lane; i < blk; i += LCTR_WAVE) acc[i] = 0.f;
+  // no __syncthreads needed: each wave owns its LDS slice; lanes of one
+  // wave execute in lockstep
+
+  int cur = -1;
+  float accw = 0.f;
+  for (int e = base; e < end; ++e) {
+    const int fid = sorted_fids[e];
+    if (fid != cur) {
+      if (cur >= 0) {
+        for (int i = lane; i < blk; i += LCTR_WAVE) {
+          if (acc[i] != 0.f) {
+            atomicAdd(&gradV[(size_t)cur * blk + i], acc[i]);
+            acc[i] = 0.f;
+          }
+        }
+        if (lane == 0) atomicAdd(&gradW[cur], accw);
+      }
+      cur = fid;
+      accw = 0.f;
+      if (lane == 0 && (e == 0 || sorted_fids[e - 1] != fid)) {
+        atomicOr(&touched[fid >> 6], 1ull << (fid & 63));
+      }
+    }
+    const int p = (int)perm[e];
+    const int r = row_of_entry[p];
+    const int Fi = fields[p];
+    const float xi = vals[p];
+    const float d = dpred[r];
+    const int beg = row_ptr[r], rend = row_ptr[r + 1];
+    for (int j = beg + jg; j < rend; j += G) {
+      if (j == p) continue;
+      const float v = V[((size_t)fids[j] * nfields + Fi) * K + k];
+      atomicAdd(&acc[fields[j] * K + k], d * xi * vals[j] * v);
+    }
+    if (lane == 0) accw += d * xi;
+  }
+  if (cur >= 0) {
+    for (int i = lane; i < blk; i += LCTR_WAVE) {
+      if (acc[i] != 0.f) atomicAdd(&gradV[(size_t)cur * blk + i], acc[i]);
+    }
+    if (lane == 0) atomicAdd(&gradW[cur], accw);
+  }
+}
+
+// entry -> row index materialization (one wave per row)
+__global__ void row_index_kernel(const int* __restrict__ row_ptr,
+                                 int* __restrict__ row_idx, int B) {
+  const int lane = threadIdx.x & 63;
+  const int row = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  if (row >= B) return;
+  for (int j = row_ptr[row] + lane; j < row_ptr[row + 1]; j += 64)
+    row_idx[j] = row;
+}
+
+void ffm_sorted_backward_launch(const int* sorted_fids, const long* perm,
+                                const int* row_of_entry, const int* row_ptr,
+                                const int* fields, const int* fids,
+                                const float* vals, const float* V,
+                                const float* dpred, float* gradW,
+                                float* gradV, unsigned long long* touched,
+                                int nfields, int nnz, int K,
+                                hipStream_t stream) {
+  const int chunk = 128;
+  const int wpb = 4;
+  const int nwaves = (nnz + chunk - 1) / chunk;
+  dim3 block(wpb * LCTR_WAVE);
+  dim3 grid((nwaves + wpb - 1) / wpb);
+  const size_t lds = (size_t)wpb * nfields * K * sizeof(float);
+  DISPATCH_FFM_K(K, hipLaunchKernelGGL((ffm_sorted_backward_kernel<KC>),
+                                       grid, block, lds, stream, sorted_fids,
+                                       perm, row_of_entry, row_ptr, fields,
+                                       fids, vals, V, dpred, gradW, gradV,
+                                       touched, nfields, nnz, chunk));
+}
+
+void row_index_launch(const int* row_ptr, int* row_idx, int B,
+                      hipStream_t stream) {
+  dim3 block(256);
+  dim3 grid((B + 3) / 4);
+  hipLaunchKernelGGL(row_index_kernel, grid, block, 0, stream, row_ptr,
+                     row_idx, B);
+}
+
 void ffm_forward_launch(const int* row_ptr, const int* fields, const int* fids,
                         const float* vals, const float* W, const float* V,
                         float* pred, int nfields, int B, int K,

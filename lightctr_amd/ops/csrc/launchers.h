@@ -41,6 +41,17 @@ void ffm_backward_launch(const int* row_ptr, const int* fields,
                          unsigned long long* touched, int nfields, int B,
                          int K, ihipStream_t* stream);
 
+void ffm_sorted_backward_launch(const int* sorted_fids, const long* perm,
+                                const int* row_of_entry, const int* row_ptr,
+                                const int* fields, const int* fids,
+                                const float* vals, const float* V,
+                                const float* dpred, float* gradW,
+                                float* gradV, unsigned long long* touched,
+                                int nfields, int nnz, int K,
+                                ihipStream_t* stream);
+void row_index_launch(const int* row_ptr, int* row_idx, int B,
+                      ihipStream_t* stream);
+
 // --- misc_kernels.hip (generic sparse fused optimizers; D = per-feature
 // latent size, runtime) ---
 void sparse_adagrad_apply_launch(const int* uniq, const int* count, float* W,
