@@ -225,15 +225,27 @@ __global__ void bitmap_compact_kernel(unsigned long long* __restrict__ bitmap,
                                       int nwords, int* __restrict__ out_fids,
                                       int* __restrict__ out_count) {
   const int w = blockIdx.x * blockDim.x + threadIdx.x;
-  if (w >= nwords) return;
-  unsigned long long m = bitmap[w];
-  if (!m) return;
-  bitmap[w] = 0ull;
-  int base = atomicAdd(out_count, __popcll(m));
+  const int lane = threadIdx.x & 63;
+  unsigned long long m = (w < nwords) ? bitmap[w] : 0ull;
+  if (m) bitmap[w] = 0ull;
+  const int n = __popcll(m);
+  // wave-aggregated append: one atomic per wave, lanes offset by prefix sum
+  int scan = n;
+#pragma unroll
+  for (int s = 1; s < LCTR_WAVE; s <<= 1) {
+    const int t = __shfl_up(scan, s);
+    if (lane >= s) scan += t;
+  }
+  const int total = __shfl(scan, LCTR_WAVE - 1);
+  int base_wave = 0;
+  if (lane == LCTR_WAVE - 1 && total > 0)
+    base_wave = atomicAdd(out_count, total);
+  base_wave = __shfl(base_wave, LCTR_WAVE - 1);
+  int off = base_wave + scan - n;
   while (m) {
     const int b = __ffsll((long long)m) - 1;
     m &= m - 1;
-    out_fids[base++] = w * 64 + b;
+    out_fids[off++] = w * 64 + b;
   }
 }
 
