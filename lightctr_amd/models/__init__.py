@@ -10,6 +10,7 @@ from .embedding import EmbedModel, EmbedHyper, vocab_from_tokens
 from .vae import VAEModel, VAEHyper
 from .cnn import CNNModel, CNNHyper, Conv2DLayer, MaxPool2DLayer
 from .rnn import RNNModel, RNNHyper, LSTMUnit, AttentionUnit
+from .lr import LRModel, LRHyper, LRTrainer
 
 __all__ = [
     "FMModel", "FMTrainer", "FMHyper",
@@ -24,4 +25,5 @@ __all__ = [
     "VAEModel", "VAEHyper",
     "CNNModel", "CNNHyper", "Conv2DLayer", "MaxPool2DLayer",
     "RNNModel", "RNNHyper", "LSTMUnit", "AttentionUnit",
+    "LRModel", "LRHyper", "LRTrainer",
 ]

@@ -22,7 +22,7 @@ from dataclasses import dataclass
 import torch
 
 from ..ops import fm_ref
-from ..ops._extension import has_hip_ops, require_hip_ops
+from ..ops._extension import require_hip_ops
 from ..utils.metrics import auc_score
 
 

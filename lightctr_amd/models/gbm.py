@@ -16,8 +16,7 @@ quantile-bucketized to uint8 once, each level builds per-(node, feature)
 
 from __future__ import annotations
 
-import math
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 
 import torch
 

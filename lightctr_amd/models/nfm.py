@@ -9,7 +9,7 @@ segment-reduce sparse backward + fused sparse optimizer.
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 
 import torch
 
