@@ -31,16 +31,30 @@ __global__ void act_backward_kernel(const float* __restrict__ dY,
   if (dZbf) dZbf[i] = (__bf16)g;
 }
 
-// db[n] = sum_m dZ[m,n]. One wave per column, grid-stride over rows.
+// db[n] = sum_m dZ[m,n]. Column-tiled: block (x = column tile, y = row
+// chunk); thread owns one column, reads coalesced across the 256-thread
+// tile, keeps a register partial over its row chunk, one global atomic per
+// (block, column). N==1 collapses to a flat wave-reduced sum.
 __global__ void colsum_kernel(const float* __restrict__ dZ,
                               float* __restrict__ db, int M, int N) {
-  const int col = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
-  const int lane = threadIdx.x & 63;
+  if (N == 1) {
+    const long total = M;
+    float s = 0.f;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+         i += (long)gridDim.x * blockDim.x)
+      s += dZ[i];
+    s = wave_reduce_sum(s);
+    if ((threadIdx.x & 63) == 0 && s != 0.f) atomicAdd(&db[0], s);
+    return;
+  }
+  const int col = blockIdx.x * blockDim.x + threadIdx.x;
   if (col >= N) return;
+  const int chunk = (M + gridDim.y - 1) / gridDim.y;
+  const int m0 = blockIdx.y * chunk;
+  const int m1 = min(M, m0 + chunk);
   float s = 0.f;
-  for (int m = lane; m < M; m += 64) s += dZ[(size_t)m * N + col];
-  s = wave_reduce_sum(s);
-  if (lane == 0) db[col] = s;
+  for (int m = m0; m < m1; ++m) s += dZ[(size_t)m * N + col];
+  if (s != 0.f) atomicAdd(&db[col], s);
 }
 
 // f32 -> bf16 convert (vectorized x4)
@@ -117,8 +131,17 @@ void act_backward_launch(const float* dY, const float* Y, float* dZ,
 
 void colsum_launch(const float* dZ, float* db, int M, int N,
                    hipStream_t stream) {
+  LCTR_CHECK_HIP(hipMemsetAsync(db, 0, (size_t)N * sizeof(float), stream));
   dim3 block(256);
-  dim3 grid((N + 3) / 4);
+  if (N == 1) {
+    dim3 grid(512);
+    hipLaunchKernelGGL(colsum_kernel, grid, block, 0, stream, dZ, db, M, N);
+    return;
+  }
+  const int xtiles = (N + 255) / 256;
+  const int ychunks = max(1, min(1024 / xtiles, (M + 255) / 256));
+  dim3 grid(xtiles * 256 / 256, ychunks);
+  grid.x = xtiles;
   hipLaunchKernelGGL(colsum_kernel, grid, block, 0, stream, dZ, db, M, N);
 }
 
