@@ -134,3 +134,46 @@ def test_sharded_fm_matches_single_model():
 def test_ring_broadcast_allreduce():
     (ok,) = _run_spawn(_ring_worker, 29532)
     assert ok
+
+
+def _ring_dp_mlp_worker(rank, port, q):
+    try:
+        dist = _init(rank, port)
+        import torch as t
+        from lightctr_amd.models.mlp import MLP
+        from lightctr_amd.parallel.ring import (RingDataParallel,
+                                                mlp_param_grads)
+
+        # identical init (same seed); different local batches
+        mlp = MLP([8, 16, 1], optimizer="adagrad", lr=0.05, seed=3,
+                  device="cpu")
+        rdp = RingDataParallel(mlp_param_grads(mlp))
+        rdp.sync_init()
+        g = t.Generator().manual_seed(50 + rank)
+        X = t.randn(32, 8, generator=g)
+        y = t.randn(32, 1, generator=g)
+        for _ in range(3):
+            out = mlp.forward(X)
+            mlp.backward((out - y) / 32)
+            rdp.sync_gradients()  # average grads over the ring
+            mlp.apply_grads()
+        # all ranks must hold identical parameters now
+        W0 = mlp.layers[0].W.clone()
+        gathered = [t.zeros_like(W0) for _ in range(WORLD)]
+        dist.all_gather(gathered, W0)
+        same = all(t.allclose(gathered[0], w, atol=1e-7) for w in gathered)
+        if rank == 0:
+            q.put(("result", bool(same)))
+        dist.destroy_process_group()
+    except Exception:  # pragma: no cover
+        import traceback
+
+        q.put(("error", rank, traceback.format_exc()))
+        raise
+
+
+def test_ring_dp_mlp_sync():
+    """Reference ring-CNN mode semantics on the dense engine: broadcast
+    init + per-step gradient all-reduce keeps replicas identical."""
+    (ok,) = _run_spawn(_ring_dp_mlp_worker, 29533)
+    assert ok
