@@ -106,8 +106,10 @@ def mlp_param_grads(mlp):
     def provider():
         out = []
         for la in mlp.layers:
-            out.append((la.W, la._dW))
-            out.append((la.b, la._db))
+            # grads exist only after a backward pass; sync_init (params
+            # only) may run before the first step
+            out.append((la.W, getattr(la, "_dW", None)))
+            out.append((la.b, getattr(la, "_db", None)))
         return out
 
     return provider
