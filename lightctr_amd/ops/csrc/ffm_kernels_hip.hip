@@ -171,6 +171,7 @@ __global__ void ffm_sorted_backward_kernel(
     const float xi = vals[p];
     const float d = dpred[r];
     const int beg = row_ptr[r], rend = row_ptr[r + 1];
+#pragma unroll 2
     for (int j = beg + jg; j < rend; j += G) {
       if (j == p) continue;
       const float v = V[((size_t)fids[j] * nfields + Fi) * K + k];
@@ -204,7 +205,8 @@ void ffm_sorted_backward_launch(const int* sorted_fids, const long* perm,
                                 float* gradV, unsigned long long* touched,
                                 int nfields, int nnz, int K,
                                 hipStream_t stream) {
-  const int chunk = 128;
+  const int chunk = 64;  // short chunks: more waves in flight to hide the
+                         // per-entry dependent V-gather latency
   const int wpb = 4;
   const int nwaves = (nnz + chunk - 1) / chunk;
   dim3 block(wpb * LCTR_WAVE);

@@ -141,3 +141,17 @@ def test_batch_predictor_report():
     rep = bp.report(pred, ds.labels)
     for key in ("auc", "logloss", "accuracy", "precision", "recall", "f1"):
         assert key in rep
+
+
+def test_cli_train_and_predict(tmp_path):
+    from lightctr_amd.cli import main
+
+    model_path = str(tmp_path / "m.pt")
+    rc = main(["train", "--model", "fm", "--data", "synthetic", "--rows",
+               "512", "--features", "8192", "--k", "8", "--epochs", "2",
+               "--device", "cpu", "--save", model_path])
+    assert rc == 0
+    rc = main(["predict", "--model", "fm", "--data", "synthetic", "--rows",
+               "256", "--features", "8192", "--k", "8", "--device", "cpu",
+               "--load", model_path])
+    assert rc == 0
