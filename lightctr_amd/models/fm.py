@@ -72,6 +72,9 @@ class FMModel:
         self.count = torch.zeros(1, dtype=torch.int32, device=self.device)
         self._use_hip = self.device.type == "cuda"
         self.backward_mode = "sorted"  # "sorted" (default) | "atomic"
+        # fused-apply: interior feature segments get their optimizer update
+        # inside the segment-reduce kernel; slab+compact handles boundaries
+        self.fused_apply = True
         if self._use_hip:
             require_hip_ops()  # fail loudly if extension missing on GPU
 
@@ -105,8 +108,24 @@ class FMModel:
                 gw, gv = ops.fm_backward_emit(row_ptr, fids, vals, self.V,
                                               sumVX, dpred)
                 sorted_fids, perm = torch.sort(fids)
-                ops.fm_sorted_apply(sorted_fids, perm, gw, gv, self.gradW,
-                                    self.gradV, self.touched)
+                if self.fused_apply:
+                    if self.h.optimizer == "ftrl":
+                        ops.fm_sorted_apply_fused(
+                            sorted_fids, perm, gw, gv, self.gradW,
+                            self.gradV, self.touched, self.W, self.V,
+                            self.nW, self.nV, self.zW, self.zV, 2,
+                            self.h.ftrl_alpha, self.h.ftrl_beta,
+                            self.h.ftrl_l1, self.h.ftrl_l2)
+                    else:
+                        ops.fm_sorted_apply_fused(
+                            sorted_fids, perm, gw, gv, self.gradW,
+                            self.gradV, self.touched, self.W, self.V,
+                            self.nW, self.nV, None, None, 1, self.h.lr,
+                            self.h.eps, self.h.l2, 0.0)
+                else:
+                    ops.fm_sorted_apply(sorted_fids, perm, gw, gv,
+                                        self.gradW, self.gradV,
+                                        self.touched)
             else:
                 ops.fm_backward(row_ptr, fids, vals, self.V, sumVX, dpred,
                                 self.gradW, self.gradV, self.touched)

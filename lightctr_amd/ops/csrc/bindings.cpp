@@ -109,7 +109,34 @@ void fm_sorted_apply(at::Tensor sorted_fids, at::Tensor perm, at::Tensor gw,
       sorted_fids.data_ptr<int>(), perm.data_ptr<long>(),
       gw.data_ptr<float>(), gv.data_ptr<float>(), gradW.data_ptr<float>(),
       gradV.data_ptr<float>(), (unsigned long long*)touched.data_ptr(),
-      (int)sorted_fids.numel(), K, cur_stream());
+      (int)sorted_fids.numel(), K, 0, nullptr, nullptr, nullptr, nullptr,
+      nullptr, nullptr, 0.f, 0.f, 0.f, 0.f, cur_stream());
+}
+
+// Fused variant: interior feature segments get their optimizer update
+// applied during the segment reduction; only boundary-spanning features go
+// through the slab + bitmap (+ the follow-up sparse apply).
+void fm_sorted_apply_fused(at::Tensor sorted_fids, at::Tensor perm,
+                           at::Tensor gw, at::Tensor gv, at::Tensor gradW,
+                           at::Tensor gradV, at::Tensor touched,
+                           at::Tensor W, at::Tensor V, at::Tensor nW,
+                           at::Tensor nV, c10::optional<at::Tensor> zW,
+                           c10::optional<at::Tensor> zV, int64_t opt_mode,
+                           double p0, double p1, double p2, double p3) {
+  check_cuda_i32(sorted_fids, "sorted_fids");
+  CHK(perm.scalar_type() == at::kLong, "perm must be int64");
+  CHK(opt_mode == 1 || opt_mode == 2, "opt_mode 1=adagrad 2=ftrl");
+  const int K = (int)gradV.size(1);
+  lightctr::fm_sorted_apply_launch(
+      sorted_fids.data_ptr<int>(), perm.data_ptr<long>(),
+      gw.data_ptr<float>(), gv.data_ptr<float>(), gradW.data_ptr<float>(),
+      gradV.data_ptr<float>(), (unsigned long long*)touched.data_ptr(),
+      (int)sorted_fids.numel(), K, (int)opt_mode, V.data_ptr<float>(),
+      W.data_ptr<float>(), nW.data_ptr<float>(),
+      zW.has_value() ? zW->data_ptr<float>() : nullptr,
+      nV.data_ptr<float>(),
+      zV.has_value() ? zV->data_ptr<float>() : nullptr, (float)p0,
+      (float)p1, (float)p2, (float)p3, cur_stream());
 }
 
 // ---- FFM ----
@@ -498,6 +525,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "FM backward phase 1: per-entry grads (no atomics)");
   m.def("fm_sorted_apply", &fm_sorted_apply,
         "FM backward phase 2: segment-reduce sorted grads into slabs");
+  m.def("fm_sorted_apply_fused", &fm_sorted_apply_fused,
+        "segment-reduce + fused optimizer for interior segments");
   m.def("ffm_forward", &ffm_forward, "FFM fused pairwise forward");
   m.def("ffm_backward", &ffm_backward, "FFM fused pairwise backward scatter");
   m.def("ffm_sorted_backward", &ffm_sorted_backward,

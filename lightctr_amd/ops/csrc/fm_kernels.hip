@@ -19,6 +19,28 @@
 
 namespace lightctr {
 
+// FTRL-proximal (per-coordinate), semantics of the reference FTRL updater
+// (gradientUpdater.h:235-278): z,n state per parameter.
+__device__ __forceinline__ void ftrl_update(float* __restrict__ w,
+                                            float* __restrict__ z,
+                                            float* __restrict__ n, float g,
+                                            float alpha, float beta, float l1,
+                                            float l2) {
+  const float g2 = g * g;
+  const float nold = *n;
+  const float sigma = (sqrtf(nold + g2) - sqrtf(nold)) / alpha;
+  const float znew = *z + g - sigma * (*w);
+  const float nnew = nold + g2;
+  *z = znew;
+  *n = nnew;
+  if (fabsf(znew) <= l1) {
+    *w = 0.f;
+  } else {
+    *w = -(znew - copysignf(l1, znew)) / ((beta + sqrtf(nnew)) / alpha + l2);
+  }
+}
+
+
 // ---------------------------------------------------------------------------
 // Forward: pred[row] = sum_j w[fid]*x + 0.5*(||sumVX||^2 - sum_j ||v*x||^2)
 // Also writes sumVX[row*K+k] (needed by backward).
@@ -155,12 +177,30 @@ __global__ void fm_backward_emit_kernel(
 // K-lane subgroups walk their stride, accumulating runs of equal fid in
 // registers and flushing with one atomicAdd per run. Segment heads also set
 // the touched bitmap (one atomicOr per unique fid).
+// opt_mode: 0 = slab-only (accumulate into gradW/gradV + touched bitmap);
+//           1 = fused Adagrad, 2 = fused FTRL. In fused modes, any feature
+//           whose sorted segment lies ENTIRELY within one subgroup's range
+//           (checked via the global neighbors sorted_fids[first-1] /
+//           sorted_fids[last+1]) holds its TOTAL batch gradient at flush
+//           time, so the optimizer update is applied in place with no
+//           atomics and the slab/bitmap/compaction/apply pass only handles
+//           the ~2 boundary-spanning features per subgroup.
+struct FmOptArgs {
+  float* W;
+  float* nW;
+  float* zW;
+  float* nV;
+  float* zV;
+  float p0, p1, p2, p3;  // adagrad: lr, eps, l2 | ftrl: alpha, beta, l1, l2
+};
+
 template <int K>
 __global__ void fm_sorted_apply_kernel(
     const int* __restrict__ sorted_fids, const long* __restrict__ perm,
     const float* __restrict__ gw, const float* __restrict__ gv,
     float* __restrict__ gradW, float* __restrict__ gradV,
-    unsigned long long* __restrict__ touched, int nnz, int chunk) {
+    unsigned long long* __restrict__ touched, int nnz, int chunk,
+    int opt_mode, float* __restrict__ V, FmOptArgs oa) {
   constexpr int G = LCTR_WAVE / K;
   const int lane = threadIdx.x & (LCTR_WAVE - 1);
   const int wave = blockIdx.x * (blockDim.x / LCTR_WAVE) + (threadIdx.x >> 6);
@@ -178,7 +218,45 @@ __global__ void fm_sorted_apply_kernel(
   const int se = min(sb + sub, end);
 
   int cur_fid = -1;
+  bool head_ok = false;  // this subgroup saw the segment's global start
   float acc = 0.f, accw = 0.f;
+
+  auto flush = [&](int tail_e) {
+    if (cur_fid < 0) return;
+    // tail_e = index of the first entry AFTER the run
+    const bool tail_ok =
+        (tail_e >= nnz) || (sorted_fids[tail_e] != cur_fid);
+    if (opt_mode != 0 && head_ok && tail_ok) {
+      // exclusive owner of this feature's total gradient: fused update
+      const size_t off = (size_t)cur_fid * K + k;
+      if (opt_mode == 1) {  // adagrad
+        const float gg = acc + oa.p2 * V[off];
+        const float a = oa.nV[off] + gg * gg;
+        oa.nV[off] = a;
+        V[off] -= oa.p0 * gg * __frsqrt_rn(a + oa.p1);
+        if (k == 0) {
+          const float gwv = accw + oa.p2 * oa.W[cur_fid];
+          const float aw = oa.nW[cur_fid] + gwv * gwv;
+          oa.nW[cur_fid] = aw;
+          oa.W[cur_fid] -= oa.p0 * gwv * __frsqrt_rn(aw + oa.p1);
+        }
+      } else {  // ftrl
+        ftrl_update(&V[off], &oa.zV[off], &oa.nV[off], acc, oa.p0, oa.p1,
+                    oa.p2, oa.p3);
+        if (k == 0) {
+          ftrl_update(&oa.W[cur_fid], &oa.zW[cur_fid], &oa.nW[cur_fid],
+                      accw, oa.p0, oa.p1, oa.p2, oa.p3);
+        }
+      }
+    } else {
+      atomicAdd(&gradV[(size_t)cur_fid * K + k], acc);
+      if (k == 0) {
+        atomicAdd(&gradW[cur_fid], accw);
+        atomicOr(&touched[cur_fid >> 6], 1ull << (cur_fid & 63));
+      }
+    }
+  };
+
   for (int e = sb; e < se; e += 4) {
     const int nvalid = min(4, se - e);
     float v[4], vw[4];
@@ -196,25 +274,17 @@ __global__ void fm_sorted_apply_kernel(
     for (int u = 0; u < 4; ++u) {
       if (u >= nvalid) break;
       if (f[u] != cur_fid) {
-        if (cur_fid >= 0) {
-          atomicAdd(&gradV[(size_t)cur_fid * K + k], acc);
-          if (k == 0) atomicAdd(&gradW[cur_fid], accw);
-        }
+        flush(e + u);
         cur_fid = f[u];
         acc = 0.f;
         accw = 0.f;
-        if (k == 0 && (e + u == 0 || sorted_fids[e + u - 1] != f[u])) {
-          atomicOr(&touched[f[u] >> 6], 1ull << (f[u] & 63));
-        }
+        head_ok = (e + u == 0 || sorted_fids[e + u - 1] != f[u]);
       }
       acc += v[u];
       accw += vw[u];
     }
   }
-  if (cur_fid >= 0) {
-    atomicAdd(&gradV[(size_t)cur_fid * K + k], acc);
-    if (k == 0) atomicAdd(&gradW[cur_fid], accw);
-  }
+  flush(se);
 }
 
 // ---------------------------------------------------------------------------
@@ -279,27 +349,6 @@ __global__ void fm_adagrad_apply_kernel(
     nW[fid] = a;
     W[fid] -= lr * gw * __frsqrt_rn(a + eps);
     gradW[fid] = 0.f;
-  }
-}
-
-// FTRL-proximal (per-coordinate), semantics of the reference FTRL updater
-// (gradientUpdater.h:235-278): z,n state per parameter.
-__device__ __forceinline__ void ftrl_update(float* __restrict__ w,
-                                            float* __restrict__ z,
-                                            float* __restrict__ n, float g,
-                                            float alpha, float beta, float l1,
-                                            float l2) {
-  const float g2 = g * g;
-  const float nold = *n;
-  const float sigma = (sqrtf(nold + g2) - sqrtf(nold)) / alpha;
-  const float znew = *z + g - sigma * (*w);
-  const float nnew = nold + g2;
-  *z = znew;
-  *n = nnew;
-  if (fabsf(znew) <= l1) {
-    *w = 0.f;
-  } else {
-    *w = -(znew - copysignf(l1, znew)) / ((beta + sqrtf(nnew)) / alpha + l2);
   }
 }
 
@@ -376,15 +425,20 @@ void fm_backward_emit_launch(const int* row_ptr, const int* fids,
 void fm_sorted_apply_launch(const int* sorted_fids, const long* perm,
                             const float* gw, const float* gv, float* gradW,
                             float* gradV, unsigned long long* touched, int nnz,
-                            int K, hipStream_t stream) {
+                            int K, int opt_mode, float* V, float* W,
+                            float* nW, float* zW, float* nV, float* zV,
+                            float p0, float p1, float p2, float p3,
+                            hipStream_t stream) {
   const int chunk = 256;
   const int wpb = waves_per_block();
   const int nwaves = (nnz + chunk - 1) / chunk;
   dim3 block(wpb * LCTR_WAVE);
   dim3 grid((nwaves + wpb - 1) / wpb);
+  FmOptArgs oa{W, nW, zW, nV, zV, p0, p1, p2, p3};
   DISPATCH_K(K, hipLaunchKernelGGL((fm_sorted_apply_kernel<KC>), grid, block,
                                    0, stream, sorted_fids, perm, gw, gv, gradW,
-                                   gradV, touched, nnz, chunk));
+                                   gradV, touched, nnz, chunk, opt_mode, V,
+                                   oa));
 }
 
 void bitmap_compact_launch(unsigned long long* bitmap, int nwords,

@@ -169,3 +169,27 @@ def test_fm_gpu_vs_cpu_one_step():
     gpu.train_step(row_ptr.cuda(), fids.cuda(), vals.cuda(), labels.cuda())
     assert torch.allclose(gpu.W.cpu(), cpu.W, atol=1e-4, rtol=1e-4)
     assert torch.allclose(gpu.V.cpu(), cpu.V, atol=1e-4, rtol=1e-4)
+
+
+def test_fused_apply_matches_two_phase():
+    """fused optimizer-in-apply == slab+compact+apply two-phase, both
+    optimizers, on skewed data (long runs exercise interior fast path)."""
+    from lightctr_amd.data.synthetic import SyntheticCriteo
+
+    for opt in ("adagrad", "ftrl"):
+        h = FMHyper(num_features=1 << 15, k=16, optimizer=opt, seed=31)
+        a = FMModel(h, device="cuda:0")
+        b = FMModel(h, device="cuda:0")
+        b.W.copy_(a.W)
+        b.V.copy_(a.V)
+        b.fused_apply = False
+        gen = SyntheticCriteo(num_features=1 << 15, seed=77, device="cuda:0")
+        for _ in range(3):
+            row_ptr, fields, fids, vals, labels = gen.batch(4096)
+            a.train_step(row_ptr, fids, vals, labels)
+            b.train_step(row_ptr, fids, vals, labels)
+        assert torch.allclose(a.W, b.W, atol=1e-5), \
+            (a.W - b.W).abs().max()
+        assert torch.allclose(a.V, b.V, atol=1e-5), \
+            (a.V - b.V).abs().max()
+        assert torch.allclose(a.nV, b.nV, atol=1e-5)
