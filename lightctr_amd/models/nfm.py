@@ -118,17 +118,8 @@ class NFMModel:
             gw, gv = ops.nfm_backward_emit(row_ptr, fids, vals, self.V, sumVX,
                                            ddeep.contiguous(), dpred)
             sorted_fids, perm = torch.sort(fids)
-            if self.h.optimizer == "ftrl":
-                ops.fm_sorted_apply_fused(
-                    sorted_fids, perm, gw, gv, self.gradW, self.gradV,
-                    self.touched, self.W, self.V, self.nW, self.nV,
-                    self.zW, self.zV, 2, self.h.ftrl_alpha,
-                    self.h.ftrl_beta, self.h.ftrl_l1, self.h.ftrl_l2)
-            else:
-                ops.fm_sorted_apply_fused(
-                    sorted_fids, perm, gw, gv, self.gradW, self.gradV,
-                    self.touched, self.W, self.V, self.nW, self.nV,
-                    None, None, 1, self.h.lr, self.h.eps, self.h.l2, 0.0)
+            ops.fm_sorted_apply(sorted_fids, perm, gw, gv, self.gradW,
+                                self.gradV, self.touched)
             self.count.zero_()
             ops.bitmap_compact(self.touched, self.uniq, self.count)
             live = self.uniq[: min(self.uniq.numel(), fids.numel())]

@@ -121,17 +121,8 @@ class WideDeepModel:
                                              dDeep.contiguous(), dpred,
                                              self.h.num_fields, self.h.k)
             sorted_fids, perm = torch.sort(fids)
-            if self.h.optimizer == "ftrl":
-                ops.fm_sorted_apply_fused(
-                    sorted_fids, perm, gw, gv, self.gradW, self.gradE,
-                    self.touched, self.W, self.E, self.nW, self.nE,
-                    self.zW, self.zE, 2, self.h.ftrl_alpha,
-                    self.h.ftrl_beta, self.h.ftrl_l1, self.h.ftrl_l2)
-            else:
-                ops.fm_sorted_apply_fused(
-                    sorted_fids, perm, gw, gv, self.gradW, self.gradE,
-                    self.touched, self.W, self.E, self.nW, self.nE,
-                    None, None, 1, self.h.lr, self.h.eps, self.h.l2, 0.0)
+            ops.fm_sorted_apply(sorted_fids, perm, gw, gv, self.gradW,
+                                self.gradE, self.touched)
             self.count.zero_()
             ops.bitmap_compact(self.touched, self.uniq, self.count)
             live = self.uniq[: min(self.uniq.numel(), fids.numel())]
