@@ -1,0 +1,170 @@
+#include "hip/hip_runtime.h"
+// 256x256-tile phase-split MFMA bf16 GEMM for MI355X/gfx950.
+//
+// An experiment toward the CDNA4 guide's 8-phase 256^2 template: bigger
+// tile (2x the compute per byte of the 128^2 kernel), 8 waves (2Mx4N),
+// per-wave 128x64 output (8x4 fragments), double-buffered 128 KiB LDS
+// staged by global_load_lds, ONE barrier + one vmcnt drain per K-tile
+// (vs 2 barriers per tile in the 128^2 kernel), compute split into 4
+// phases (one 2x4-fragment quadrant x full K-tile each) wrapped in
+// s_setprio so the CU scheduler favors MFMA-entering waves (T5; needs the
+// phase role-split to matter). B fragments for the whole tile are kept in
+// registers across phases (read once per tile).
+//
+// Eligibility: TRANSA=TRANSB=0, M%256==0, N%256==0, K%64==0, K%8==0 —
+// dispatched from gemm_bf16_launch; everything else takes the 128^2 path.
+#include "common.h"
+
+namespace lightctr {
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+#define G256_BK 64
+
+__device__ __forceinline__ int swz256(int row, int ke) {
+  return ke ^ ((row & 7) << 3);
+}
+
+// stage one 128-row x 64-col bf16 half-tile with 512 threads (2 glds each)
+__device__ __forceinline__ void stage_half_512(
+    const __bf16* __restrict__ gbase, long stride, __bf16* dst) {
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;  // 0..7
+#pragma unroll
+  for (int it = 0; it < 2; ++it) {
+    const int seg = wave * 2 + it;  // 16 segments of 8 rows
+    const int row = seg * 8 + (lane >> 3);
+    const int ke = swz256(row, (lane & 7) * 8);
+    const __bf16* src = gbase + (long)row * stride + ke;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)src,
+        (__attribute__((address_space(3))) void*)(dst + seg * 512), 16, 0,
+        0);
+  }
+}
+
+__device__ __forceinline__ float act_apply256(float v, int act) {
+  if (act == 1) return fmaxf(v, 0.f);
+  if (act == 2) return sigmoidf_clamped(v);
+  return v;
+}
+
+__global__ __launch_bounds__(512, 1) void gemm256_bf16_kernel(
+    const __bf16* __restrict__ A, const __bf16* __restrict__ Bst,
+    const float* __restrict__ bias, float* __restrict__ C,
+    __bf16* __restrict__ Cbf, int M, int N, int K, int act) {
+  // one __shared__ object: [buf][A(256x64) then B(256x64)]
+  __shared__ __bf16 smem[2 * 2 * 256 * G256_BK];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wm = wave >> 2;      // 0..1
+  const int wn = wave & 3;       // 0..3
+  const int M0 = blockIdx.y * 256;
+  const int N0 = blockIdx.x * 256;
+
+  __bf16* bufA[2] = {smem, smem + 2 * 256 * G256_BK};
+  __bf16* bufB[2] = {smem + 256 * G256_BK, smem + 3 * 256 * G256_BK};
+
+  f32x4 acc[8][4] = {};
+
+  const int NT = K / G256_BK;
+  // prologue: stage tile 0 (4 half-tiles: A-top, A-bot, B-top, B-bot)
+  stage_half_512(A + (long)M0 * K, K, bufA[0]);
+  stage_half_512(A + (long)(M0 + 128) * K, K, bufA[0] + 128 * G256_BK);
+  stage_half_512(Bst + (long)N0 * K, K, bufB[0]);
+  stage_half_512(Bst + (long)(N0 + 128) * K, K, bufB[0] + 128 * G256_BK);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  int cur = 0;
+  for (int t = 0; t < NT; ++t) {
+    const int k0 = (t + 1) * G256_BK;
+    const bool more = (t + 1) < NT;
+    // B fragments for this wave's 64 columns, whole K-tile, kept in regs
+    bf16x8 b[4][2];
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn)
+#pragma unroll
+      for (int kc = 0; kc < 2; ++kc) {
+        const int rb = wn * 64 + fn * 16 + (lane & 15);
+        const int ks = kc * 32 + (lane >> 4) * 8;
+        b[fn][kc] = *(const bf16x8*)&bufB[cur][rb * G256_BK + swz256(rb, ks)];
+      }
+    // 4 phases: quadrant p = fm in [2p, 2p+2), all fn, full K-tile
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      if (more) {  // prefetch half-tile p of tile t+1
+        if (p == 0)
+          stage_half_512(A + (long)M0 * K + k0, K, bufA[cur ^ 1]);
+        else if (p == 1)
+          stage_half_512(A + (long)(M0 + 128) * K + k0, K,
+                         bufA[cur ^ 1] + 128 * G256_BK);
+        else if (p == 2)
+          stage_half_512(Bst + (long)N0 * K + k0, K, bufB[cur ^ 1]);
+        else
+          stage_half_512(Bst + (long)(N0 + 128) * K + k0, K,
+                         bufB[cur ^ 1] + 128 * G256_BK);
+      }
+      bf16x8 a[2][2];
+#pragma unroll
+      for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+        for (int kc = 0; kc < 2; ++kc) {
+          const int ra = wm * 128 + (p * 2 + fm) * 16 + (lane & 15);
+          const int ks = kc * 32 + (lane >> 4) * 8;
+          a[fm][kc] =
+              *(const bf16x8*)&bufA[cur][ra * G256_BK + swz256(ra, ks)];
+        }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < 4; ++fn)
+#pragma unroll
+          for (int kc = 0; kc < 2; ++kc)
+            acc[p * 2 + fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a[fm][kc], b[fn][kc], acc[p * 2 + fm][fn], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    // drain the prefetch, then one barrier before buffers swap roles
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  // epilogue (C/D map: col=lane&15, row=(lane>>4)*4+r)
+#pragma unroll
+  for (int fm = 0; fm < 8; ++fm) {
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn) {
+      const int col = N0 + wn * 64 + fn * 16 + (lane & 15);
+      const float bv = bias ? bias[col] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = M0 + wm * 128 + fm * 16 + (lane >> 4) * 4 + r;
+        const float v = act_apply256(acc[fm][fn][r] + bv, act);
+        C[(size_t)row * N + col] = v;
+        if (Cbf) Cbf[(size_t)row * N + col] = (__bf16)v;
+      }
+    }
+  }
+}
+
+bool gemm256_eligible(int M, int N, int K, int transA, int transB) {
+  return transA == 0 && transB == 0 && M % 256 == 0 && N % 256 == 0 &&
+         K % G256_BK == 0;
+}
+
+void gemm256_bf16_launch(const void* A, const void* Bst, const float* bias,
+                         float* C, void* Cbf, int M, int N, int K, int act,
+                         hipStream_t stream) {
+  dim3 block(512);
+  dim3 grid(N / 256, M / 256);
+  hipLaunchKernelGGL(gemm256_bf16_kernel, grid, block, 0, stream,
+                     (const __bf16*)A, (const __bf16*)Bst, bias, C,
+                     (__bf16*)Cbf, M, N, K, act);
+}
+
+}  // namespace lightctr

@@ -81,6 +81,11 @@ void gemm_bf16_launch(const void* A, const void* Bst, const float* bias,
                       float* C, void* Cbf, int M, int N, int K, int transA,
                       int transB, int act, ihipStream_t* stream);
 
+bool gemm256_eligible(int M, int N, int K, int transA, int transB);
+void gemm256_bf16_launch(const void* A, const void* Bst, const float* bias,
+                         float* C, void* Cbf, int M, int N, int K, int act,
+                         ihipStream_t* stream);
+
 // --- nn_kernels.hip ---
 void act_backward_launch(const float* dY, const float* Y, float* dZ,
                          void* dZbf, long n, int act, ihipStream_t* stream);

@@ -79,7 +79,8 @@ def test_widedeep_cpu_convergence():
 @pytest.mark.parametrize("shape", [(128, 128, 64), (256, 384, 156),
                                    (1000, 250, 100), (64, 1, 256),
                                    (513, 129, 65), (512, 256, 128),
-                                   (384, 128, 624)])
+                                   (384, 128, 624), (256, 256, 64),
+                                   (512, 512, 192)])
 def test_gemm_bf16_parity(shape):
     """MFMA GEMM vs torch bf16 matmul (fp32 accumulate) incl. odd tails.
 
