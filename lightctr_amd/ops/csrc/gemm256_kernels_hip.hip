@@ -64,22 +64,24 @@ __global__ __launch_bounds__(512, 1) void gemm256_bf16_kernel(
   const int M0 = blockIdx.y * 256;
   const int N0 = blockIdx.x * 256;
 
-  __bf16* bufA[2] = {smem, smem + 2 * 256 * G256_BK};
-  __bf16* bufB[2] = {smem + 256 * G256_BK, smem + 3 * 256 * G256_BK};
-
   f32x4 acc[8][4] = {};
 
   const int NT = K / G256_BK;
+  // buffer b: A at smem + b*32768, B at A + 16384 elems
   // prologue: stage tile 0 (4 half-tiles: A-top, A-bot, B-top, B-bot)
-  stage_half_512(A + (long)M0 * K, K, bufA[0]);
-  stage_half_512(A + (long)(M0 + 128) * K, K, bufA[0] + 128 * G256_BK);
-  stage_half_512(Bst + (long)N0 * K, K, bufB[0]);
-  stage_half_512(Bst + (long)(N0 + 128) * K, K, bufB[0] + 128 * G256_BK);
+  stage_half_512(A + (long)M0 * K, K, smem);
+  stage_half_512(A + (long)(M0 + 128) * K, K, smem + 128 * G256_BK);
+  stage_half_512(Bst + (long)N0 * K, K, smem + 256 * G256_BK);
+  stage_half_512(Bst + (long)(N0 + 128) * K, K, smem + 384 * G256_BK);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
 
   int cur = 0;
   for (int t = 0; t < NT; ++t) {
+    __bf16* curA = smem + cur * 512 * G256_BK;
+    __bf16* curB = curA + 256 * G256_BK;
+    __bf16* nxtA = smem + (cur ^ 1) * 512 * G256_BK;
+    __bf16* nxtB = nxtA + 256 * G256_BK;
     const int k0 = (t + 1) * G256_BK;
     const bool more = (t + 1) < NT;
     // B fragments for this wave's 64 columns, whole K-tile, kept in regs
@@ -90,22 +92,22 @@ __global__ __launch_bounds__(512, 1) void gemm256_bf16_kernel(
       for (int kc = 0; kc < 2; ++kc) {
         const int rb = wn * 64 + fn * 16 + (lane & 15);
         const int ks = kc * 32 + (lane >> 4) * 8;
-        b[fn][kc] = *(const bf16x8*)&bufB[cur][rb * G256_BK + swz256(rb, ks)];
+        b[fn][kc] = *(const bf16x8*)&curB[rb * G256_BK + swz256(rb, ks)];
       }
     // 4 phases: quadrant p = fm in [2p, 2p+2), all fn, full K-tile
 #pragma unroll
     for (int p = 0; p < 4; ++p) {
       if (more) {  // prefetch half-tile p of tile t+1
         if (p == 0)
-          stage_half_512(A + (long)M0 * K + k0, K, bufA[cur ^ 1]);
+          stage_half_512(A + (long)M0 * K + k0, K, nxtA);
         else if (p == 1)
           stage_half_512(A + (long)(M0 + 128) * K + k0, K,
-                         bufA[cur ^ 1] + 128 * G256_BK);
+                         nxtA + 128 * G256_BK);
         else if (p == 2)
-          stage_half_512(Bst + (long)N0 * K + k0, K, bufB[cur ^ 1]);
+          stage_half_512(Bst + (long)N0 * K + k0, K, nxtB);
         else
           stage_half_512(Bst + (long)(N0 + 128) * K + k0, K,
-                         bufB[cur ^ 1] + 128 * G256_BK);
+                         nxtB + 128 * G256_BK);
       }
       bf16x8 a[2][2];
 #pragma unroll
@@ -115,7 +117,7 @@ __global__ __launch_bounds__(512, 1) void gemm256_bf16_kernel(
           const int ra = wm * 128 + (p * 2 + fm) * 16 + (lane & 15);
           const int ks = kc * 32 + (lane >> 4) * 8;
           a[fm][kc] =
-              *(const bf16x8*)&bufA[cur][ra * G256_BK + swz256(ra, ks)];
+              *(const bf16x8*)&curA[ra * G256_BK + swz256(ra, ks)];
         }
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
