@@ -69,6 +69,15 @@ class LibffmDataset:
 
 
 def load_libffm(path: str, max_rows: int | None = None) -> LibffmDataset:
+    try:  # native C++ parser (ops extension); Python fallback below
+        from ..ops._extension import has_hip_ops, require_hip_ops
+
+        if has_hip_ops():
+            rp, fl, fi, v, lb = require_hip_ops().parse_libffm(
+                path, -1 if max_rows is None else max_rows)
+            return LibffmDataset(rp, fl, fi, v, lb)
+    except ImportError:
+        pass
     row_ptr = [0]
     fields: list[int] = []
     fids: list[int] = []
