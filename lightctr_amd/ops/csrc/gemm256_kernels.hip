@@ -60,22 +60,11 @@ __global__ __launch_bounds__(512, 1) void gemm256_bf16_kernel(
   const int wave = threadIdx.x >> 6;
   const int wm = wave >> 2;      // 0..1
   const int wn = wave & 3;       // 0..3
-  // XCD-aware tile remap (T1): dispatch places block b on XCD b%8; give
-  // each XCD a contiguous chunk of the logical tile grid so neighboring
-  // tiles (sharing A rows / B cols) hit the same per-XCD L2. Bijective for
-  // any grid size.
-  const int nwg = gridDim.x * gridDim.y;
-  const int bid = blockIdx.y * gridDim.x + blockIdx.x;
-  int tile;
-  if (nwg >= 16) {
-    const int q = nwg / 8, r = nwg % 8;
-    const int xcd = bid % 8, idx = bid / 8;
-    tile = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
-  } else {
-    tile = bid;
-  }
-  const int M0 = (tile / gridDim.x) * 256;
-  const int N0 = (tile % gridDim.x) * 256;
+  // (XCD-aware tile remap was tried here and measured -5% at 4k^3 and 8k^3
+  // on this schedule — 1 block/CU with 512-thread blocks leaves little
+  // cross-block L2 reuse to recover; plain mapping kept.)
+  const int M0 = blockIdx.y * 256;
+  const int N0 = blockIdx.x * 256;
 
   f32x4 acc[8][4] = {};
 
