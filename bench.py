@@ -60,6 +60,72 @@ def build_model(args, device, world):
     raise SystemExit(f"unknown model {args.model}")
 
 
+def run_ps_mode(args, device, world, rank, dist):
+    """Async PS benchmark (BASELINE config #5): ranks [0, world/2) serve
+    table shards, ranks [world/2, world) train FM workers at their own pace
+    with SSP/DCASGD semantics and int8 gradient compression on the wire.
+    Whole-job examples/sec = total worker examples / wall time (max over
+    workers)."""
+    import time as _t
+
+    from lightctr_amd.data.synthetic import SyntheticCriteo
+    from lightctr_amd.parallel.ps import (PSConfig, PSShard, ps_train_fm,
+                                          setup_pair_groups)
+
+    n_ps = max(1, world // 2)
+    cfg = PSConfig(num_features=args.features, k=args.k, ps_shards=n_ps,
+                   updater="dcasgd", lr=0.05, wire="int8", staleness=10)
+    groups = setup_pair_groups(cfg)
+    dist.barrier()
+    t0 = _t.perf_counter()
+    if rank < n_ps:
+        shard = PSShard(cfg, device=device)
+        shard.serve(groups)
+        elapsed = _t.perf_counter() - t0
+    else:
+        gen = SyntheticCriteo(num_features=args.features, seed=77 + rank,
+                              device=device)
+        pool = [gen.batch(args.batch) for _ in range(2)]
+
+        def gen_batch(step):
+            rp, fl, fi, v, lb = pool[step % 2]
+            return rp, fi, v, lb
+
+        ps_train_fm(cfg, groups[rank], gen_batch, steps=args.warmup,
+                    batch_size=args.batch, device=device, fin=False)
+        t0 = _t.perf_counter()
+        ps_train_fm(cfg, groups[rank], gen_batch, steps=args.steps,
+                    batch_size=args.batch, device=device,
+                    epoch_base=args.warmup)
+        elapsed = _t.perf_counter() - t0
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+    t = torch.tensor([elapsed if rank >= n_ps else 0.0],
+                     device=device if torch.cuda.is_available() else "cpu")
+    dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed = float(t.item())
+    if rank == 0:
+        n_workers = world - n_ps
+        total = args.batch * args.steps * n_workers
+        import json as _json
+
+        print(_json.dumps({
+            "metric": "examples/sec (whole node), FM async-PS training on "
+                      "synthetic Criteo-shaped sparse",
+            "value": total / elapsed, "unit": "examples/sec",
+            "n_gpus": world, "steps": args.steps, "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000,
+            "higher_is_better": True, "scaling": "weak",
+            "vs_baseline": None, "dtype": "fp32", "data": "synthetic",
+            "config": {"model": f"FM k={args.k}, async PS (DCASGD, SSP, "
+                                "int8 wire)",
+                       "global_batch": args.batch * n_workers,
+                       "num_features": args.features, "num_fields": 39,
+                       "parallelism": f"{n_ps} PS shards + {n_workers} "
+                                      "workers, P2P"}}))
+    dist.destroy_process_group()
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -74,6 +140,11 @@ def main():
                     choices=["fm", "ffm", "nfm", "widedeep"])
     ap.add_argument("--no-graph", action="store_true",
                     help="disable hipGraph capture of the step")
+    ap.add_argument("--mode", default="sharded",
+                    choices=["sharded", "ps"],
+                    help="N>1 layout: synchronous hash-sharded all-to-all, "
+                         "or async PS (half the ranks serve shards, half "
+                         "train; BASELINE config #5)")
     args = ap.parse_args()
 
     from lightctr_amd.data.synthetic import SyntheticCriteo
@@ -92,6 +163,9 @@ def main():
 
         dist = dist_mod
         dist.init_process_group(backend="nccl" if have_gpu else "gloo")
+
+    if world > 1 and args.mode == "ps" and args.model == "fm":
+        return run_ps_mode(args, device, world, rank, dist)
 
     model, flavor = build_model(args, device, world)
 

@@ -333,7 +333,8 @@ class PSWorker:
 
 def ps_train_fm(cfg: PSConfig, group, gen_batch, steps: int,
                 batch_size: int, device: str = "cpu",
-                epoch_per_step: int = 1):
+                epoch_per_step: int = 1, fin: bool = True,
+                epoch_base: int = 0):
     """Worker-side FM training loop against the PS (reference
     Distributed_Algo_Abst::Train / batchGradCompute shape,
     distributed_algo_abst.h:130-280). gen_batch(step) -> CSR batch.
@@ -345,7 +346,7 @@ def ps_train_fm(cfg: PSConfig, group, gen_batch, steps: int,
     losses = []
     use_hip = torch.device(device).type == "cuda"
     for step in range(steps):
-        epoch = step * epoch_per_step
+        epoch = epoch_base + step * epoch_per_step
         row_ptr, fids, vals, labels = gen_batch(step)
         uniq, inverse = torch.unique(fids.long(), return_inverse=True)
         Wl, Vl = worker.pull(uniq, epoch)
@@ -372,5 +373,6 @@ def ps_train_fm(cfg: PSConfig, group, gen_batch, steps: int,
                                               sumVX, dpred)
         worker.push(uniq, gWl, gVl, epoch)
         losses.append(float(loss.mean()))
-    worker.fin()
+    if fin:
+        worker.fin()
     return losses
