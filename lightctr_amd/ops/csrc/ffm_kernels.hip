@@ -138,9 +138,12 @@ __global__ void ffm_sorted_backward_kernel(
   const int base = wave * chunk;
   if (base >= nnz) return;
   const int end = min(base + chunk, nnz);
-  float* acc = &lds_acc[wave_in_blk * nfields * K];
+  // padded field stride (K+1): field rows land on coprime-64 bank offsets,
+  // so the 64-lane (field-group, k) atomicAdd pattern is conflict-free
+  const int fstride = K + 1;
+  float* acc = &lds_acc[wave_in_blk * nfields * fstride];
   const int blk = nfields * K;
-  for (int i = lane; i < blk; i += LCTR_WAVE) acc[i] = 0.f;
+  for (int i = lane; i < nfields * fstride; i += LCTR_WAVE) acc[i] = 0.f;
   // no __syncthreads needed: each wave owns its LDS slice; lanes of one
   // wave execute in lockstep
 
@@ -151,9 +154,10 @@ __global__ void ffm_sorted_backward_kernel(
     if (fid != cur) {
       if (cur >= 0) {
         for (int i = lane; i < blk; i += LCTR_WAVE) {
-          if (acc[i] != 0.f) {
-            atomicAdd(&gradV[(size_t)cur * blk + i], acc[i]);
-            acc[i] = 0.f;
+          const int src = (i / K) * fstride + (i % K);
+          if (acc[src] != 0.f) {
+            atomicAdd(&gradV[(size_t)cur * blk + i], acc[src]);
+            acc[src] = 0.f;
           }
         }
         if (lane == 0) atomicAdd(&gradW[cur], accw);
@@ -174,13 +178,15 @@ __global__ void ffm_sorted_backward_kernel(
     for (int j = beg + jg; j < rend; j += G) {
       if (j == p) continue;
       const float v = V[((size_t)fids[j] * nfields + Fi) * K + k];
-      atomicAdd(&acc[fields[j] * K + k], d * xi * vals[j] * v);
+      atomicAdd(&acc[fields[j] * fstride + k], d * xi * vals[j] * v);
     }
     if (lane == 0) accw += d * xi;
   }
   if (cur >= 0) {
     for (int i = lane; i < blk; i += LCTR_WAVE) {
-      if (acc[i] != 0.f) atomicAdd(&gradV[(size_t)cur * blk + i], acc[i]);
+      const int src = (i / K) * fstride + (i % K);
+      if (acc[src] != 0.f)
+        atomicAdd(&gradV[(size_t)cur * blk + i], acc[src]);
     }
     if (lane == 0) atomicAdd(&gradW[cur], accw);
   }
@@ -210,7 +216,7 @@ void ffm_sorted_backward_launch(const int* sorted_fids, const long* perm,
   const int nwaves = (nnz + chunk - 1) / chunk;
   dim3 block(wpb * LCTR_WAVE);
   dim3 grid((nwaves + wpb - 1) / wpb);
-  const size_t lds = (size_t)wpb * nfields * K * sizeof(float);
+  const size_t lds = (size_t)wpb * nfields * (K + 1) * sizeof(float);
   DISPATCH_FFM_K(K, hipLaunchKernelGGL((ffm_sorted_backward_kernel<KC>),
                                        grid, block, lds, stream, sorted_fids,
                                        perm, row_of_entry, row_ptr, fields,
