@@ -211,7 +211,7 @@ void ffm_sorted_backward_launch(const int* sorted_fids, const long* perm,
                                 float* gradV, unsigned long long* touched,
                                 int nfields, int nnz, int K,
                                 hipStream_t stream) {
-  const int chunk = 64;  // short chunks: more waves in flight to hide the
+  const int chunk = 32;  // short chunks: more waves in flight to hide the
                          // per-entry dependent V-gather latency
   const int wpb = 4;
   const int nwaves = (nnz + chunk - 1) / chunk;
