@@ -115,6 +115,27 @@ def mlp_param_grads(mlp):
     return provider
 
 
+def cnn_param_grads(cnn):
+    """Provider for models.cnn.CNNModel — the reference's ring-mode CNN
+    (train_cnn_algo.h:64-70,91-97: registerInitializer/registerGradient
+    walking the layer chain) expressed as a (param, grad) walk over the
+    conv stack + head MLP."""
+
+    def provider():
+        out = []
+        for la in cnn.layers:
+            fc = getattr(la, "fc", None)
+            if fc is not None:
+                out.append((fc.W, getattr(fc, "_dW", None)))
+                out.append((fc.b, getattr(fc, "_db", None)))
+        for dl in cnn.head.layers:
+            out.append((dl.W, getattr(dl, "_dW", None)))
+            out.append((dl.b, getattr(dl, "_db", None)))
+        return out
+
+    return provider
+
+
 def mlp_params(mlp):
     out = []
     for la in mlp.layers:

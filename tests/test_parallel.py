@@ -249,3 +249,51 @@ def test_sharded_fm_world4_multi_step():
     msg = q.get()
     assert msg[0] == "result", f"worker error: {msg}"
     assert msg[1], f"3-step world-4 sharded FTRL diverged: {msg[2]}"
+
+
+def _ring_cnn_worker(rank, port, q):
+    try:
+        dist = _init(rank, port)
+        import torch as t
+        from lightctr_amd.models.cnn import CNNHyper, CNNModel
+        from lightctr_amd.parallel.ring import (RingDataParallel,
+                                                cnn_param_grads)
+
+        m = CNNModel(CNNHyper(in_shape=(1, 8, 8), n_classes=2, lr=1e-2,
+                              seed=9))
+        rdp = RingDataParallel(cnn_param_grads(m))
+        rdp.sync_init()
+        g = t.Generator().manual_seed(70 + rank)
+        X = t.rand(16, 1, 8, 8, generator=g)
+        y = t.randint(0, 2, (16,), generator=g)
+        for _ in range(2):
+            # reference ring-CNN semantics: local fwd/bwd, ring all-reduce
+            # of the fused grads, then the per-layer optimizer step
+            logits = m.forward(X, train=True)
+            p = t.softmax(logits, dim=1)
+            yk = t.nn.functional.one_hot(y, 2).float()
+            dx = m.head.backward((p - yk) / 16)
+            for la in reversed(m.layers):
+                dx = la.backward(dx)
+            rdp.sync_gradients()
+            m.head.apply_grads()
+            for la in m.layers:
+                la.apply_grads()
+        W0 = m.layers[0].fc.W.clone()
+        gathered = [t.zeros_like(W0) for _ in range(WORLD)]
+        dist.all_gather(gathered, W0)
+        same = all(t.allclose(gathered[0], w, atol=1e-6) for w in gathered)
+        if rank == 0:
+            q.put(("result", bool(same)))
+        dist.destroy_process_group()
+    except Exception:  # pragma: no cover
+        import traceback
+
+        q.put(("error", rank, traceback.format_exc()))
+        raise
+
+
+def test_ring_cnn_replicas_stay_synced():
+    """Reference WORKER_RING CNN mode (train_cnn_algo.h) on this engine."""
+    (ok,) = _run_spawn(_ring_cnn_worker, 29535)
+    assert ok
