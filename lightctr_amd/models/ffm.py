@@ -60,7 +60,7 @@ class FFMModel:
         self.uniq = torch.zeros(cap, dtype=torch.int32, device=self.device)
         self.count = torch.zeros(1, dtype=torch.int32, device=self.device)
         self._use_hip = self.device.type == "cuda"
-        self.backward_mode = "sorted"  # sorted (default) | atomic
+        self.backward_mode = "blocks"  # blocks (default) | sorted | atomic
         if self._use_hip:
             require_hip_ops()
 
@@ -94,7 +94,18 @@ class FFMModel:
             ops = require_hip_ops()
             pred = ops.ffm_forward(row_ptr, fields, fids, vals, self.W, self.V)
             loss, dpred = ops.logloss_grad(pred, labels, scale)
-            if self.backward_mode == "sorted":
+            if self.backward_mode == "blocks":
+                row_of_entry = ops.row_index(row_ptr, fids.numel())
+                gw, gblocks = ops.ffm_block_emit(row_of_entry, row_ptr,
+                                                 fields, fids, vals, self.V,
+                                                 dpred)
+                sorted_fids, perm = torch.sort(fids)
+                ops.ffm_blocks_apply(sorted_fids, perm, gblocks, gw,
+                                     self.gradW,
+                                     self.gradV.view(self.h.num_features,
+                                                     -1),
+                                     self.touched)
+            elif self.backward_mode == "sorted":
                 sorted_fids, perm = torch.sort(fids)
                 row_of_entry = ops.row_index(row_ptr, fids.numel())
                 ops.ffm_sorted_backward(sorted_fids, perm, row_of_entry,

@@ -198,6 +198,40 @@ void ffm_sorted_backward(at::Tensor sorted_fids, at::Tensor perm,
       nfields, (int)sorted_fids.numel(), K, cur_stream());
 }
 
+std::vector<at::Tensor> ffm_block_emit(at::Tensor row_of_entry,
+                                       at::Tensor row_ptr, at::Tensor fields,
+                                       at::Tensor fids, at::Tensor vals,
+                                       at::Tensor V, at::Tensor dpred) {
+  check_cuda_i32(row_of_entry, "row_of_entry");
+  check_cuda_f32(V, "V");
+  const int nfields = (int)V.size(1);
+  const int K = (int)V.size(2);
+  const auto nnz = fids.numel();
+  auto gblocks = at::empty({nnz, (long)nfields * K}, V.options());
+  auto gw = at::empty({nnz}, V.options());
+  lightctr::ffm_block_emit_launch(
+      row_of_entry.data_ptr<int>(), row_ptr.data_ptr<int>(),
+      fields.data_ptr<int>(), fids.data_ptr<int>(), vals.data_ptr<float>(),
+      V.data_ptr<float>(), dpred.data_ptr<float>(),
+      gblocks.data_ptr<float>(), gw.data_ptr<float>(), nfields, (int)nnz, K,
+      cur_stream());
+  return {gw, gblocks};
+}
+
+void ffm_blocks_apply(at::Tensor sorted_fids, at::Tensor perm,
+                      at::Tensor gblocks, at::Tensor gw, at::Tensor gradW,
+                      at::Tensor gradV, at::Tensor touched) {
+  check_cuda_i32(sorted_fids, "sorted_fids");
+  CHK(perm.scalar_type() == at::kLong, "perm must be int64");
+  const int D = (int)gblocks.size(1);
+  lightctr::ffm_blocks_apply_launch(
+      sorted_fids.data_ptr<int>(), perm.data_ptr<long>(),
+      gblocks.data_ptr<float>(), gw.data_ptr<float>(),
+      gradW.data_ptr<float>(), gradV.data_ptr<float>(),
+      (unsigned long long*)touched.data_ptr(), D,
+      (int)sorted_fids.numel(), cur_stream());
+}
+
 at::Tensor row_index(at::Tensor row_ptr, int64_t nnz) {
   check_cuda_i32(row_ptr, "row_ptr");
   auto out = at::empty({nnz}, row_ptr.options());
@@ -606,6 +640,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ffm_sorted_backward", &ffm_sorted_backward,
         "FFM sorted segment-reduce backward (LDS block accumulate)");
   m.def("row_index", &row_index, "entry -> row index from row_ptr");
+  m.def("ffm_block_emit", &ffm_block_emit,
+        "FFM per-entry [nf,K] gradient blocks (wave/entry)");
+  m.def("ffm_blocks_apply", &ffm_blocks_apply,
+        "segment-reduce sorted FFM blocks into slabs");
   m.def("parse_libffm", &parse_libffm, "native libffm text parser",
         py::arg("path"), py::arg("max_rows") = -1);
   m.def("sparse_adagrad_apply", &sparse_adagrad_apply,

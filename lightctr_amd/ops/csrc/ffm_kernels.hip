@@ -192,6 +192,130 @@ __global__ void ffm_sorted_backward_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Block-emit FFM backward, phase 1: ONE WAVE PER ENTRY (maximal
+// parallelism — no serial chunk walk), computing the entry's full
+// [nfields, K] gradient block in its private LDS slice and writing it
+// coalesced to gblocks[p*D..]. Phase 2 (ffm_blocks_apply_kernel) walks the
+// fid-sorted order gathering CONTIGUOUS blocks (sequential 1.2 KB reads,
+// unlike the recompute path's dependent random gathers) and
+// segment-reduces into the slabs.
+// ---------------------------------------------------------------------------
+template <int K>
+__global__ void ffm_block_emit_kernel(
+    const int* __restrict__ row_of_entry, const int* __restrict__ row_ptr,
+    const int* __restrict__ fields, const int* __restrict__ fids,
+    const float* __restrict__ vals, const float* __restrict__ V,
+    const float* __restrict__ dpred, float* __restrict__ gblocks,
+    float* __restrict__ gw, int nfields, int nnz) {
+  extern __shared__ float lds_acc[];
+  constexpr int G = LCTR_WAVE / K;
+  const int lane = threadIdx.x & 63;
+  const int wave_in_blk = threadIdx.x >> 6;
+  const int p = blockIdx.x * (blockDim.x >> 6) + wave_in_blk;
+  if (p >= nnz) return;
+  const int jg = lane / K;
+  const int k = lane % K;
+  const int fstride = K + 1;
+  float* acc = &lds_acc[wave_in_blk * nfields * fstride];
+  for (int i = lane; i < nfields * fstride; i += LCTR_WAVE) acc[i] = 0.f;
+
+  const int r = row_of_entry[p];
+  const int Fi = fields[p];
+  const float xi = vals[p];
+  const float d = dpred[r];
+  const int beg = row_ptr[r], rend = row_ptr[r + 1];
+  for (int j = beg + jg; j < rend; j += G) {
+    if (j == p) continue;
+    const float v = V[((size_t)fids[j] * nfields + Fi) * K + k];
+    atomicAdd(&acc[fields[j] * fstride + k], d * xi * vals[j] * v);
+  }
+  const int D = nfields * K;
+  for (int i = lane; i < D; i += LCTR_WAVE)
+    gblocks[(size_t)p * D + i] = acc[(i / K) * fstride + (i % K)];
+  if (lane == 0) gw[p] = d * xi;
+}
+
+// Phase 2: segment-reduce fid-sorted blocks into the grad slabs.
+__global__ void ffm_blocks_apply_kernel(
+    const int* __restrict__ sorted_fids, const long* __restrict__ perm,
+    const float* __restrict__ gblocks, const float* __restrict__ gw,
+    float* __restrict__ gradW, float* __restrict__ gradV,
+    unsigned long long* __restrict__ touched, int D, int nnz, int chunk) {
+  extern __shared__ float lds_acc[];
+  const int lane = threadIdx.x & 63;
+  const int wave_in_blk = threadIdx.x >> 6;
+  const int wave = blockIdx.x * (blockDim.x >> 6) + wave_in_blk;
+  const int base = wave * chunk;
+  if (base >= nnz) return;
+  const int end = min(base + chunk, nnz);
+  float* acc = &lds_acc[wave_in_blk * D];
+  for (int i = lane; i < D; i += LCTR_WAVE) acc[i] = 0.f;
+
+  int cur = -1;
+  float accw = 0.f;
+  for (int e = base; e < end; ++e) {
+    const int fid = sorted_fids[e];
+    if (fid != cur) {
+      if (cur >= 0) {
+        for (int i = lane; i < D; i += LCTR_WAVE) {
+          if (acc[i] != 0.f) {
+            atomicAdd(&gradV[(size_t)cur * D + i], acc[i]);
+            acc[i] = 0.f;
+          }
+        }
+        if (lane == 0) atomicAdd(&gradW[cur], accw);
+      }
+      cur = fid;
+      accw = 0.f;
+      if (lane == 0 && (e == 0 || sorted_fids[e - 1] != fid)) {
+        atomicOr(&touched[fid >> 6], 1ull << (fid & 63));
+      }
+    }
+    const long p = perm[e];
+    for (int i = lane; i < D; i += LCTR_WAVE)
+      acc[i] += gblocks[(size_t)p * D + i];
+    if (lane == 0) accw += gw[p];
+  }
+  if (cur >= 0) {
+    for (int i = lane; i < D; i += LCTR_WAVE) {
+      if (acc[i] != 0.f) atomicAdd(&gradV[(size_t)cur * D + i], acc[i]);
+    }
+    if (lane == 0) atomicAdd(&gradW[cur], accw);
+  }
+}
+
+void ffm_block_emit_launch(const int* row_of_entry, const int* row_ptr,
+                           const int* fields, const int* fids,
+                           const float* vals, const float* V,
+                           const float* dpred, float* gblocks, float* gw,
+                           int nfields, int nnz, int K, hipStream_t stream) {
+  const int wpb = 4;
+  dim3 block(wpb * LCTR_WAVE);
+  dim3 grid((nnz + wpb - 1) / wpb);
+  const size_t lds = (size_t)wpb * nfields * (K + 1) * sizeof(float);
+  DISPATCH_FFM_K(K, hipLaunchKernelGGL((ffm_block_emit_kernel<KC>), grid,
+                                       block, lds, stream, row_of_entry,
+                                       row_ptr, fields, fids, vals, V, dpred,
+                                       gblocks, gw, nfields, nnz));
+}
+
+void ffm_blocks_apply_launch(const int* sorted_fids, const long* perm,
+                             const float* gblocks, const float* gw,
+                             float* gradW, float* gradV,
+                             unsigned long long* touched, int D, int nnz,
+                             hipStream_t stream) {
+  const int chunk = 64;
+  const int wpb = 4;
+  const int nwaves = (nnz + chunk - 1) / chunk;
+  dim3 block(wpb * LCTR_WAVE);
+  dim3 grid((nwaves + wpb - 1) / wpb);
+  const size_t lds = (size_t)wpb * D * sizeof(float);
+  hipLaunchKernelGGL(ffm_blocks_apply_kernel, grid, block, lds, stream,
+                     sorted_fids, perm, gblocks, gw, gradW, gradV, touched,
+                     D, nnz, chunk);
+}
+
 // entry -> row index materialization (one wave per row)
 __global__ void row_index_kernel(const int* __restrict__ row_ptr,
                                  int* __restrict__ row_idx, int B) {

@@ -54,6 +54,16 @@ void ffm_sorted_backward_launch(const int* sorted_fids, const long* perm,
                                 ihipStream_t* stream);
 void row_index_launch(const int* row_ptr, int* row_idx, int B,
                       ihipStream_t* stream);
+void ffm_block_emit_launch(const int* row_of_entry, const int* row_ptr,
+                           const int* fields, const int* fids,
+                           const float* vals, const float* V,
+                           const float* dpred, float* gblocks, float* gw,
+                           int nfields, int nnz, int K, ihipStream_t* stream);
+void ffm_blocks_apply_launch(const int* sorted_fids, const long* perm,
+                             const float* gblocks, const float* gw,
+                             float* gradW, float* gradV,
+                             unsigned long long* touched, int D, int nnz,
+                             ihipStream_t* stream);
 
 // --- misc_kernels.hip (generic sparse fused optimizers; D = per-feature
 // latent size, runtime) ---

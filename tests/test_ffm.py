@@ -190,3 +190,36 @@ def test_ffm_sorted_backward_parity():
         (gradW - gW_ref).abs().max()
     assert torch.allclose(gradV, gV_ref, atol=1e-5, rtol=1e-4), \
         (gradV - gV_ref).abs().max()
+
+
+@pytest.mark.gpu
+def test_ffm_blocks_backward_parity():
+    """blocks-mode (emit + sorted block reduce) == dense reference."""
+    from lightctr_amd.ops import hip_ops
+
+    row_ptr, fids, vals, labels = make_random_csr(B=96, F_total=3000,
+                                                  min_f=2, max_f=25, seed=19,
+                                                  device="cuda:0",
+                                                  binary_vals=False)
+    nf, K = 6, 8
+    fields = (fids.long() % nf).int()
+    g = torch.Generator().manual_seed(8)
+    W = torch.randn(3000, generator=g).cuda()
+    V = (torch.randn(3000, nf, K, generator=g) * 0.1).cuda()
+    pred = hip_ops.ffm_forward(row_ptr, fields, fids, vals, W, V)
+    _, dpred = hip_ops.logloss_grad(pred, labels, 1.0 / 96)
+    row_of_entry = hip_ops.row_index(row_ptr, fids.numel())
+    gw, gblocks = hip_ops.ffm_block_emit(row_of_entry, row_ptr, fields,
+                                         fids, vals, V, dpred)
+    sorted_fids, perm = torch.sort(fids)
+    gradW = torch.zeros(3000).cuda()
+    gradV = torch.zeros(3000, nf, K).cuda()
+    touched = torch.zeros((3000 + 63) // 64, dtype=torch.int64).cuda()
+    hip_ops.ffm_blocks_apply(sorted_fids, perm, gblocks, gw, gradW,
+                             gradV.view(3000, -1), touched)
+    gW_ref, gV_ref = ffm_ref.ffm_backward_ref(row_ptr, fields, fids, vals,
+                                              V, dpred)
+    assert torch.allclose(gradW, gW_ref, atol=1e-5, rtol=1e-4), \
+        (gradW - gW_ref).abs().max()
+    assert torch.allclose(gradV, gV_ref, atol=1e-5, rtol=1e-4), \
+        (gradV - gV_ref).abs().max()
