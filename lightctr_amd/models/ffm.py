@@ -60,7 +60,13 @@ class FFMModel:
         self.uniq = torch.zeros(cap, dtype=torch.int32, device=self.device)
         self.count = torch.zeros(1, dtype=torch.int32, device=self.device)
         self._use_hip = self.device.type == "cuda"
-        self.backward_mode = "blocks"  # blocks (default) | sorted | atomic
+        # backward variants, all parity-tested (tests/test_ffm.py):
+        #   sorted (default) — per-run LDS-block recompute, 4.0 ms
+        #   blocks — wave/entry emit + coalesced block reduce, 4.4+1.1 ms
+        #            (measured slower: the bound is the per-(pair,k)
+        #            instruction stream, not scheduling; kept as evidence)
+        #   atomic — naive scatter (hot-feature serialization)
+        self.backward_mode = "sorted"
         if self._use_hip:
             require_hip_ops()
 
