@@ -134,6 +134,7 @@ __global__ void nfm_backward_emit_kernel(
 void embed_gather_launch(const int* row_ptr, const int* fids,
                          const float* vals, const float* E, void* out_bf,
                          int nf, int B, int K, hipStream_t stream) {
+  if (B <= 0) return;
   dim3 block(256);
   dim3 grid((B + 3) / 4);
   DISPATCH_K(K, hipLaunchKernelGGL((embed_gather_kernel<KC>), grid, block, 0,
@@ -145,6 +146,7 @@ void embed_backward_emit_launch(const int* row_ptr, const float* vals,
                                 const float* dOut, const float* dwide,
                                 float* gv, float* gw, int nf, int B, int K,
                                 hipStream_t stream) {
+  if (B <= 0) return;
   dim3 block(256);
   dim3 grid((B + 3) / 4);
   DISPATCH_K(K, hipLaunchKernelGGL((embed_backward_emit_kernel<KC>), grid,
@@ -171,6 +173,7 @@ __global__ void wide_forward_kernel(const int* __restrict__ row_ptr,
 void wide_forward_launch(const int* row_ptr, const int* fids,
                          const float* vals, const float* W, float* wide,
                          int B, hipStream_t stream) {
+  if (B <= 0) return;
   dim3 block(256);
   dim3 grid((B + 3) / 4);
   hipLaunchKernelGGL(wide_forward_kernel, grid, block, 0, stream, row_ptr,
@@ -181,6 +184,7 @@ void nfm_forward_launch(const int* row_ptr, const int* fids, const float* vals,
                         const float* W, const float* V, float* wide,
                         float* sumVX, float* vec, void* vec_bf, int B, int K,
                         hipStream_t stream) {
+  if (B <= 0) return;
   dim3 block(256);
   dim3 grid((B + 3) / 4);
   DISPATCH_K(K, hipLaunchKernelGGL((nfm_forward_kernel<KC>), grid, block, 0,
@@ -193,6 +197,7 @@ void nfm_backward_emit_launch(const int* row_ptr, const int* fids,
                               const float* sumVX, const float* dvec,
                               const float* dwide, float* gw, float* gv, int B,
                               int K, hipStream_t stream) {
+  if (B <= 0) return;
   dim3 block(256);
   dim3 grid((B + 3) / 4);
   DISPATCH_K(K, hipLaunchKernelGGL((nfm_backward_emit_kernel<KC>), grid,

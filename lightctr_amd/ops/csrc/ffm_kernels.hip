@@ -290,6 +290,7 @@ void ffm_block_emit_launch(const int* row_of_entry, const int* row_ptr,
                            const float* vals, const float* V,
                            const float* dpred, float* gblocks, float* gw,
                            int nfields, int nnz, int K, hipStream_t stream) {
+  if (nnz <= 0) return;
   const int wpb = 4;
   dim3 block(wpb * LCTR_WAVE);
   dim3 grid((nnz + wpb - 1) / wpb);
@@ -305,6 +306,7 @@ void ffm_blocks_apply_launch(const int* sorted_fids, const long* perm,
                              float* gradW, float* gradV,
                              unsigned long long* touched, int D, int nnz,
                              hipStream_t stream) {
+  if (nnz <= 0) return;
   const int chunk = 64;
   const int wpb = 4;
   const int nwaves = (nnz + chunk - 1) / chunk;
@@ -334,6 +336,7 @@ void ffm_sorted_backward_launch(const int* sorted_fids, const long* perm,
                                 float* gradV, unsigned long long* touched,
                                 int nfields, int nnz, int K,
                                 hipStream_t stream) {
+  if (nnz <= 0) return;
   const int chunk = 32;  // short chunks: more waves in flight to hide the
                          // per-entry dependent V-gather latency
   const int wpb = 4;
@@ -350,6 +353,7 @@ void ffm_sorted_backward_launch(const int* sorted_fids, const long* perm,
 
 void row_index_launch(const int* row_ptr, int* row_idx, int B,
                       hipStream_t stream) {
+  if (B <= 0) return;
   dim3 block(256);
   dim3 grid((B + 3) / 4);
   hipLaunchKernelGGL(row_index_kernel, grid, block, 0, stream, row_ptr,
@@ -360,6 +364,7 @@ void ffm_forward_launch(const int* row_ptr, const int* fields, const int* fids,
                         const float* vals, const float* W, const float* V,
                         float* pred, int nfields, int B, int K,
                         hipStream_t stream) {
+  if (B <= 0) return;
   dim3 block(256);
   dim3 grid((B + 3) / 4);
   DISPATCH_FFM_K(K, hipLaunchKernelGGL((ffm_forward_kernel<KC>), grid, block,
@@ -372,6 +377,7 @@ void ffm_backward_launch(const int* row_ptr, const int* fields,
                          const float* dpred, float* gradW, float* gradV,
                          unsigned long long* touched, int nfields, int B,
                          int K, hipStream_t stream) {
+  if (B <= 0) return;
   dim3 block(256);
   dim3 grid((B + 3) / 4);
   DISPATCH_FFM_K(K, hipLaunchKernelGGL((ffm_backward_kernel<KC>), grid, block,

@@ -381,6 +381,7 @@ static inline int waves_per_block() { return 4; }  // 256 threads
 void fm_forward_launch(const int* row_ptr, const int* fids, const float* vals,
                        const float* W, const float* V, float* pred,
                        float* sumVX, int B, int K, hipStream_t stream) {
+  if (B <= 0) return;
   const int wpb = waves_per_block();
   dim3 block(wpb * LCTR_WAVE);
   dim3 grid((B + wpb - 1) / wpb);
@@ -392,6 +393,7 @@ void fm_forward_launch(const int* row_ptr, const int* fids, const float* vals,
 void logloss_grad_launch(const float* pred, const float* label, float* loss,
                          float* dpred, float scale, int B,
                          hipStream_t stream) {
+  if (B <= 0) return;
   dim3 block(256);
   dim3 grid((B + 255) / 256);
   hipLaunchKernelGGL(logloss_grad_kernel, grid, block, 0, stream, pred, label,
@@ -402,6 +404,7 @@ void fm_backward_launch(const int* row_ptr, const int* fids, const float* vals,
                         const float* V, const float* sumVX, const float* dpred,
                         float* gradW, float* gradV, unsigned long long* touched,
                         int B, int K, hipStream_t stream) {
+  if (B <= 0) return;
   const int wpb = waves_per_block();
   dim3 block(wpb * LCTR_WAVE);
   dim3 grid((B + wpb - 1) / wpb);
@@ -414,6 +417,7 @@ void fm_backward_emit_launch(const int* row_ptr, const int* fids,
                              const float* vals, const float* V,
                              const float* sumVX, const float* dpred, float* gw,
                              float* gv, int B, int K, hipStream_t stream) {
+  if (B <= 0) return;
   const int wpb = waves_per_block();
   dim3 block(wpb * LCTR_WAVE);
   dim3 grid((B + wpb - 1) / wpb);
@@ -429,6 +433,7 @@ void fm_sorted_apply_launch(const int* sorted_fids, const long* perm,
                             float* nW, float* zW, float* nV, float* zV,
                             float p0, float p1, float p2, float p3,
                             hipStream_t stream) {
+  if (nnz <= 0) return;
   const int chunk = 256;
   const int wpb = waves_per_block();
   const int nwaves = (nnz + chunk - 1) / chunk;

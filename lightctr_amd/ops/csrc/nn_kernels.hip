@@ -123,6 +123,7 @@ __global__ void dense_adagrad_kernel(float* __restrict__ W,
 
 void act_backward_launch(const float* dY, const float* Y, float* dZ,
                          void* dZbf, long n, int act, hipStream_t stream) {
+  if (n <= 0) return;
   dim3 block(256);
   dim3 grid((unsigned)((n + 255) / 256));
   hipLaunchKernelGGL(act_backward_kernel, grid, block, 0, stream, dY, Y, dZ,
@@ -146,6 +147,7 @@ void colsum_launch(const float* dZ, float* db, int M, int N,
 }
 
 void to_bf16_launch(const float* x, void* y, long n, hipStream_t stream) {
+  if (n <= 0) return;
   dim3 block(256);
   dim3 grid((unsigned)((n / 4 + 256) / 256) + 1);
   hipLaunchKernelGGL(to_bf16_kernel, grid, block, 0, stream, x, (__bf16*)y,

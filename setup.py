@@ -19,8 +19,10 @@ CSRC = os.path.join(os.path.dirname(os.path.abspath(__file__)),
 
 import glob
 
+# exclude torch-hipify's generated *_hip.* copies (build artifacts)
 sources = [os.path.join(CSRC, "bindings.cpp")] + sorted(
-    glob.glob(os.path.join(CSRC, "*.hip")))
+    p for p in glob.glob(os.path.join(CSRC, "*.hip"))
+    if not p.endswith("_hip.hip"))
 
 ext = CUDAExtension(
     name="lightctr_amd.ops._hip_ops",
