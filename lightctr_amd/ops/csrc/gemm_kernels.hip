@@ -91,18 +91,32 @@ __device__ __forceinline__ void stage_scalar(const __bf16* __restrict__ g,
       dst[r * 64 + swz(r, ke)] = v;
     }
   } else {
-    // stored [K, rows]: thread t covers k=t/8 (+32), rowpart=(t%8)*16
+    // stored [K, rows]: thread t covers k=t/8 (+32), rowpart=(t%8)*16;
+    // fast path loads the 16 contiguous elements as two bf16x8 vectors
+    // (the wgrad operands are large and almost always fully in-bounds)
+    typedef __attribute__((ext_vector_type(8))) __bf16 v8;
     const int kk = tid >> 3;
     const int rp = (tid & 7) * 16;
     for (int kr = 0; kr < 2; ++kr) {
       const int ke = kk + kr * 32;
       const int gk = k0 + ke;
-      for (int e = 0; e < 16; ++e) {
-        const int r = rp + e;
-        const int grow = base_row + r;
-        __bf16 v = (__bf16)0.f;
-        if (gk < K && grow < max_row) v = g[(long)gk * stride + grow];
-        dst[r * 64 + swz(r, ke)] = v;
+      if (gk < K && base_row + rp + 15 < max_row &&
+          (((uintptr_t)&g[(long)gk * stride + base_row + rp]) & 15) == 0) {
+        const v8 lo = *(const v8*)&g[(long)gk * stride + base_row + rp];
+        const v8 hi = *(const v8*)&g[(long)gk * stride + base_row + rp + 8];
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          dst[(rp + e) * 64 + swz(rp + e, ke)] = lo[e];
+          dst[(rp + 8 + e) * 64 + swz(rp + 8 + e, ke)] = hi[e];
+        }
+      } else {
+        for (int e = 0; e < 16; ++e) {
+          const int r = rp + e;
+          const int grow = base_row + r;
+          __bf16 v = (__bf16)0.f;
+          if (gk < K && grow < max_row) v = g[(long)gk * stride + grow];
+          dst[r * 64 + swz(r, ke)] = v;
+        }
       }
     }
   }
