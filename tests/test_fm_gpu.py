@@ -193,3 +193,45 @@ def test_fused_apply_matches_two_phase():
         assert torch.allclose(a.V, b.V, atol=1e-5), \
             (a.V - b.V).abs().max()
         assert torch.allclose(a.nV, b.nV, atol=1e-5)
+
+
+def test_sorted_apply_single_hot_feature():
+    """Extreme skew: every entry is the SAME fid (one segment spanning all
+    chunks/subgroups) — partial flushes must sum exactly."""
+    from lightctr_amd.ops import hip_ops
+
+    B, F, K = 1024, 100, 16
+    row_ptr = torch.arange(0, (B + 1) * 4, 4, dtype=torch.int32).cuda()
+    fids = torch.full((B * 4,), 7, dtype=torch.int32).cuda()
+    g = torch.Generator().manual_seed(3)
+    vals = torch.rand(B * 4, generator=g).cuda()
+    labels = torch.randint(0, 2, (B,), generator=g).float().cuda()
+    W = torch.randn(F, generator=g).cuda()
+    V = (torch.randn(F, K, generator=g) * 0.1).cuda()
+    pred, sumVX = hip_ops.fm_forward(row_ptr, fids, vals, W, V)
+    _, dpred = hip_ops.logloss_grad(pred, labels, 1.0 / B)
+    gw, gv = hip_ops.fm_backward_emit(row_ptr, fids, vals, V, sumVX, dpred)
+    sorted_fids, perm = torch.sort(fids)
+    gradW = torch.zeros(F).cuda()
+    gradV = torch.zeros(F, K).cuda()
+    touched = torch.zeros((F + 63) // 64, dtype=torch.int64).cuda()
+    hip_ops.fm_sorted_apply(sorted_fids, perm, gw, gv, gradW, gradV, touched)
+    gW_ref, gV_ref = fm_ref.fm_backward_ref(row_ptr, fids, vals, V, sumVX,
+                                            dpred)
+    assert torch.allclose(gradW, gW_ref, atol=1e-3, rtol=1e-4)
+    assert torch.allclose(gradV, gV_ref, atol=1e-3, rtol=1e-4)
+    assert int(touched.cpu()[0]) == 1 << 7  # only fid 7 marked
+
+
+def test_fm_forward_k64():
+    from lightctr_amd.ops import hip_ops
+
+    row_ptr, fids, vals, _ = _csr(B=64, seed=64)
+    F, K = 50_000, 64
+    g = torch.Generator().manual_seed(64)
+    W = torch.randn(F, generator=g).cuda()
+    V = (torch.randn(F, K, generator=g) * 0.1).cuda()
+    pred, sumVX = hip_ops.fm_forward(row_ptr, fids, vals, W, V)
+    pred_ref, sumVX_ref = fm_ref.fm_forward_ref(row_ptr, fids, vals, W, V)
+    assert torch.allclose(sumVX, sumVX_ref, atol=1e-4, rtol=1e-4)
+    assert torch.allclose(pred, pred_ref, atol=2e-3, rtol=1e-4)

@@ -212,3 +212,19 @@ def test_nfm_gpu_forward_parity():
     assert torch.allclose(sumVX, sumVX_ref, atol=1e-4)
     # wide + vec.sum() must equal the FM prediction
     assert torch.allclose(wide + vec.sum(dim=1), pred_ref, atol=1e-3)
+
+
+@pytest.mark.gpu
+def test_gemm_sigmoid_epilogue_and_256path_bf16_emit():
+    from lightctr_amd.ops import hip_ops
+
+    M, N, K = 512, 256, 128  # routes through the 256^2 kernel
+    g = torch.Generator().manual_seed(21)
+    A = (torch.randn(M, K, generator=g) * 0.3).to(torch.bfloat16).cuda()
+    Bst = (torch.randn(N, K, generator=g) * 0.3).to(torch.bfloat16).cuda()
+    bias = torch.randn(N, generator=g).cuda()
+    C, Cbf = hip_ops.gemm_bf16_full(A, Bst, bias, M, N, K, 0, 0, 2)
+    ref = torch.sigmoid((A.float() @ Bst.float().t() + bias)
+                        .clamp(-16, 16))
+    assert torch.allclose(C, ref, atol=0.02, rtol=1e-2)
+    assert torch.allclose(Cbf.float(), C, atol=0.01, rtol=1e-2)
