@@ -52,11 +52,14 @@ def build_model(args, device, world):
         from lightctr_amd.models.wide_deep import (WideDeepHyper,
                                                    WideDeepModel)
 
-        assert world == 1
-        return WideDeepModel(
-            WideDeepHyper(num_features=args.features, num_fields=39,
-                          k=args.k, hidden=(256, 128), seed=1234),
-            device=device), "sparse"
+        hyper = WideDeepHyper(num_features=args.features, num_fields=39,
+                              k=args.k, hidden=(256, 128), seed=1234)
+        if world > 1:
+            from lightctr_amd.parallel.sharded_widedeep import (
+                ShardedWideDeepModel)
+
+            return ShardedWideDeepModel(hyper, device=device), "sparse"
+        return WideDeepModel(hyper, device=device), "sparse"
     raise SystemExit(f"unknown model {args.model}")
 
 

@@ -297,3 +297,78 @@ def test_ring_cnn_replicas_stay_synced():
     """Reference WORKER_RING CNN mode (train_cnn_algo.h) on this engine."""
     (ok,) = _run_spawn(_ring_cnn_worker, 29535)
     assert ok
+
+
+def _sharded_wd_worker(rank, port, q):
+    try:
+        dist = _init(rank, port)
+        import torch as t
+        from lightctr_amd.models.wide_deep import (WideDeepHyper,
+                                                   WideDeepModel)
+        from lightctr_amd.parallel.sharded_widedeep import (
+            ShardedWideDeepModel)
+        from conftest import make_random_csr
+
+        F, K, nf = 1200, 8, 10
+        h = WideDeepHyper(num_features=F, num_fields=nf, k=K,
+                          optimizer="adagrad", hidden=(16,), seed=21,
+                          mlp_optimizer="adagrad", mlp_lr=0.01)
+        sharded = ShardedWideDeepModel(h, device="cpu")
+        batches = [make_random_csr(B=32, F_total=F, min_f=nf, max_f=nf,
+                                   seed=50 + 3 * s + rank,
+                                   binary_vals=False)
+                   for s in range(2)]
+        for rp, fi, v, lb in batches:
+            loss = sharded.train_step(rp, fi, v, lb)
+            assert t.isfinite(loss).all()
+        gathered_W = [t.zeros_like(sharded.W) for _ in range(WORLD)]
+        gathered_E = [t.zeros_like(sharded.E) for _ in range(WORLD)]
+        dist.all_gather(gathered_W, sharded.W)
+        dist.all_gather(gathered_E, sharded.E)
+        if rank == 0:
+            W_full = t.zeros(F)
+            E_full = t.zeros(F, K)
+            for r in range(WORLD):
+                idx = t.arange(r, F, WORLD)
+                W_full[idx] = gathered_W[r][: idx.numel()]
+                E_full[idx] = gathered_E[r][: idx.numel()]
+            single = WideDeepModel(h, device="cpu")
+            for r in range(WORLD):
+                g = t.Generator().manual_seed(h.seed + 17 * r)
+                Fl = (F + WORLD - 1) // WORLD
+                Er = t.randn(Fl, K, generator=g) * h.init_sigma
+                idx = t.arange(r, F, WORLD)
+                single.E[idx] = Er[: idx.numel()]
+            for s in range(2):
+                bs = [make_random_csr(B=32, F_total=F, min_f=nf, max_f=nf,
+                                      seed=50 + 3 * s + r,
+                                      binary_vals=False)
+                      for r in range(WORLD)]
+                rp, fi, v, lb = bs[0]
+                for r in range(1, WORLD):
+                    rp = t.cat([rp[:-1], bs[r][0] + rp[-1]])
+                    fi = t.cat([fi, bs[r][1]])
+                    v = t.cat([v, bs[r][2]])
+                    lb = t.cat([lb, bs[r][3]])
+                single.train_step(rp, fi, v, lb)
+            ok_e = t.allclose(E_full, single.E, atol=1e-5)
+            ok_w = t.allclose(W_full, single.W, atol=1e-5)
+            ok_mlp = t.allclose(sharded.mlp.layers[0].W,
+                                single.mlp.layers[0].W, atol=1e-5)
+            q.put(("result", bool(ok_e and ok_w and ok_mlp),
+                   float((E_full - single.E).abs().max()),
+                   float((sharded.mlp.layers[0].W
+                          - single.mlp.layers[0].W).abs().max())))
+        dist.destroy_process_group()
+    except Exception:  # pragma: no cover
+        import traceback
+
+        q.put(("error", rank, traceback.format_exc()))
+        raise
+
+
+def test_sharded_widedeep_matches_single():
+    """Sharded embedding + SUM-allreduced replicated MLP == single model
+    on the union batch (2 steps, adagrad both sides)."""
+    ok, de, dmlp = _run_spawn(_sharded_wd_worker, 29536)
+    assert ok, f"maxdiff E={de} mlp={dmlp}"
