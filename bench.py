@@ -37,10 +37,15 @@ def build_model(args, device, world):
     if args.model == "ffm":
         from lightctr_amd.models.ffm import FFMHyper, FFMModel
 
-        assert world == 1, "ffm bench is single-GPU in this revision"
-        return FFMModel(FFMHyper(num_features=args.features, num_fields=39,
-                                 k=min(args.k, 8), optimizer=args.optimizer,
-                                 seed=1234), device=device), "fieldaware"
+        hyper = FFMHyper(num_features=args.features, num_fields=39,
+                         k=min(args.k, 8), optimizer=args.optimizer,
+                         seed=1234)
+        if world > 1:
+            from lightctr_amd.parallel.sharded_ffm import ShardedFFMModel
+
+            return (ShardedFFMModel(hyper, device=device, wire="fp16"),
+                    "fieldaware")
+        return FFMModel(hyper, device=device), "fieldaware"
     if args.model == "nfm":
         from lightctr_amd.models.nfm import NFMHyper, NFMModel
 

@@ -372,3 +372,69 @@ def test_sharded_widedeep_matches_single():
     on the union batch (2 steps, adagrad both sides)."""
     ok, de, dmlp = _run_spawn(_sharded_wd_worker, 29536)
     assert ok, f"maxdiff E={de} mlp={dmlp}"
+
+
+def _sharded_ffm_worker(rank, port, q):
+    try:
+        dist = _init(rank, port)
+        import torch as t
+        from lightctr_amd.models.ffm import FFMHyper, FFMModel
+        from lightctr_amd.parallel.sharded_ffm import ShardedFFMModel
+        from conftest import make_random_csr
+
+        F, nf, K = 800, 6, 4
+        h = FFMHyper(num_features=F, num_fields=nf, k=K,
+                     optimizer="adagrad", seed=33)
+        sharded = ShardedFFMModel(h, device="cpu")
+        batches = []
+        for r in range(WORLD):
+            rp, fi, v, lb = make_random_csr(B=24, F_total=F, min_f=3,
+                                            max_f=nf, seed=90 + r,
+                                            binary_vals=False)
+            fl = (fi.long() % nf).int()
+            batches.append((rp, fl, fi, v, lb))
+        rp, fl, fi, v, lb = batches[rank]
+        loss = sharded.train_step(rp, fl, fi, v, lb)
+        assert t.isfinite(loss).all()
+        gathered_W = [t.zeros_like(sharded.W) for _ in range(WORLD)]
+        gathered_V = [t.zeros_like(sharded.V) for _ in range(WORLD)]
+        dist.all_gather(gathered_W, sharded.W)
+        dist.all_gather(gathered_V, sharded.V)
+        if rank == 0:
+            W_full = t.zeros(F)
+            V_full = t.zeros(F, nf, K)
+            for r in range(WORLD):
+                idx = t.arange(r, F, WORLD)
+                W_full[idx] = gathered_W[r][: idx.numel()]
+                V_full[idx] = gathered_V[r][: idx.numel()]
+            single = FFMModel(h, device="cpu")
+            for r in range(WORLD):
+                g = t.Generator().manual_seed(h.seed + 17 * r)
+                Fl = (F + WORLD - 1) // WORLD
+                Vr = t.randn(Fl, nf, K, generator=g) * h.init_sigma
+                idx = t.arange(r, F, WORLD)
+                single.V[idx] = Vr[: idx.numel()]
+            rp, fl, fi, v, lb = batches[0]
+            for r in range(1, WORLD):
+                rp2, fl2, fi2, v2, lb2 = batches[r]
+                rp = t.cat([rp[:-1], rp2 + rp[-1]])
+                fl = t.cat([fl, fl2])
+                fi = t.cat([fi, fi2])
+                v = t.cat([v, v2])
+                lb = t.cat([lb, lb2])
+            single.train_step(rp, fl, fi, v, lb)
+            ok = t.allclose(W_full, single.W, atol=1e-5) and \
+                t.allclose(V_full, single.V, atol=1e-5)
+            q.put(("result", bool(ok),
+                   float((V_full - single.V).abs().max())))
+        dist.destroy_process_group()
+    except Exception:  # pragma: no cover
+        import traceback
+
+        q.put(("error", rank, traceback.format_exc()))
+        raise
+
+
+def test_sharded_ffm_matches_single():
+    ok, dv = _run_spawn(_sharded_ffm_worker, 29537)
+    assert ok, f"V maxdiff {dv}"
