@@ -30,7 +30,7 @@ class NFMHyper:
     mlp_lr: float = 1e-3
     eps: float = 1e-8
     l2: float = 1e-5
-    ftrl_alpha: float = 0.05
+    ftrl_alpha: float = 0.15
     ftrl_beta: float = 1.0
     ftrl_l1: float = 1e-4
     ftrl_l2: float = 1e-4
