@@ -240,6 +240,24 @@ at::Tensor row_index(at::Tensor row_ptr, int64_t nnz) {
   return out;
 }
 
+void ps_apply(at::Tensor lidx, at::Tensor gW, at::Tensor gV, at::Tensor W,
+              at::Tensor V, c10::optional<at::Tensor> nW,
+              c10::optional<at::Tensor> nV,
+              c10::optional<at::Tensor> shadowW,
+              c10::optional<at::Tensor> shadowV, int64_t updater, double lr,
+              double lam, double eps) {
+  CHK(lidx.is_cuda() && lidx.scalar_type() == at::kLong, "lidx i64 GPU");
+  const int K = (int)V.size(1);
+  lightctr::ps_apply_launch(
+      lidx.data_ptr<long>(), (int)lidx.numel(), gW.data_ptr<float>(),
+      gV.data_ptr<float>(), W.data_ptr<float>(), V.data_ptr<float>(),
+      nW.has_value() ? nW->data_ptr<float>() : nullptr,
+      nV.has_value() ? nV->data_ptr<float>() : nullptr,
+      shadowW.has_value() ? shadowW->data_ptr<float>() : nullptr,
+      shadowV.has_value() ? shadowV->data_ptr<float>() : nullptr, K,
+      (int)updater, (float)lr, (float)lam, (float)eps, cur_stream());
+}
+
 // ---- generic sparse optimizers (D = latent block width, runtime) ----
 
 void sparse_adagrad_apply(at::Tensor uniq, at::Tensor count, at::Tensor W,
@@ -648,6 +666,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("path"), py::arg("max_rows") = -1);
   m.def("sparse_adagrad_apply", &sparse_adagrad_apply,
         "generic sparse fused Adagrad (runtime D)");
+  m.def("ps_apply", &ps_apply,
+        "PS-side fused updater (sgd/adagrad/dcasgd/dcasgda)");
   m.def("sparse_ftrl_apply", &sparse_ftrl_apply,
         "generic sparse fused FTRL (runtime D)");
   m.def("gemm_bf16", &gemm_bf16, "MFMA bf16 GEMM (C fp32), fused bias+act");

@@ -67,6 +67,11 @@ void ffm_blocks_apply_launch(const int* sorted_fids, const long* perm,
 
 // --- misc_kernels.hip (generic sparse fused optimizers; D = per-feature
 // latent size, runtime) ---
+void ps_apply_launch(const long* lidx, int n, const float* gW,
+                     const float* gV, float* W, float* V, float* nW,
+                     float* nV, float* shadowW, float* shadowV, int K,
+                     int updater, float lr, float lam, float eps,
+                     ihipStream_t* stream);
 void sparse_adagrad_apply_launch(const int* uniq, const int* count, float* W,
                                  float* V, float* nW, float* nV, float* gradW,
                                  float* gradV, float lr, float eps, float l2,

@@ -77,6 +77,70 @@ __global__ void sparse_ftrl_apply_g_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// PS-side fused updaters (reference paramserver.h:217-306): applied to the
+// shard tables for one push's key list. updater: 0=sgd, 1=adagrad,
+// 2=dcasgd, 3=dcasgda. DCASGD keeps a per-worker shadow copy of each
+// pulled weight and compensates w -= lr*(g + lambda*g^2*(w - shadow));
+// DCASGDA uses lambda/sqrt(eps + EMA(g^2)). One wave per key, lanes
+// stride the K latent params; lane 0 handles the scalar W.
+// ---------------------------------------------------------------------------
+__global__ void ps_apply_kernel(const long* __restrict__ lidx, int n,
+                                const float* __restrict__ gW,
+                                const float* __restrict__ gV,
+                                float* __restrict__ W, float* __restrict__ V,
+                                float* __restrict__ nW,
+                                float* __restrict__ nV,
+                                float* __restrict__ shadowW,
+                                float* __restrict__ shadowV, int K,
+                                int updater, float lr, float lam,
+                                float eps) {
+  const int i = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  const int lane = threadIdx.x & 63;
+  if (i >= n) return;
+  const long f = lidx[i];
+  for (int k = lane; k < K + 1; k += LCTR_WAVE) {
+    const bool is_w = (k == K);
+    float* p = is_w ? &W[f] : &V[f * K + k];
+    float* st = is_w ? (nW ? &nW[f] : nullptr)
+                     : (nV ? &nV[f * K + k] : nullptr);
+    float* sh = is_w ? (shadowW ? &shadowW[f] : nullptr)
+                     : (shadowV ? &shadowV[f * K + k] : nullptr);
+    const float g = is_w ? gW[i] : gV[(size_t)i * K + k];
+    if (updater == 0) {  // sgd
+      *p -= lr * g;
+    } else if (updater == 1) {  // adagrad
+      const float acc = *st + g * g;
+      *st = acc;
+      *p -= lr * g * __frsqrt_rn(acc + eps);
+    } else {  // dcasgd / dcasgda
+      float l = lam;
+      if (updater == 3) {
+        const float acc = 0.95f * (*st) + 0.05f * g * g;
+        *st = acc;
+        l = lam * __frsqrt_rn(acc + eps);
+      }
+      const float w = *p;
+      const float wn = w - lr * (g + l * g * g * (w - *sh));
+      *p = wn;
+      *sh = wn;
+    }
+  }
+}
+
+void ps_apply_launch(const long* lidx, int n, const float* gW,
+                     const float* gV, float* W, float* V, float* nW,
+                     float* nV, float* shadowW, float* shadowV, int K,
+                     int updater, float lr, float lam, float eps,
+                     hipStream_t stream) {
+  if (n <= 0) return;
+  dim3 block(256);
+  dim3 grid((n + 3) / 4);
+  hipLaunchKernelGGL(ps_apply_kernel, grid, block, 0, stream, lidx, n, gW,
+                     gV, W, V, nW, nV, shadowW, shadowV, K, updater, lr,
+                     lam, eps);
+}
+
 void sparse_adagrad_apply_launch(const int* uniq, const int* count, float* W,
                                  float* V, float* nW, float* nV, float* gradW,
                                  float* gradV, float lr, float eps, float l2,

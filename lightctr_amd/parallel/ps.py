@@ -199,6 +199,21 @@ class PSShard:
     def _apply(self, wi, lidx, gW, gV):
         cfg = self.cfg
         lr = cfg.lr
+        if self.device.type == "cuda":
+            # fused per-shard updater kernel (GPU-resident state; reference
+            # paramserver.h:217-306 semantics)
+            from ..ops._extension import require_hip_ops
+
+            ops = require_hip_ops()
+            upd = {"sgd": 0, "adagrad": 1, "dcasgd": 2, "dcasgda": 3}[
+                cfg.updater]
+            ops.ps_apply(
+                lidx, gW.contiguous(), gV.contiguous(), self.W, self.V,
+                getattr(self, "nW", None), getattr(self, "nV", None),
+                self.shadowW[wi] if hasattr(self, "shadowW") else None,
+                self.shadowV[wi] if hasattr(self, "shadowV") else None,
+                upd, lr, cfg.dc_lambda, cfg.eps)
+            return
         if cfg.updater == "sgd":
             self.W.index_add_(0, lidx, -lr * gW)
             self.V.index_add_(0, lidx, -lr * gV)

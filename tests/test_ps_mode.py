@@ -149,3 +149,64 @@ def test_lowbit_codec_gpu_parity():
         y_cpu = c.decode(w_cpu, meta_cpu, n)
         y_gpu = c.decode(w_gpu, meta_gpu, n)
         assert torch.allclose(y_gpu.cpu(), y_cpu, atol=1e-6)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("updater", ["sgd", "adagrad", "dcasgd", "dcasgda"])
+def test_ps_apply_kernel_parity(updater):
+    """GPU fused PS updater kernel vs the torch reference math."""
+    from lightctr_amd.ops import hip_ops
+
+    F, K, n = 500, 8, 120
+    g = torch.Generator().manual_seed(17)
+    lidx = torch.randint(0, F, (n,), generator=g).unique().long().cuda()
+    n = lidx.numel()
+    gW = torch.randn(n, generator=g).cuda() * 0.01
+    gV = torch.randn(n, K, generator=g).cuda() * 0.01
+    W = torch.randn(F, generator=g).cuda()
+    V = (torch.randn(F, K, generator=g) * 0.1).cuda()
+    nW = torch.rand(F, generator=g).cuda() * 0.1
+    nV = torch.rand(F, K, generator=g).cuda() * 0.1
+    shW = torch.randn(F, generator=g).cuda() * 0.1
+    shV = torch.randn(F, K, generator=g).cuda() * 0.1
+    lr, lam, eps = 0.05, 0.1, 1e-8
+    # torch reference (mirrors parallel/ps.PSShard._apply CPU math)
+    rW, rV = W.clone(), V.clone()
+    rnW, rnV, rshW, rshV = nW.clone(), nV.clone(), shW.clone(), shV.clone()
+    if updater == "sgd":
+        rW.index_add_(0, lidx, -lr * gW)
+        rV.index_add_(0, lidx, -lr * gV)
+    elif updater == "adagrad":
+        rnW[lidx] += gW * gW
+        rnV[lidx] += gV * gV
+        rW[lidx] -= lr * gW / (rnW[lidx] + eps).sqrt()
+        rV[lidx] -= lr * gV / (rnV[lidx] + eps).sqrt()
+    else:
+        lamW = lamV = lam
+        if updater == "dcasgda":
+            rnW[lidx] = 0.95 * rnW[lidx] + 0.05 * gW * gW
+            rnV[lidx] = 0.95 * rnV[lidx] + 0.05 * gV * gV
+            lamW = lam / (rnW[lidx] + eps).sqrt()
+            lamV = lam / (rnV[lidx] + eps).sqrt()
+        newW = rW[lidx] - lr * (gW + lamW * gW * gW
+                                * (rW[lidx] - rshW[lidx]))
+        newV = rV[lidx] - lr * (gV + lamV * gV * gV
+                                * (rV[lidx] - rshV[lidx]))
+        rW[lidx] = newW
+        rV[lidx] = newV
+        rshW[lidx] = newW
+        rshV[lidx] = newV
+    upd = {"sgd": 0, "adagrad": 1, "dcasgd": 2, "dcasgda": 3}[updater]
+    hip_ops.ps_apply(lidx, gW, gV, W, V,
+                     nW if updater in ("adagrad", "dcasgda") else None,
+                     nV if updater in ("adagrad", "dcasgda") else None,
+                     shW if updater.startswith("dcasgd") else None,
+                     shV if updater.startswith("dcasgd") else None,
+                     upd, lr, lam, eps)
+    assert torch.allclose(W, rW, atol=1e-5, rtol=1e-5), updater
+    assert torch.allclose(V, rV, atol=1e-5, rtol=1e-5), updater
+    if updater in ("adagrad", "dcasgda"):
+        assert torch.allclose(nW, rnW, atol=1e-6)
+        assert torch.allclose(nV, rnV, atol=1e-6)
+    if updater.startswith("dcasgd"):
+        assert torch.allclose(shV, rshV, atol=1e-5)
