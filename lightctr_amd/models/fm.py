@@ -38,6 +38,9 @@ class FMHyper:
     ftrl_beta: float = 1.0
     ftrl_l1: float = 1e-4
     ftrl_l2: float = 1e-4
+    # latent-factor updater under optimizer="ftrl": "adagrad" (default —
+    # FTRL's L1 starves interactions, measured 0.66 vs 0.79 AUC) | "ftrl"
+    ftrl_v: str = "adagrad"
     init_sigma: float = 0.01
     seed: int = 1234
 
@@ -111,7 +114,9 @@ class FMModel:
                 gw, gv = ops.fm_backward_emit(row_ptr, fids, vals, self.V,
                                               sumVX, dpred)
                 sorted_fids, perm = torch.sort(fids)
-                if self.fused_apply:
+                if self.fused_apply and not (
+                        self.h.optimizer == "ftrl"
+                        and self.h.ftrl_v == "adagrad"):
                     if self.h.optimizer == "ftrl":
                         ops.fm_sorted_apply_fused(
                             sorted_fids, perm, gw, gv, self.gradW,
@@ -140,7 +145,9 @@ class FMModel:
                                   self.zW, self.nW, self.zV, self.nV,
                                   self.gradW, self.gradV, self.h.ftrl_alpha,
                                   self.h.ftrl_beta, self.h.ftrl_l1,
-                                  self.h.ftrl_l2)
+                                  self.h.ftrl_l2,
+                                  1 if self.h.ftrl_v == "adagrad" else 0,
+                                  self.h.lr, self.h.eps, self.h.l2)
             else:
                 ops.fm_adagrad_apply(live, self.count, self.W, self.V,
                                      self.nW, self.nV, self.gradW, self.gradV,
@@ -157,7 +164,10 @@ class FMModel:
             fm_ref.ftrl_apply_ref(uniq, self.W, self.V, self.zW, self.nW,
                                   self.zV, self.nV, self.gradW, self.gradV,
                                   self.h.ftrl_alpha, self.h.ftrl_beta,
-                                  self.h.ftrl_l1, self.h.ftrl_l2)
+                                  self.h.ftrl_l1, self.h.ftrl_l2,
+                                  v_adagrad=self.h.ftrl_v == "adagrad",
+                                  v_lr=self.h.lr, v_eps=self.h.eps,
+                                  v_l2=self.h.l2)
         else:
             fm_ref.adagrad_apply_ref(uniq, self.W, self.V, self.nW, self.nV,
                                      self.gradW, self.gradV, self.h.lr,

@@ -317,7 +317,9 @@ void fm_adagrad_apply(at::Tensor uniq, at::Tensor count, at::Tensor W,
 void fm_ftrl_apply(at::Tensor uniq, at::Tensor count, at::Tensor W,
                    at::Tensor V, at::Tensor zW, at::Tensor nW, at::Tensor zV,
                    at::Tensor nV, at::Tensor gradW, at::Tensor gradV,
-                   double alpha, double beta, double l1, double l2) {
+                   double alpha, double beta, double l1, double l2,
+                   int64_t v_adagrad, double v_lr, double v_eps,
+                   double v_l2) {
   check_cuda_i32(uniq, "uniq");
   check_cuda_i32(count, "count");
   const int K = (int)V.size(1);
@@ -326,7 +328,8 @@ void fm_ftrl_apply(at::Tensor uniq, at::Tensor count, at::Tensor W,
       V.data_ptr<float>(), zW.data_ptr<float>(), nW.data_ptr<float>(),
       zV.data_ptr<float>(), nV.data_ptr<float>(), gradW.data_ptr<float>(),
       gradV.data_ptr<float>(), (float)alpha, (float)beta, (float)l1,
-      (float)l2, (int)uniq.numel(), K, cur_stream());
+      (float)l2, (int)uniq.numel(), K, (int)v_adagrad, (float)v_lr,
+      (float)v_eps, (float)v_l2, cur_stream());
 }
 
 // ---- dense NN ops ----
@@ -694,5 +697,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "NFM per-entry grads for sorted apply");
   m.def("bitmap_compact", &bitmap_compact, "touched bitmap -> fid list");
   m.def("fm_adagrad_apply", &fm_adagrad_apply, "sparse fused Adagrad");
-  m.def("fm_ftrl_apply", &fm_ftrl_apply, "sparse fused FTRL-proximal");
+  m.def("fm_ftrl_apply", &fm_ftrl_apply,
+        "sparse fused FTRL-proximal (optionally Adagrad on V)",
+        py::arg("uniq"), py::arg("count"), py::arg("W"), py::arg("V"),
+        py::arg("zW"), py::arg("nW"), py::arg("zV"), py::arg("nV"),
+        py::arg("gradW"), py::arg("gradV"), py::arg("alpha"),
+        py::arg("beta"), py::arg("l1"), py::arg("l2"),
+        py::arg("v_adagrad") = 0, py::arg("v_lr") = 0.05,
+        py::arg("v_eps") = 1e-8, py::arg("v_l2") = 1e-5);
 }

@@ -89,6 +89,7 @@ void fm_ftrl_apply_launch(const int* uniq, const int* count, float* W,
                           float* V, float* zW, float* nW, float* zV, float* nV,
                           float* gradW, float* gradV, float alpha, float beta,
                           float l1, float l2, int capacity, int K,
+                          int v_adagrad, float v_lr, float v_eps, float v_l2,
                           ihipStream_t* stream);
 
 // --- gemm_kernels.hip ---

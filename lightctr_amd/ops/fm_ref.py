@@ -80,9 +80,12 @@ def adagrad_apply_ref(uniq, W, V, nW, nV, gradW, gradV, lr, eps, l2):
 
 
 def ftrl_apply_ref(uniq, W, V, zW, nW, zV, nV, gradW, gradV,
-                   alpha, beta, l1, l2):
+                   alpha, beta, l1, l2, v_adagrad=False, v_lr=0.05,
+                   v_eps=1e-8, v_l2=1e-5):
     """Sparse FTRL-proximal over unique fids (per-coordinate; reference
-    gradientUpdater.h:235-278 semantics). Zeroes applied grad slots."""
+    gradientUpdater.h:235-278 semantics). Zeroes applied grad slots.
+    v_adagrad: FTRL on W only, Adagrad on the latent V block (the classic
+    CTR split; see fm_kernels.hip)."""
     u = uniq.long()
 
     def upd(w, z, n, g):
@@ -98,7 +101,12 @@ def ftrl_apply_ref(uniq, W, V, zW, nW, zV, nV, gradW, gradV,
         )
         return w_new, z_new, n_new
 
-    V[u], zV[u], nV[u] = upd(V[u], zV[u], nV[u], gradV[u])
+    if v_adagrad:
+        gv = gradV[u] + v_l2 * V[u]
+        nV[u] = nV[u] + gv * gv
+        V[u] = V[u] - v_lr * gv / torch.sqrt(nV[u] + v_eps)
+    else:
+        V[u], zV[u], nV[u] = upd(V[u], zV[u], nV[u], gradV[u])
     gradV[u] = 0
     W[u], zW[u], nW[u] = upd(W[u], zW[u], nW[u], gradW[u])
     gradW[u] = 0

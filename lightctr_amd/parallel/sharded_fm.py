@@ -142,7 +142,9 @@ class ShardedFMModel:
                 ops.fm_ftrl_apply(self.uniq, self.count, self.W, self.V,
                                   self.zW, self.nW, self.zV, self.nV,
                                   self.gradW, self.gradV, h.ftrl_alpha,
-                                  h.ftrl_beta, h.ftrl_l1, h.ftrl_l2)
+                                  h.ftrl_beta, h.ftrl_l1, h.ftrl_l2,
+                                  1 if h.ftrl_v == "adagrad" else 0, h.lr,
+                                  h.eps, h.l2)
             else:
                 ops.fm_adagrad_apply(self.uniq, self.count, self.W, self.V,
                                      self.nW, self.nV, self.gradW, self.gradV,
@@ -155,7 +157,9 @@ class ShardedFMModel:
                 fm_ref.ftrl_apply_ref(own_uniq, self.W, self.V, self.zW,
                                       self.nW, self.zV, self.nV, self.gradW,
                                       self.gradV, h.ftrl_alpha, h.ftrl_beta,
-                                      h.ftrl_l1, h.ftrl_l2)
+                                      h.ftrl_l1, h.ftrl_l2,
+                                      v_adagrad=h.ftrl_v == "adagrad",
+                                      v_lr=h.lr, v_eps=h.eps, v_l2=h.l2)
             else:
                 fm_ref.adagrad_apply_ref(own_uniq, self.W, self.V, self.nW,
                                          self.nV, self.gradW, self.gradV,

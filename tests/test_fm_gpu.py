@@ -177,8 +177,10 @@ def test_fused_apply_matches_two_phase():
     from lightctr_amd.data.synthetic import SyntheticCriteo
 
     for opt in ("adagrad", "ftrl"):
-        h = FMHyper(num_features=1 << 15, k=16, optimizer=opt, seed=31)
+        h = FMHyper(num_features=1 << 15, k=16, optimizer=opt, seed=31,
+                    ftrl_v="ftrl")
         a = FMModel(h, device="cuda:0")
+        a.fused_apply = True
         b = FMModel(h, device="cuda:0")
         b.W.copy_(a.W)
         b.V.copy_(a.V)
