@@ -177,6 +177,16 @@ class EmbedModel:
                 self.neg_cdf,
                 torch.rand(B, h.negatives, generator=self._gen)
                 .to(self.device)).clamp(max=self.E.shape[0] - 1)
+            if self.device.type == "cuda" and h.dim <= 64:
+                # fused CDNA4 kernel (Hogwild across examples, like the
+                # reference's lock-free update)
+                from ..ops._extension import require_hip_ops
+
+                loss_t = require_hip_ops().w2v_negsample_step(
+                    self.E, self.O, centers.long().contiguous(),
+                    ctx.long().contiguous(), neg.long().contiguous(),
+                    h.lr, 1.0 / B)
+                return float(loss_t.mean())
             tgt = torch.cat([centers.unsqueeze(1), neg], dim=1)  # [B, 1+N]
             lbl = torch.zeros(B, 1 + h.negatives, device=self.device)
             lbl[:, 0] = 1.0

@@ -465,6 +465,25 @@ std::vector<at::Tensor> embed_backward_emit(at::Tensor row_ptr,
   return {gw, gv};
 }
 
+at::Tensor w2v_negsample_step(at::Tensor E, at::Tensor O,
+                              at::Tensor centers, at::Tensor ctx,
+                              at::Tensor negs, double lr, double scale) {
+  check_cuda_f32(E, "E");
+  check_cuda_f32(O, "O");
+  CHK(centers.scalar_type() == at::kLong, "centers i64");
+  const int B = (int)centers.numel();
+  const int C = (int)ctx.size(1);
+  const int N = (int)negs.size(1);
+  const int D = (int)E.size(1);
+  CHK(D <= 64, "w2v kernel supports dim <= 64");
+  auto loss = at::empty({B}, E.options());
+  lightctr::w2v_negsample_launch(
+      E.data_ptr<float>(), O.data_ptr<float>(), centers.data_ptr<long>(),
+      ctx.data_ptr<long>(), negs.data_ptr<long>(), loss.data_ptr<float>(),
+      B, C, N, D, (float)lr, (float)scale, cur_stream());
+  return loss;
+}
+
 at::Tensor wide_forward(at::Tensor row_ptr, at::Tensor fids, at::Tensor vals,
                         at::Tensor W) {
   check_cuda_i32(row_ptr, "row_ptr");
@@ -686,6 +705,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("embed_backward_emit", &embed_backward_emit,
         "per-entry embedding grads (gw, gv) for sorted apply");
   m.def("wide_forward", &wide_forward, "LR wide forward sum");
+  m.def("w2v_negsample_step", &w2v_negsample_step,
+        "fused CBOW negative-sampling train step (Hogwild)");
   m.def("quantile_encode", &quantile_encode, "int8 quantile encode");
   m.def("quantile_decode", &quantile_decode, "int8 quantile decode");
   m.def("lowbit_encode", &lowbit_encode, "1/2-bit sign quantize");

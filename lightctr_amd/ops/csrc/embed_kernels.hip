@@ -154,6 +154,72 @@ void embed_backward_emit_launch(const int* row_ptr, const float* vals,
                                    dwide, gv, gw, nf, B));
 }
 
+// ---------------------------------------------------------------------------
+// Fused CBOW negative-sampling step (word2vec). One wave per example,
+// lane = embedding dim (D <= 64; strided loop above):
+//   x = mean(E[ctx]); for t in {center} + negs: p = sigmoid(x . O[t]),
+//   g = (p - y)*scale; O[t] -= lr*g*x (atomic); gx += g*O[t];
+//   E[c] -= lr*gx/C for each ctx word (atomic).
+// Updates are Hogwild-style across examples — the reference's own
+// semantics ("unsafe multi-thread update", train_embed_algo.cpp:195-200).
+// Replaces ~6 torch kernels per step with one fused launch.
+// ---------------------------------------------------------------------------
+__global__ void w2v_negsample_kernel(
+    float* __restrict__ E, float* __restrict__ O,
+    const long* __restrict__ centers, const long* __restrict__ ctx,
+    const long* __restrict__ negs, float* __restrict__ loss, int B, int C,
+    int N, int D, float lr, float scale) {
+  const int lane = threadIdx.x & 63;
+  const int ex = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  if (ex >= B) return;
+  // D <= 64 fast path: lane == dim
+  if (lane >= D) {
+    // still participate in shuffles with zeros
+  }
+  const float inv_c = 1.f / C;
+  float x = 0.f;
+  for (int c = 0; c < C; ++c) {
+    const long w = ctx[(size_t)ex * C + c];
+    if (lane < D) x += E[(size_t)w * D + lane];
+  }
+  x *= inv_c;
+  float gx = 0.f;
+  float lsum = 0.f;
+  for (int t = 0; t < N + 1; ++t) {
+    const long tgt = (t == 0) ? centers[ex]
+                              : negs[(size_t)ex * N + (t - 1)];
+    const float y = (t == 0) ? 1.f : 0.f;
+    const float o = (lane < D) ? O[(size_t)tgt * D + lane] : 0.f;
+    float dot = wave_reduce_sum(x * o);
+    dot = fminf(16.f, fmaxf(-16.f, dot));
+    const float p = 1.f / (1.f + __expf(-dot));
+    if (lane == 0) {
+      lsum += -(y * __logf(fmaxf(p, 1e-7f))
+                + (1.f - y) * __logf(fmaxf(1.f - p, 1e-7f)));
+    }
+    const float g = (p - y) * scale;
+    gx += g * o;
+    if (lane < D) atomicAdd(&O[(size_t)tgt * D + lane], -lr * g * x);
+  }
+  const float ge = lr * gx * inv_c;
+  for (int c = 0; c < C; ++c) {
+    const long w = ctx[(size_t)ex * C + c];
+    if (lane < D) atomicAdd(&E[(size_t)w * D + lane], -ge);
+  }
+  if (lane == 0) loss[ex] = lsum;
+}
+
+void w2v_negsample_launch(float* E, float* O, const long* centers,
+                          const long* ctx, const long* negs, float* loss,
+                          int B, int C, int N, int D, float lr, float scale,
+                          hipStream_t stream) {
+  if (B <= 0) return;
+  dim3 block(256);
+  dim3 grid((B + 3) / 4);
+  hipLaunchKernelGGL(w2v_negsample_kernel, grid, block, 0, stream, E, O,
+                     centers, ctx, negs, loss, B, C, N, D, lr, scale);
+}
+
 // wide (LR) forward: wide[row] = sum_j W[fid_j] * x_j
 __global__ void wide_forward_kernel(const int* __restrict__ row_ptr,
                                     const int* __restrict__ fids,

@@ -131,6 +131,10 @@ void nfm_forward_launch(const int* row_ptr, const int* fids, const float* vals,
                         const float* W, const float* V, float* wide,
                         float* sumVX, float* vec, void* vec_bf, int B, int K,
                         ihipStream_t* stream);
+void w2v_negsample_launch(float* E, float* O, const long* centers,
+                          const long* ctx, const long* negs, float* loss,
+                          int B, int C, int N, int D, float lr, float scale,
+                          ihipStream_t* stream);
 void nfm_backward_emit_launch(const int* row_ptr, const int* fids,
                               const float* vals, const float* V,
                               const float* sumVX, const float* dvec,
