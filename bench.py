@@ -32,7 +32,8 @@ def build_model(args, device, world):
         if world > 1:
             from lightctr_amd.parallel.sharded_fm import ShardedFMModel
 
-            return ShardedFMModel(hyper, device=device), "sparse"
+            return (ShardedFMModel(hyper, device=device, wire="fp16"),
+                    "sparse")
         return FMModel(hyper, device=device), "sparse"
     if args.model == "ffm":
         from lightctr_amd.models.ffm import FFMHyper, FFMModel

@@ -26,10 +26,15 @@ from ..ops._extension import require_hip_ops
 
 
 class ShardedFMModel:
-    def __init__(self, hyper: FMHyper, device: str = "cpu", group=None):
+    def __init__(self, hyper: FMHyper, device: str = "cpu", group=None,
+                 wire: str = "fp32"):
+        """wire="fp16" sends parameter/gradient payloads in half precision
+        (the reference's fp16 PS wire, float16.h), halving xGMI bytes;
+        fp32 is exact and is what the equivalence tests pin."""
         self.h = hyper
         self.device = torch.device(device)
         self.group = group
+        self.wire = wire
         self.world = dist.get_world_size(group)
         self.rank = dist.get_rank(group)
         F, K = hyper.num_features, hyper.k
@@ -58,12 +63,16 @@ class ShardedFMModel:
     # ---- parameter exchange ----
     def _exchange(self, send: torch.Tensor, send_counts, recv_counts):
         """all_to_all_single with known splits; send/recv along dim 0."""
+        if self.wire == "fp16" and send.is_floating_point():
+            send = send.to(torch.float16)
         out_shape = (sum(recv_counts),) + tuple(send.shape[1:])
         out = torch.empty(out_shape, dtype=send.dtype, device=send.device)
         dist.all_to_all_single(out, send.contiguous(),
                                output_split_sizes=recv_counts,
                                input_split_sizes=send_counts,
                                group=self.group)
+        if out.dtype == torch.float16:
+            out = out.float()
         return out
 
     def train_step(self, row_ptr, fids, vals, labels) -> torch.Tensor:

@@ -438,3 +438,34 @@ def _sharded_ffm_worker(rank, port, q):
 def test_sharded_ffm_matches_single():
     ok, dv = _run_spawn(_sharded_ffm_worker, 29537)
     assert ok, f"V maxdiff {dv}"
+
+
+def _sharded_fm_fp16_worker(rank, port, q):
+    try:
+        dist = _init(rank, port)
+        from lightctr_amd.models.fm import FMHyper
+        from lightctr_amd.parallel.sharded_fm import ShardedFMModel
+        from conftest import make_random_csr
+
+        h = FMHyper(num_features=1000, k=8, optimizer="adagrad", seed=99)
+        m = ShardedFMModel(h, device="cpu", wire="fp16")
+        for step in range(3):
+            rp, fi, v, lb = make_random_csr(B=64, F_total=1000,
+                                            seed=step * 5 + rank,
+                                            binary_vals=False)
+            loss = m.train_step(rp, fi, v, lb)
+            assert torch.isfinite(loss).all()
+        if rank == 0:
+            q.put(("result", bool(torch.isfinite(m.V).all()
+                                  and m.V.abs().sum() > 0)))
+        dist.destroy_process_group()
+    except Exception:  # pragma: no cover
+        import traceback
+
+        q.put(("error", rank, traceback.format_exc()))
+        raise
+
+
+def test_sharded_fm_fp16_wire_trains():
+    (ok,) = _run_spawn(_sharded_fm_fp16_worker, 29538)
+    assert ok
