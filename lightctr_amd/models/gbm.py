@@ -162,14 +162,18 @@ class GBMModel:
             # nodes at the deepest level must close as leaves (their values
             # come from this level's histograms)
             split = (best_gain > h.gamma) & (depth < h.max_depth - 1)
+            # one device->host copy per level (not per node)
+            split_h = split.cpu()
+            bf_h, bt_h, bnl_h = bf.cpu(), bt.cpu(), bnl.cpu()
+            G_h, H_h = G[:, 0].cpu(), H[:, 0].cpu()
             new_level = []
             child_of = torch.full((nl, 2), -1, dtype=torch.int32)
             for j in range(nl):
                 gid = level_global[j]
-                if bool(split[j]):
-                    feature[gid] = int(bf[j])
-                    thr[gid] = int(bt[j])
-                    nanl[gid] = bool(bnl[j])
+                if bool(split_h[j]):
+                    feature[gid] = int(bf_h[j])
+                    thr[gid] = int(bt_h[j])
+                    nanl[gid] = bool(bnl_h[j])
                     lgid = self._emit(feature, thr, nanl, left, right, value)
                     rgid = self._emit(feature, thr, nanl, left, right, value)
                     left[gid], right[gid] = lgid, rgid
@@ -179,7 +183,7 @@ class GBMModel:
                     new_level.append(rgid)
                 else:
                     lam = h.reg_lambda
-                    value[gid] = float(-G[j, 0] / (H[j, 0] + lam)
+                    value[gid] = float(-G_h[j] / (H_h[j] + lam)
                                        * h.learning_rate)
             if not new_level:
                 break
