@@ -142,10 +142,10 @@ def test_embedding_gpu_fused_kernel_trains():
         toks.extend(f"{grp}{int(i)}" for i in torch.randperm(5, generator=g))
     vocab, counts = vocab_from_tokens(toks)
     m = EmbedModel(vocab, counts,
-                   EmbedHyper(dim=32, window=2, lr=0.5, subsample_t=1e2),
+                   EmbedHyper(dim=32, window=2, lr=2.0, subsample_t=1e2),
                    device="cuda:0")
     ids = torch.tensor([m.word2id[t] for t in toks])
-    m.train_stream(ids, epochs=4, batch=512)
+    m.train_stream(ids, epochs=6, batch=512)
     m.normalize()
     sims = m.E @ m.E.t()
     a = [m.word2id[f"a{i}"] for i in range(5)]
