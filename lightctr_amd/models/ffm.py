@@ -31,6 +31,7 @@ class FFMHyper:
     ftrl_beta: float = 1.0
     ftrl_l1: float = 1e-4
     ftrl_l2: float = 1e-4
+    ftrl_v: str = "adagrad"  # latent updater under ftrl (see models/fm.py)
     init_sigma: float = 0.01
     seed: int = 1234
 
@@ -87,7 +88,9 @@ class FFMModel:
                                   self.zW, self.nW, self.zV, self.nV,
                                   self.gradW, self.gradV, self.h.ftrl_alpha,
                                   self.h.ftrl_beta, self.h.ftrl_l1,
-                                  self.h.ftrl_l2)
+                                  self.h.ftrl_l2,
+                                  1 if self.h.ftrl_v == "adagrad" else 0,
+                                  self.h.lr, self.h.eps, self.h.l2)
         else:
             ops.sparse_adagrad_apply(live, self.count, self.W, self.V,
                                      self.nW, self.nV, self.gradW, self.gradV,
@@ -142,7 +145,10 @@ class FFMModel:
             fm_ref.ftrl_apply_ref(uniq, self.W, Vf, self.zW, self.nW,
                                   self.zV.view(F, -1), nVf, self.gradW, gVf,
                                   self.h.ftrl_alpha, self.h.ftrl_beta,
-                                  self.h.ftrl_l1, self.h.ftrl_l2)
+                                  self.h.ftrl_l1, self.h.ftrl_l2,
+                                  v_adagrad=self.h.ftrl_v == "adagrad",
+                                  v_lr=self.h.lr, v_eps=self.h.eps,
+                                  v_l2=self.h.l2)
         else:
             fm_ref.adagrad_apply_ref(uniq, self.W, Vf, self.nW, nVf,
                                      self.gradW, gVf, self.h.lr, self.h.eps,

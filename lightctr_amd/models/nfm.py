@@ -34,6 +34,7 @@ class NFMHyper:
     ftrl_beta: float = 1.0
     ftrl_l1: float = 1e-4
     ftrl_l2: float = 1e-4
+    ftrl_v: str = "adagrad"  # latent updater under ftrl (see models/fm.py)
     dropout: float = 0.0
     init_sigma: float = 0.01
     seed: int = 1234
@@ -128,7 +129,9 @@ class NFMModel:
                                   self.zW, self.nW, self.zV, self.nV,
                                   self.gradW, self.gradV, self.h.ftrl_alpha,
                                   self.h.ftrl_beta, self.h.ftrl_l1,
-                                  self.h.ftrl_l2)
+                                  self.h.ftrl_l2,
+                                  1 if self.h.ftrl_v == "adagrad" else 0,
+                                  self.h.lr, self.h.eps, self.h.l2)
             else:
                 ops.fm_adagrad_apply(live, self.count, self.W, self.V,
                                      self.nW, self.nV, self.gradW, self.gradV,
@@ -155,7 +158,10 @@ class NFMModel:
             fm_ref.ftrl_apply_ref(uniq, self.W, self.V, self.zW, self.nW,
                                   self.zV, self.nV, self.gradW, self.gradV,
                                   self.h.ftrl_alpha, self.h.ftrl_beta,
-                                  self.h.ftrl_l1, self.h.ftrl_l2)
+                                  self.h.ftrl_l1, self.h.ftrl_l2,
+                                  v_adagrad=self.h.ftrl_v == "adagrad",
+                                  v_lr=self.h.lr, v_eps=self.h.eps,
+                                  v_l2=self.h.l2)
         else:
             fm_ref.adagrad_apply_ref(uniq, self.W, self.V, self.nW, self.nV,
                                      self.gradW, self.gradV, self.h.lr,

@@ -277,7 +277,8 @@ void sparse_ftrl_apply(at::Tensor uniq, at::Tensor count, at::Tensor W,
                        at::Tensor V, at::Tensor zW, at::Tensor nW,
                        at::Tensor zV, at::Tensor nV, at::Tensor gradW,
                        at::Tensor gradV, double alpha, double beta, double l1,
-                       double l2) {
+                       double l2, int64_t v_adagrad, double v_lr,
+                       double v_eps, double v_l2) {
   check_cuda_i32(uniq, "uniq");
   const int D = (int)(V.numel() / V.size(0));
   lightctr::sparse_ftrl_apply_launch(
@@ -285,7 +286,8 @@ void sparse_ftrl_apply(at::Tensor uniq, at::Tensor count, at::Tensor W,
       V.data_ptr<float>(), zW.data_ptr<float>(), nW.data_ptr<float>(),
       zV.data_ptr<float>(), nV.data_ptr<float>(), gradW.data_ptr<float>(),
       gradV.data_ptr<float>(), (float)alpha, (float)beta, (float)l1,
-      (float)l2, (int)uniq.numel(), D, cur_stream());
+      (float)l2, (int)uniq.numel(), D, (int)v_adagrad, (float)v_lr,
+      (float)v_eps, (float)v_l2, cur_stream());
 }
 
 at::Tensor bitmap_compact(at::Tensor bitmap, at::Tensor out_fids,
@@ -691,7 +693,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ps_apply", &ps_apply,
         "PS-side fused updater (sgd/adagrad/dcasgd/dcasgda)");
   m.def("sparse_ftrl_apply", &sparse_ftrl_apply,
-        "generic sparse fused FTRL (runtime D)");
+        "generic sparse fused FTRL (runtime D; optional Adagrad on V)",
+        py::arg("uniq"), py::arg("count"), py::arg("W"), py::arg("V"),
+        py::arg("zW"), py::arg("nW"), py::arg("zV"), py::arg("nV"),
+        py::arg("gradW"), py::arg("gradV"), py::arg("alpha"),
+        py::arg("beta"), py::arg("l1"), py::arg("l2"),
+        py::arg("v_adagrad") = 0, py::arg("v_lr") = 0.05,
+        py::arg("v_eps") = 1e-8, py::arg("v_l2") = 1e-5);
   m.def("gemm_bf16", &gemm_bf16, "MFMA bf16 GEMM (C fp32), fused bias+act");
   m.def("gemm_bf16_full", &gemm_bf16_full,
         "MFMA bf16 GEMM returning (C fp32, C bf16)");

@@ -80,7 +80,8 @@ void sparse_ftrl_apply_launch(const int* uniq, const int* count, float* W,
                               float* V, float* zW, float* nW, float* zV,
                               float* nV, float* gradW, float* gradV,
                               float alpha, float beta, float l1, float l2,
-                              int capacity, int D, ihipStream_t* stream);
+                              int capacity, int D, int v_adagrad, float v_lr,
+                              float v_eps, float v_l2, ihipStream_t* stream);
 void fm_adagrad_apply_launch(const int* uniq, const int* count, float* W,
                              float* V, float* nW, float* nV, float* gradW,
                              float* gradV, float lr, float eps, float l2,

@@ -59,15 +59,24 @@ __global__ void sparse_ftrl_apply_g_kernel(
     float* __restrict__ W, float* __restrict__ V, float* __restrict__ zW,
     float* __restrict__ nW, float* __restrict__ zV, float* __restrict__ nV,
     float* __restrict__ gradW, float* __restrict__ gradV, float alpha,
-    float beta, float l1, float l2, int capacity, int D) {
+    float beta, float l1, float l2, int capacity, int D, int v_adagrad,
+    float v_lr, float v_eps, float v_l2) {
   const int i = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   const int lane = threadIdx.x & (LCTR_WAVE - 1);
   if (i >= capacity || i >= *count) return;
   const int fid = uniq[i];
   const size_t base = (size_t)fid * D;
   for (int d = lane; d < D; d += LCTR_WAVE) {
-    ftrl_update_g(&V[base + d], &zV[base + d], &nV[base + d], gradV[base + d],
-                  alpha, beta, l1, l2);
+    if (v_adagrad) {  // FTRL on W, Adagrad on the latent block (see
+                      // fm_kernels.hip rationale)
+      const float g = gradV[base + d] + v_l2 * V[base + d];
+      const float acc = nV[base + d] + g * g;
+      nV[base + d] = acc;
+      V[base + d] -= v_lr * g * __frsqrt_rn(acc + v_eps);
+    } else {
+      ftrl_update_g(&V[base + d], &zV[base + d], &nV[base + d],
+                    gradV[base + d], alpha, beta, l1, l2);
+    }
     gradV[base + d] = 0.f;
   }
   if (lane == 0 && W != nullptr) {
@@ -156,12 +165,13 @@ void sparse_ftrl_apply_launch(const int* uniq, const int* count, float* W,
                               float* V, float* zW, float* nW, float* zV,
                               float* nV, float* gradW, float* gradV,
                               float alpha, float beta, float l1, float l2,
-                              int capacity, int D, hipStream_t stream) {
+                              int capacity, int D, int v_adagrad, float v_lr,
+                              float v_eps, float v_l2, hipStream_t stream) {
   dim3 block(256);
   dim3 grid((capacity + 3) / 4);
   hipLaunchKernelGGL(sparse_ftrl_apply_g_kernel, grid, block, 0, stream, uniq,
                      count, W, V, zW, nW, zV, nV, gradW, gradV, alpha, beta,
-                     l1, l2, capacity, D);
+                     l1, l2, capacity, D, v_adagrad, v_lr, v_eps, v_l2);
 }
 
 }  // namespace lightctr

@@ -133,7 +133,10 @@ class ShardedFFMModel:
                 ops.sparse_ftrl_apply(live, self.count, self.W, self.V,
                                       self.zW, self.nW, self.zV, self.nV,
                                       self.gradW, self.gradV, h.ftrl_alpha,
-                                      h.ftrl_beta, h.ftrl_l1, h.ftrl_l2)
+                                      h.ftrl_beta, h.ftrl_l1, h.ftrl_l2,
+                                      1 if getattr(h, "ftrl_v", "adagrad")
+                                      == "adagrad" else 0, h.lr, h.eps,
+                                      h.l2)
             else:
                 ops.sparse_adagrad_apply(live, self.count, self.W, self.V,
                                          self.nW, self.nV, self.gradW,
@@ -150,7 +153,11 @@ class ShardedFFMModel:
                                       self.nV.view(Fl, -1), self.gradW,
                                       self.gradV.view(Fl, -1),
                                       h.ftrl_alpha, h.ftrl_beta, h.ftrl_l1,
-                                      h.ftrl_l2)
+                                      h.ftrl_l2,
+                                      v_adagrad=getattr(h, "ftrl_v",
+                                                        "adagrad")
+                                      == "adagrad", v_lr=h.lr, v_eps=h.eps,
+                                      v_l2=h.l2)
             else:
                 fm_ref.adagrad_apply_ref(own, self.W, self.V.view(Fl, -1),
                                          self.nW, self.nV.view(Fl, -1),
