@@ -44,6 +44,21 @@ __device__ __forceinline__ void stage_half_512(
   }
 }
 
+// scalar staging for the K-tail tile (row-major source, 512 threads):
+// thread t covers row=t/2, k-half=(t%2)*32; zero-fills out-of-range K
+__device__ __forceinline__ void stage_tail_512(
+    const __bf16* __restrict__ g, long stride, int k0, int K, __bf16* dst) {
+  const int tid = threadIdx.x;
+  const int r = tid >> 1;  // 0..255
+  const int kk = (tid & 1) * 32;
+  for (int u = 0; u < 32; ++u) {
+    const int ke = kk + u;
+    __bf16 v = (__bf16)0.f;
+    if (k0 + ke < K) v = g[(long)r * stride + k0 + ke];
+    dst[r * G256_BK + swz256(r, ke)] = v;
+  }
+}
+
 __device__ __forceinline__ float act_apply256(float v, int act) {
   if (act == 1) return fmaxf(v, 0.f);
   if (act == 2) return sigmoidf_clamped(v);
@@ -68,13 +83,18 @@ __global__ __launch_bounds__(512, 1) void gemm256_bf16_kernel(
 
   f32x4 acc[8][4] = {};
 
-  const int NT = K / G256_BK;
+  const int NT = (K + G256_BK - 1) / G256_BK;
   // buffer b: A at smem + b*32768, B at A + 16384 elems
   // prologue: stage tile 0 (4 half-tiles: A-top, A-bot, B-top, B-bot)
-  stage_half_512(A + (long)M0 * K, K, smem);
-  stage_half_512(A + (long)(M0 + 128) * K, K, smem + 128 * G256_BK);
-  stage_half_512(Bst + (long)N0 * K, K, smem + 256 * G256_BK);
-  stage_half_512(Bst + (long)(N0 + 128) * K, K, smem + 384 * G256_BK);
+  if (G256_BK <= K) {
+    stage_half_512(A + (long)M0 * K, K, smem);
+    stage_half_512(A + (long)(M0 + 128) * K, K, smem + 128 * G256_BK);
+    stage_half_512(Bst + (long)N0 * K, K, smem + 256 * G256_BK);
+    stage_half_512(Bst + (long)(N0 + 128) * K, K, smem + 384 * G256_BK);
+  } else {
+    stage_tail_512(A + (long)M0 * K, K, 0, K, smem);
+    stage_tail_512(Bst + (long)N0 * K, K, 0, K, smem + 256 * G256_BK);
+  }
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
 
@@ -100,16 +120,25 @@ __global__ __launch_bounds__(512, 1) void gemm256_bf16_kernel(
 #pragma unroll
     for (int p = 0; p < 4; ++p) {
       if (more) {  // prefetch half-tile p of tile t+1
-        if (p == 0)
-          stage_half_512(A + (long)M0 * K + k0, K, nxtA);
-        else if (p == 1)
-          stage_half_512(A + (long)(M0 + 128) * K + k0, K,
-                         nxtA + 128 * G256_BK);
-        else if (p == 2)
-          stage_half_512(Bst + (long)N0 * K + k0, K, nxtB);
-        else
-          stage_half_512(Bst + (long)(N0 + 128) * K + k0, K,
-                         nxtB + 128 * G256_BK);
+        const bool next_full = (k0 + G256_BK <= K);
+        if (next_full) {
+          if (p == 0)
+            stage_half_512(A + (long)M0 * K + k0, K, nxtA);
+          else if (p == 1)
+            stage_half_512(A + (long)(M0 + 128) * K + k0, K,
+                           nxtA + 128 * G256_BK);
+          else if (p == 2)
+            stage_half_512(Bst + (long)N0 * K + k0, K, nxtB);
+          else
+            stage_half_512(Bst + (long)(N0 + 128) * K + k0, K,
+                           nxtB + 128 * G256_BK);
+        } else {  // K tail: zero-filled scalar staging covers all 256
+                  // rows per call — phases 0 (A) and 2 (B) only
+          if (p == 0)
+            stage_tail_512(A + (long)M0 * K, K, k0, K, nxtA);
+          else if (p == 2)
+            stage_tail_512(Bst + (long)N0 * K, K, k0, K, nxtB);
+        }
       }
       bf16x8 a[2][2];
 #pragma unroll
@@ -157,8 +186,10 @@ __global__ __launch_bounds__(512, 1) void gemm256_bf16_kernel(
 }
 
 bool gemm256_eligible(int M, int N, int K, int transA, int transB) {
+  // K%8 keeps the glds row stride 16B-aligned for the full tiles; the
+  // K%64 tail is handled by scalar staging
   return transA == 0 && transB == 0 && M % 256 == 0 && N % 256 == 0 &&
-         K % G256_BK == 0;
+         K % 8 == 0 && K >= 32;
 }
 
 void gemm256_bf16_launch(const void* A, const void* Bst, const float* bias,
