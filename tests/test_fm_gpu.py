@@ -237,3 +237,28 @@ def test_fm_forward_k64():
     pred_ref, sumVX_ref = fm_ref.fm_forward_ref(row_ptr, fids, vals, W, V)
     assert torch.allclose(sumVX, sumVX_ref, atol=1e-4, rtol=1e-4)
     assert torch.allclose(pred, pred_ref, atol=2e-3, rtol=1e-4)
+
+
+@pytest.mark.parametrize("seed", [101, 202, 303, 404])
+def test_fm_step_fuzz_parity(seed):
+    """Randomized CSR shapes/values: one full GPU step == CPU oracle step."""
+    g = torch.Generator().manual_seed(seed)
+    F = int(torch.randint(500, 60_000, (1,), generator=g))
+    K = [4, 8, 16, 32][seed % 4]
+    B = int(torch.randint(16, 700, (1,), generator=g))
+    min_f = int(torch.randint(1, 5, (1,), generator=g))
+    max_f = min_f + int(torch.randint(1, 60, (1,), generator=g))
+    row_ptr, fids, vals, labels = make_random_csr(
+        B=B, F_total=F, min_f=min_f, max_f=max_f, seed=seed,
+        binary_vals=bool(seed % 2))
+    h = FMHyper(num_features=F, k=K,
+                optimizer="ftrl" if seed % 2 else "adagrad", seed=seed)
+    cpu = FMModel(h, device="cpu")
+    gpu = FMModel(h, device="cuda:0")
+    gpu.W.copy_(cpu.W)
+    gpu.V.copy_(cpu.V)
+    cpu.train_step(row_ptr, fids, vals, labels)
+    gpu.train_step(row_ptr.cuda(), fids.cuda(), vals.cuda(), labels.cuda())
+    assert torch.allclose(gpu.W.cpu(), cpu.W, atol=2e-4, rtol=1e-3), \
+        (F, K, B, (gpu.W.cpu() - cpu.W).abs().max())
+    assert torch.allclose(gpu.V.cpu(), cpu.V, atol=2e-4, rtol=1e-3)
