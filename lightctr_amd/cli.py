@@ -77,9 +77,14 @@ def cmd_train(args) -> int:
                                                lr=args.lr),
                              device=dev, batch_size=args.batch,
                              epochs=args.epochs)
+    elif args.model == "lr":
+        from .models.lr import LRHyper, LRTrainer
+
+        tr = LRTrainer(ds, LRHyper(num_features=F, optimizer=args.optimizer,
+                                   lr=args.lr), device=dev,
+                       batch_size=args.batch, epochs=args.epochs)
     else:
-        print(f"unknown model {args.model}", file=sys.stderr)
-        return 2
+        return _train_other(args, dev)
     tr.train(log=lambda msg: print(msg, flush=True))
     metrics = tr.evaluate()
     print(json.dumps({"final": metrics}))
@@ -87,6 +92,197 @@ def cmd_train(args) -> int:
         tr.model.save(args.save)
         print(f"saved -> {args.save}")
     return 0
+
+
+def _dense_data(args):
+    """(X, y) from a dense CSV or a synthetic quadrant-image set."""
+    import torch as t
+
+    if args.data == "synthetic":
+        g = t.Generator().manual_seed(args.seed)
+        size = 16
+        X = t.rand(args.rows, size * size, generator=g) * 0.2
+        y = t.randint(0, 4, (args.rows,), generator=g)
+        hs = size // 2
+        img = X.view(args.rows, size, size)
+        for i in range(args.rows):
+            q = int(y[i])
+            img[i, (q // 2) * hs:(q // 2 + 1) * hs,
+                (q % 2) * hs:(q % 2 + 1) * hs] += 0.8
+        return X, y
+    from .data import load_dense_csv
+
+    return load_dense_csv(args.data, max_rows=args.rows or None)
+
+
+def _train_other(args, dev) -> int:
+    """The rest of the zoo (reference main.cpp TEST_* coverage): gbm, cnn,
+    rnn, vae, gmm, embed, plsa."""
+    import math
+
+    import torch as t
+
+    if args.model == "gbm":
+        from .models.gbm import GBMHyper, GBMModel
+        from .utils.metrics import auc_score
+
+        X, y = _dense_data(args)
+        ncls = int(y.max()) + 1
+        m = GBMModel(GBMHyper(n_rounds=args.epochs * 5, max_depth=5,
+                              n_classes=max(2, ncls), seed=args.seed),
+                     device=dev)
+        m.fit(X.to(dev), y.float().to(dev),
+              log=lambda msg: print(msg, flush=True))
+        p = m.predict_proba(X.to(dev))
+        if p.dim() == 1:
+            print(json.dumps({"final": {"auc": auc_score(p.cpu(),
+                                                         y.float())}}))
+        else:
+            acc = float((p.argmax(dim=1).cpu() == y).float().mean())
+            print(json.dumps({"final": {"accuracy": acc}}))
+        if args.save:
+            m.save(args.save)
+        return 0
+    if args.model == "cnn":
+        from .models.cnn import CNNHyper, CNNModel
+
+        X, y = _dense_data(args)
+        side = int(math.isqrt(X.shape[1]))
+        ncls = int(y.max()) + 1
+        m = CNNModel(CNNHyper(in_shape=(1, side, side), n_classes=ncls,
+                              seed=args.seed), device=dev)
+        Xi = X.view(-1, 1, side, side).to(dev)
+        yi = y.to(dev)
+        for ep in range(args.epochs):
+            perm = t.randperm(Xi.shape[0])
+            tot, nb = 0.0, 0
+            for s0 in range(0, Xi.shape[0], args.batch):
+                idx = perm[s0:s0 + args.batch]
+                tot += m.train_step(Xi[idx], yi[idx])
+                nb += 1
+            print(f"epoch {ep}: loss={tot / max(nb, 1):.5f}", flush=True)
+        acc = float((m.predict_proba(Xi).argmax(dim=1) == yi)
+                    .float().mean())
+        print(json.dumps({"final": {"accuracy": acc}}))
+        if args.save:
+            m.save(args.save)
+        return 0
+    if args.model == "rnn":
+        from .models.rnn import RNNHyper, RNNModel
+
+        X, y = _dense_data(args)
+        side = int(math.isqrt(X.shape[1]))
+        ncls = int(y.max()) + 1
+        m = RNNModel(RNNHyper(in_dim=side, seq_len=side, n_classes=ncls,
+                              seed=args.seed), device=dev)
+        Xi = X.view(-1, side, side).to(dev)
+        yi = y.to(dev)
+        for ep in range(args.epochs):
+            perm = t.randperm(Xi.shape[0])
+            tot, nb = 0.0, 0
+            for s0 in range(0, Xi.shape[0], args.batch):
+                idx = perm[s0:s0 + args.batch]
+                tot += m.train_step(Xi[idx], yi[idx])
+                nb += 1
+            print(f"epoch {ep}: loss={tot / max(nb, 1):.5f}", flush=True)
+        acc = float((m.predict_proba(Xi).argmax(dim=1) == yi)
+                    .float().mean())
+        print(json.dumps({"final": {"accuracy": acc}}))
+        if args.save:
+            m.save(args.save)
+        return 0
+    if args.model == "vae":
+        from .models.vae import VAEHyper, VAEModel
+
+        X, _ = _dense_data(args)
+        Xi = X.to(dev)
+        m = VAEModel(VAEHyper(in_dim=X.shape[1], seed=args.seed),
+                     device=dev)
+        for ep in range(args.epochs):
+            stats = {}
+            for s0 in range(0, Xi.shape[0], args.batch):
+                stats = m.train_step(Xi[s0:s0 + args.batch])
+            print(f"epoch {ep}: {stats}", flush=True)
+        print(json.dumps({"final": stats}))
+        if args.save:
+            m.save(args.save)
+        return 0
+    if args.model == "gmm":
+        from .models.gmm import GMMHyper, GMMModel
+
+        X, _ = _dense_data(args)
+        m = GMMModel(GMMHyper(n_components=max(2, args.k), seed=args.seed),
+                     device=dev)
+        m.fit(X.to(dev), log=lambda msg: print(msg, flush=True))
+        print(json.dumps({"final": {"weights": m.weights.cpu().tolist()}}))
+        if args.save:
+            m.save(args.save)
+        return 0
+    if args.model == "embed":
+        from .models.embedding import (EmbedHyper, EmbedModel,
+                                       vocab_from_tokens)
+
+        if args.data == "synthetic":
+            import torch as t2
+
+            g = t2.Generator().manual_seed(args.seed)
+            toks = []
+            for _ in range(max(200, args.rows)):
+                grp = "a" if t2.rand(1, generator=g) < 0.5 else "b"
+                toks.extend(f"{grp}{int(i)}"
+                            for i in t2.randperm(5, generator=g))
+        else:
+            toks = open(args.data).read().split()
+        vocab, counts = vocab_from_tokens(toks)
+        m = EmbedModel(vocab, counts, EmbedHyper(dim=min(args.k * 4, 64),
+                                                 seed=args.seed),
+                       device=dev)
+        ids = t.tensor([m.word2id[tok] for tok in toks])
+        m.train_stream(ids, epochs=args.epochs,
+                       log=lambda msg: print(msg, flush=True))
+        m.normalize()
+        print(json.dumps({"final": {"vocab": len(vocab),
+                                    "top": m.most_similar(vocab[0], 3)}}))
+        if args.save:
+            m.save(args.save)
+        return 0
+    if args.model == "plsa":
+        from .models.plsa import PLSAHyper, PLSAModel
+
+        if args.data == "synthetic":
+            g = t.Generator().manual_seed(args.seed)
+            docs, words, cnts = [], [], []
+            for d in range(200):
+                for _ in range(30):
+                    w = int(t.randint(0, 10, (1,), generator=g)) \
+                        + (d % 2) * 10
+                    docs.append(d)
+                    words.append(w)
+                    cnts.append(1.0)
+            n_docs, n_words = 200, 20
+        else:
+            vocab = {}
+            docs, words, cnts = [], [], []
+            lines = open(args.data).read().splitlines()
+            for d, line in enumerate(lines):
+                for tok in line.split():
+                    wid = vocab.setdefault(tok, len(vocab))
+                    docs.append(d)
+                    words.append(wid)
+                    cnts.append(1.0)
+            n_docs, n_words = len(lines), len(vocab)
+        m = PLSAModel(PLSAHyper(n_topics=max(2, args.k), seed=args.seed),
+                      device=dev)
+        m.fit(t.tensor(docs), t.tensor(words), t.tensor(cnts), n_docs,
+              n_words, log=lambda msg: print(msg, flush=True))
+        print(json.dumps({"final": {"topics": [m.top_words(z, 5)
+                                               for z in
+                                               range(m.h.n_topics)]}}))
+        if args.save:
+            m.save(args.save)
+        return 0
+    print(f"unknown model {args.model}", file=sys.stderr)
+    return 2
 
 
 def cmd_predict(args) -> int:
