@@ -65,6 +65,7 @@ __device__ __forceinline__ float act_apply256(float v, int act) {
   return v;
 }
 
+template <bool TAIL>
 __global__ __launch_bounds__(512, 1) void gemm256_bf16_kernel(
     const __bf16* __restrict__ A, const __bf16* __restrict__ Bst,
     const float* __restrict__ bias, float* __restrict__ C,
@@ -86,7 +87,7 @@ __global__ __launch_bounds__(512, 1) void gemm256_bf16_kernel(
   const int NT = (K + G256_BK - 1) / G256_BK;
   // buffer b: A at smem + b*32768, B at A + 16384 elems
   // prologue: stage tile 0 (4 half-tiles: A-top, A-bot, B-top, B-bot)
-  if (G256_BK <= K) {
+  if (!TAIL || G256_BK <= K) {
     stage_half_512(A + (long)M0 * K, K, smem);
     stage_half_512(A + (long)(M0 + 128) * K, K, smem + 128 * G256_BK);
     stage_half_512(Bst + (long)N0 * K, K, smem + 256 * G256_BK);
@@ -120,7 +121,7 @@ __global__ __launch_bounds__(512, 1) void gemm256_bf16_kernel(
 #pragma unroll
     for (int p = 0; p < 4; ++p) {
       if (more) {  // prefetch half-tile p of tile t+1
-        const bool next_full = (k0 + G256_BK <= K);
+        const bool next_full = !TAIL || (k0 + G256_BK <= K);
         if (next_full) {
           if (p == 0)
             stage_half_512(A + (long)M0 * K + k0, K, nxtA);
@@ -197,9 +198,15 @@ void gemm256_bf16_launch(const void* A, const void* Bst, const float* bias,
                          hipStream_t stream) {
   dim3 block(512);
   dim3 grid(N / 256, M / 256);
-  hipLaunchKernelGGL(gemm256_bf16_kernel, grid, block, 0, stream,
-                     (const __bf16*)A, (const __bf16*)Bst, bias, C,
-                     (__bf16*)Cbf, M, N, K, act);
+  if (K % G256_BK == 0) {
+    hipLaunchKernelGGL((gemm256_bf16_kernel<false>), grid, block, 0, stream,
+                       (const __bf16*)A, (const __bf16*)Bst, bias, C,
+                       (__bf16*)Cbf, M, N, K, act);
+  } else {
+    hipLaunchKernelGGL((gemm256_bf16_kernel<true>), grid, block, 0, stream,
+                       (const __bf16*)A, (const __bf16*)Bst, bias, C,
+                       (__bf16*)Cbf, M, N, K, act);
+  }
 }
 
 }  // namespace lightctr
