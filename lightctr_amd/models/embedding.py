@@ -111,20 +111,6 @@ class EmbedModel:
         return (torch.rand(ids.numel(), generator=self._gen)
                 < keep_p).to(ids.device)
 
-    def cbow_batch(self, ids: torch.Tensor):
-        """Sliding-window (center, context) pairs from a token-id stream."""
-        W = self.h.window
-        n = ids.numel()
-        ctx_idx = []
-        centers = []
-        for off in range(-W, W + 1):
-            if off == 0:
-                continue
-            lo, hi = max(0, -off), min(n, n - off)
-            ctx_idx.append(torch.arange(lo, hi) + off)
-            centers.append(torch.arange(lo, hi))
-        return ids, centers, ctx_idx
-
     def train_stream(self, ids: torch.Tensor, epochs: int = 1,
                      batch: int = 4096, log=None):
         """ids: long tensor token stream."""

@@ -11,13 +11,13 @@ import sys
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
+import glob  # noqa: E402
+
 from setuptools import setup  # noqa: E402
 from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
 
 CSRC = os.path.join(os.path.dirname(os.path.abspath(__file__)),
                     "lightctr_amd", "ops", "csrc")
-
-import glob
 
 # exclude torch-hipify's generated *_hip.* copies (build artifacts)
 sources = [os.path.join(CSRC, "bindings.cpp")] + sorted(
