@@ -28,6 +28,10 @@ bool gemm256_eligible(int M, int N, int K, int transA, int transB);
 void gemm256_bf16_launch(const void* A, const void* Bst, const float* bias,
                          float* C, void* Cbf, int M, int N, int K, int act,
                          hipStream_t stream);
+bool gemm256v2_eligible(int M, int N, int K, int transA, int transB);
+void gemm256v2_bf16_launch(const void* A, const void* Bst, const float* bias,
+                           float* C, void* Cbf, int M, int N, int K, int act,
+                           hipStream_t stream);
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
@@ -215,6 +219,10 @@ __global__ __launch_bounds__(256) void gemm_bf16_kernel(
 void gemm_bf16_launch(const void* A, const void* Bst, const float* bias,
                       float* C, void* Cbf, int M, int N, int K, int transA,
                       int transB, int act, hipStream_t stream) {
+  if (gemm256v2_eligible(M, N, K, transA, transB)) {
+    gemm256v2_bf16_launch(A, Bst, bias, C, Cbf, M, N, K, act, stream);
+    return;
+  }
   if (gemm256_eligible(M, N, K, transA, transB)) {
     gemm256_bf16_launch(A, Bst, bias, C, Cbf, M, N, K, act, stream);
     return;

@@ -228,3 +228,20 @@ def test_gemm_sigmoid_epilogue_and_256path_bf16_emit():
                         .clamp(-16, 16))
     assert torch.allclose(C, ref, atol=0.02, rtol=1e-2)
     assert torch.allclose(Cbf.float(), C, atol=0.01, rtol=1e-2)
+
+
+@pytest.mark.gpu
+def test_gemm_v2_race_screen():
+    """Sync-structure kernels need multi-run race screens (guide two-lane
+    discipline): 10 repeated runs of a v2-routed shape must all match."""
+    from lightctr_amd.ops import hip_ops
+
+    M, N, K = 2048, 2048, 2048
+    g = torch.Generator().manual_seed(33)
+    A = (torch.randn(M, K, generator=g) * 0.4).to(torch.bfloat16).cuda()
+    Bst = (torch.randn(N, K, generator=g) * 0.4).to(torch.bfloat16).cuda()
+    ref = A.float() @ Bst.float().t()
+    for it in range(10):
+        C = hip_ops.gemm_bf16(A, Bst, None, M, N, K, 0, 0, 0, False)
+        assert torch.allclose(C, ref, atol=2e-2 * K ** 0.5, rtol=1e-2), \
+            (it, (C - ref).abs().max())
