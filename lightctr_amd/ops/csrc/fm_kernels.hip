@@ -257,12 +257,12 @@ __global__ void fm_sorted_apply_kernel(
     }
   };
 
-  for (int e = sb; e < se; e += 4) {
-    const int nvalid = min(4, se - e);
-    float v[4], vw[4];
-    int f[4];
+  for (int e = sb; e < se; e += 8) {
+    const int nvalid = min(8, se - e);
+    float v[8], vw[8];
+    int f[8];
 #pragma unroll
-    for (int u = 0; u < 4; ++u) {
+    for (int u = 0; u < 8; ++u) {
       if (u < nvalid) {
         f[u] = sorted_fids[e + u];
         const long p = perm[e + u];
@@ -271,7 +271,7 @@ __global__ void fm_sorted_apply_kernel(
       }
     }
 #pragma unroll
-    for (int u = 0; u < 4; ++u) {
+    for (int u = 0; u < 8; ++u) {
       if (u >= nvalid) break;
       if (f[u] != cur_fid) {
         flush(e + u);
