@@ -81,12 +81,13 @@ def test_voting_and_adaboost():
 
 def test_watchdog_detects_stale_rank():
     events = []
-    wd = Watchdog(2, soft_s=0.1, dead_s=0.25, period_s=0.05,
+    # generous margins so a briefly-starved CI box cannot flake the test
+    wd = Watchdog(2, soft_s=0.5, dead_s=1.0, period_s=0.1,
                   on_soft=lambda r: events.append(("soft", r)),
                   on_dead=lambda r: events.append(("dead", r))).start()
-    for _ in range(8):
+    for _ in range(16):
         wd.heartbeat(0)
-        time.sleep(0.05)
+        time.sleep(0.1)
     wd.stop()
     states = wd.snapshot()
     assert states[0] == "alive"
