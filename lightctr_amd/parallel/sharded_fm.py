@@ -175,6 +175,150 @@ class ShardedFMModel:
                                          h.lr, h.eps, h.l2)
         return loss
 
+    # ---- SSP-1 pipelined stepping ------------------------------------
+    # The reference's async PS tolerates bounded staleness (SSP); the
+    # pipelined flavor prefetches batch t+1's parameter working set while
+    # step t is still computing/pushing, so the value all-to-all overlaps
+    # compute at the cost of params being one optimizer step stale
+    # (staleness 1 << the reference's threshold of 10). Enable by calling
+    # prefetch() for the next batch before finishing the current step, or
+    # just call train_step_pipelined per batch.
+
+    def _pull_issue(self, fids):
+        """Dedup + routing + issue the value exchanges with async_op; the
+        returned state is finished by _pull_wait."""
+        world = self.world
+        uniq, inverse = torch.unique(fids, return_inverse=True)
+        owner = uniq.long() % world
+        order = torch.argsort(owner, stable=True)
+        uniq_o = uniq[order]
+        cnt_t = torch.bincount(owner, minlength=world).to(self.device)
+        send_counts = cnt_t.cpu().tolist()
+        recv_cnt_t = torch.empty(world, dtype=cnt_t.dtype,
+                                 device=self.device)
+        dist.all_to_all_single(recv_cnt_t, cnt_t, group=self.group)
+        recv_counts = recv_cnt_t.cpu().tolist()
+        req = self._exchange(uniq_o, send_counts, recv_counts)
+        lidx = req.long() // world
+        Wv = self.W[lidx].contiguous()
+        Vv = self.V[lidx].contiguous()
+        if self.wire == "fp16":
+            Wv = Wv.to(torch.float16)
+            Vv = Vv.to(torch.float16)
+        Wl_o = torch.empty((sum(send_counts),), dtype=Wv.dtype,
+                           device=self.device)
+        Vl_o = torch.empty((sum(send_counts), self.h.k), dtype=Vv.dtype,
+                           device=self.device)
+        hW = dist.all_to_all_single(Wl_o, Wv,
+                                    output_split_sizes=send_counts,
+                                    input_split_sizes=recv_counts,
+                                    group=self.group, async_op=True)
+        hV = dist.all_to_all_single(Vl_o, Vv,
+                                    output_split_sizes=send_counts,
+                                    input_split_sizes=recv_counts,
+                                    group=self.group, async_op=True)
+        return {"uniq": uniq, "inverse": inverse, "order": order,
+                "lidx": lidx, "send_counts": send_counts,
+                "recv_counts": recv_counts, "Wl_o": Wl_o, "Vl_o": Vl_o,
+                "hW": hW, "hV": hV}
+
+    def _pull_wait(self, st):
+        st["hW"].wait()
+        st["hV"].wait()
+        U = st["uniq"].numel()
+        Wl = torch.empty(U, device=self.device)
+        Vl = torch.empty(U, self.h.k, device=self.device)
+        Wl[st["order"]] = st["Wl_o"].float()
+        Vl[st["order"]] = st["Vl_o"].float()
+        return Wl, Vl
+
+    def prefetch(self, fids):
+        """Issue the parameter pull for an upcoming batch (params as of the
+        LAST completed optimizer step — SSP staleness 1)."""
+        self._prefetched = self._pull_issue(fids)
+
+    def train_step_pipelined(self, row_ptr, fids, vals, labels,
+                             next_fids=None):
+        """Step with prefetched params; issues next_fids' pull right after
+        this step's local compute so the value exchange overlaps the grad
+        routing + owner apply."""
+        h = self.h
+        B = row_ptr.numel() - 1
+        st = getattr(self, "_prefetched", None)
+        if st is None:
+            st = self._pull_issue(fids)
+        self._prefetched = None
+        Wl, Vl = self._pull_wait(st)
+        scale = 1.0 / (B * self.world)
+        fids_local = st["inverse"].to(torch.int32)
+        U = st["uniq"].numel()
+        if self._use_hip:
+            ops = require_hip_ops()
+            pred, sumVX = ops.fm_forward(row_ptr, fids_local, vals, Wl, Vl)
+            loss, dpred = ops.logloss_grad(pred, labels, scale)
+            gw, gv = ops.fm_backward_emit(row_ptr, fids_local, vals, Vl,
+                                          sumVX, dpred)
+            sorted_l, perm = torch.sort(fids_local)
+            gWl = torch.zeros(U, device=self.device)
+            gVl = torch.zeros(U, h.k, device=self.device)
+            scratch = torch.zeros((U + 63) // 64, dtype=torch.int64,
+                                  device=self.device)
+            ops.fm_sorted_apply(sorted_l, perm, gw, gv, gWl, gVl, scratch)
+        else:
+            pred, sumVX = fm_ref.fm_forward_ref(row_ptr, fids_local, vals,
+                                                Wl, Vl)
+            loss, dpred = fm_ref.logloss_grad_ref(pred, labels, scale)
+            gWl, gVl = fm_ref.fm_backward_ref(row_ptr, fids_local, vals, Vl,
+                                              sumVX, dpred)
+        if next_fids is not None:
+            # overlap: next batch's pull (stale by this step's update)
+            self.prefetch(next_fids)
+        self._route_and_apply(st, gWl, gVl)
+        return loss
+
+    def _route_and_apply(self, st, gWl, gVl):
+        h = self.h
+        gW_recv = self._exchange(gWl[st["order"]], st["send_counts"],
+                                 st["recv_counts"])
+        gV_recv = self._exchange(gVl[st["order"]], st["send_counts"],
+                                 st["recv_counts"])
+        lidx = st["lidx"]
+        lidx32 = lidx.to(torch.int32)
+        if self._use_hip:
+            ops = require_hip_ops()
+            sorted_own, perm_own = torch.sort(lidx32)
+            ops.fm_sorted_apply(sorted_own, perm_own, gW_recv.contiguous(),
+                                gV_recv.contiguous(), self.gradW, self.gradV,
+                                self.touched)
+            self.count.zero_()
+            ops.bitmap_compact(self.touched, self.uniq, self.count)
+            if h.optimizer == "ftrl":
+                ops.fm_ftrl_apply(self.uniq, self.count, self.W, self.V,
+                                  self.zW, self.nW, self.zV, self.nV,
+                                  self.gradW, self.gradV, h.ftrl_alpha,
+                                  h.ftrl_beta, h.ftrl_l1, h.ftrl_l2,
+                                  1 if h.ftrl_v == "adagrad" else 0, h.lr,
+                                  h.eps, h.l2)
+            else:
+                ops.fm_adagrad_apply(self.uniq, self.count, self.W, self.V,
+                                     self.nW, self.nV, self.gradW,
+                                     self.gradV, h.lr, h.eps, h.l2)
+        else:
+            self.gradW.index_add_(0, lidx, gW_recv)
+            self.gradV.index_add_(0, lidx, gV_recv)
+            own_uniq = torch.unique(lidx).int()
+            if h.optimizer == "ftrl":
+                fm_ref.ftrl_apply_ref(own_uniq, self.W, self.V, self.zW,
+                                      self.nW, self.zV, self.nV, self.gradW,
+                                      self.gradV, h.ftrl_alpha, h.ftrl_beta,
+                                      h.ftrl_l1, h.ftrl_l2,
+                                      v_adagrad=h.ftrl_v == "adagrad",
+                                      v_lr=h.lr, v_eps=h.eps, v_l2=h.l2)
+            else:
+                fm_ref.adagrad_apply_ref(own_uniq, self.W, self.V, self.nW,
+                                         self.nV, self.gradW, self.gradV,
+                                         h.lr, h.eps, h.l2)
+
     # ---- inference (pull-only) ----
     def predict_proba(self, row_ptr, fids, vals):
         world = self.world

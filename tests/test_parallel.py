@@ -469,3 +469,60 @@ def _sharded_fm_fp16_worker(rank, port, q):
 def test_sharded_fm_fp16_wire_trains():
     (ok,) = _run_spawn(_sharded_fm_fp16_worker, 29538)
     assert ok
+
+
+def _sharded_fm_pipelined_worker(rank, port, q):
+    try:
+        dist = _init(rank, port)
+        from lightctr_amd.models.fm import FMHyper
+        from lightctr_amd.parallel.sharded_fm import ShardedFMModel
+        from conftest import make_random_csr
+
+        h = FMHyper(num_features=1000, k=8, optimizer="adagrad", seed=99)
+        sync = ShardedFMModel(h, device="cpu")
+        pipe = ShardedFMModel(h, device="cpu")       # same seed -> same init
+        over = ShardedFMModel(h, device="cpu")       # SSP-1 overlap flavor
+
+        batches = [make_random_csr(B=64, F_total=1000, seed=step * 7 + rank,
+                                   binary_vals=False) for step in range(3)]
+
+        # pipelined WITHOUT overlap (pull issued at step start) must equal
+        # the synchronous step bit-for-bit across multiple steps
+        sync_losses, pipe_losses = [], []
+        for rp, fi, v, lb in batches:
+            sync_losses.append(float(sync.train_step(rp, fi, v, lb).sum()))
+        for rp, fi, v, lb in batches:
+            pipe_losses.append(
+                float(pipe.train_step_pipelined(rp, fi, v, lb).sum()))
+        ok_eq = (torch.allclose(sync.W, pipe.W, atol=1e-6)
+                 and torch.allclose(sync.V, pipe.V, atol=1e-6))
+
+        # SSP-1 overlap: prefetch batch t+1 before batch t's apply. Step 1
+        # trains on identical params -> identical loss; later steps use
+        # params stale by one update but must stay finite and train.
+        over_losses = []
+        for step, (rp, fi, v, lb) in enumerate(batches):
+            nxt = batches[step + 1][1] if step + 1 < len(batches) else None
+            over_losses.append(
+                float(over.train_step_pipelined(rp, fi, v, lb,
+                                                next_fids=nxt).sum()))
+        ok_first = abs(over_losses[0] - sync_losses[0]) < 1e-6
+        ok_finite = all(torch.isfinite(torch.tensor(over_losses)))
+
+        if rank == 0:
+            q.put(("result", bool(ok_eq), bool(ok_first), bool(ok_finite),
+                   sync_losses, pipe_losses, over_losses))
+        dist.destroy_process_group()
+    except Exception:  # pragma: no cover
+        import traceback
+
+        q.put(("error", rank, traceback.format_exc()))
+        raise
+
+
+def test_sharded_fm_pipelined():
+    ok_eq, ok_first, ok_finite, sl, pl, ol = _run_spawn(
+        _sharded_fm_pipelined_worker, 29539)
+    assert ok_eq, f"pipelined(no overlap) != sync: {sl} vs {pl}"
+    assert ok_first, f"SSP-1 first step differs: {sl[0]} vs {ol[0]}"
+    assert ok_finite, f"SSP-1 losses not finite: {ol}"
