@@ -14,7 +14,7 @@ from dataclasses import dataclass
 import torch
 
 from ..ops import ffm_ref, fm_ref
-from ..ops._extension import require_hip_ops
+from ..ops._extension import require_hip_ops, sort_ids
 from ..utils.metrics import auc_score
 
 
@@ -108,14 +108,14 @@ class FFMModel:
                 gw, gblocks = ops.ffm_block_emit(row_of_entry, row_ptr,
                                                  fields, fids, vals, self.V,
                                                  dpred)
-                sorted_fids, perm = torch.sort(fids)
+                sorted_fids, perm = sort_ids(fids, self.h.num_features)
                 ops.ffm_blocks_apply(sorted_fids, perm, gblocks, gw,
                                      self.gradW,
                                      self.gradV.view(self.h.num_features,
                                                      -1),
                                      self.touched)
             elif self.backward_mode == "sorted":
-                sorted_fids, perm = torch.sort(fids)
+                sorted_fids, perm = sort_ids(fids, self.h.num_features)
                 row_of_entry = ops.row_index(row_ptr, fids.numel())
                 ops.ffm_sorted_backward(sorted_fids, perm, row_of_entry,
                                         row_ptr, fields, fids, vals, self.V,

@@ -22,7 +22,7 @@ from dataclasses import dataclass
 import torch
 
 from ..ops import fm_ref
-from ..ops._extension import require_hip_ops
+from ..ops._extension import require_hip_ops, sort_ids
 from ..utils.metrics import auc_score
 
 
@@ -113,7 +113,7 @@ class FMModel:
                 # profiles/r01_fm_atomic_backward.txt for why)
                 gw, gv = ops.fm_backward_emit(row_ptr, fids, vals, self.V,
                                               sumVX, dpred)
-                sorted_fids, perm = torch.sort(fids)
+                sorted_fids, perm = sort_ids(fids, self.h.num_features)
                 if self.fused_apply and not (
                         self.h.optimizer == "ftrl"
                         and self.h.ftrl_v == "adagrad"):

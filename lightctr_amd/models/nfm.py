@@ -14,7 +14,7 @@ from dataclasses import dataclass
 import torch
 
 from ..ops import fm_ref
-from ..ops._extension import require_hip_ops
+from ..ops._extension import require_hip_ops, sort_ids
 from ..utils.metrics import auc_score
 from .mlp import MLP
 
@@ -118,7 +118,7 @@ class NFMModel:
             ddeep = self.mlp.backward(dpred.unsqueeze(1))  # [B,K] fp32
             gw, gv = ops.nfm_backward_emit(row_ptr, fids, vals, self.V, sumVX,
                                            ddeep.contiguous(), dpred)
-            sorted_fids, perm = torch.sort(fids)
+            sorted_fids, perm = sort_ids(fids, self.h.num_features)
             ops.fm_sorted_apply(sorted_fids, perm, gw, gv, self.gradW,
                                 self.gradV, self.touched)
             self.count.zero_()

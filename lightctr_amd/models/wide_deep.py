@@ -15,7 +15,7 @@ from dataclasses import dataclass
 import torch
 
 from ..ops import fm_ref
-from ..ops._extension import require_hip_ops
+from ..ops._extension import require_hip_ops, sort_ids
 from ..utils.metrics import auc_score
 from .mlp import MLP
 
@@ -121,7 +121,7 @@ class WideDeepModel:
             gw, gv = ops.embed_backward_emit(row_ptr, vals,
                                              dDeep.contiguous(), dpred,
                                              self.h.num_fields, self.h.k)
-            sorted_fids, perm = torch.sort(fids)
+            sorted_fids, perm = sort_ids(fids, self.h.num_features)
             ops.fm_sorted_apply(sorted_fids, perm, gw, gv, self.gradW,
                                 self.gradE, self.touched)
             self.count.zero_()

@@ -387,6 +387,26 @@ std::vector<at::Tensor> act_backward(at::Tensor dY, at::Tensor Y,
   return {dZ, dZbf};
 }
 
+// Bit-range radix sort: drop-in for `torch.sort(fids)` on nonnegative
+// int32 keys < 2^end_bit. Returns (sorted_keys int32, perm int64).
+std::vector<at::Tensor> radix_sort_index(at::Tensor keys, int64_t end_bit) {
+  check_cuda_i32(keys, "keys");
+  CHK(end_bit >= 1 && end_bit <= 32, "end_bit in [1,32]");
+  const int n = (int)keys.numel();
+  auto sorted = at::empty_like(keys);
+  auto perm = at::empty({n}, keys.options().dtype(at::kLong));
+  auto iota = at::empty({n}, keys.options().dtype(at::kLong));
+  lightctr::iota_i64_launch(iota.data_ptr<long>(), n, cur_stream());
+  const unsigned long bytes =
+      lightctr::radix_sort_pairs_i32_temp_bytes(n, (int)end_bit);
+  auto temp = at::empty({(long)bytes}, keys.options().dtype(at::kByte));
+  lightctr::radix_sort_pairs_i32_launch(
+      temp.data_ptr(), bytes, keys.data_ptr<int>(), sorted.data_ptr<int>(),
+      iota.data_ptr<long>(), perm.data_ptr<long>(), n, (int)end_bit,
+      cur_stream());
+  return {sorted, perm};
+}
+
 at::Tensor colsum(at::Tensor dZ) {
   check_cuda_f32(dZ, "dZ");
   const int M = (int)dZ.size(0), N = (int)dZ.size(1);
@@ -704,6 +724,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_bf16_full", &gemm_bf16_full,
         "MFMA bf16 GEMM returning (C fp32, C bf16)");
   m.def("act_backward", &act_backward, "dZ = dY*act'(Y), + bf16 mirror");
+  m.def("radix_sort_index", &radix_sort_index,
+        "bit-range radix sort -> (sorted, perm)");
   m.def("colsum", &colsum, "bias gradient column sum");
   m.def("to_bf16", &to_bf16, "f32 -> bf16 convert");
   m.def("dense_adam", &dense_adam, "dense Adam + bf16 mirror refresh");

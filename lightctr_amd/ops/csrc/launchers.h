@@ -154,6 +154,14 @@ void lowbit_encode_launch(const float* x, unsigned int* words, float thresh,
 void lowbit_decode_launch(const unsigned int* words, float* x, float lo,
                           float hi, int bits, long n, ihipStream_t* stream);
 
+// --- sort_kernels.hip ---
+void iota_i64_launch(long* out, int n, ihipStream_t* stream);
+unsigned long radix_sort_pairs_i32_temp_bytes(int n, int end_bit);
+void radix_sort_pairs_i32_launch(void* temp, unsigned long temp_bytes,
+                                 const int* keys_in, int* keys_out,
+                                 const long* vals_in, long* vals_out, int n,
+                                 int end_bit, ihipStream_t* stream);
+
 // --- gbm_kernels.hip ---
 void gbm_hist_launch(const unsigned char* bins, const float* grad,
                      const float* hess, const int* node_of_row, float* hist,

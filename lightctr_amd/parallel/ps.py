@@ -355,7 +355,7 @@ def ps_train_fm(cfg: PSConfig, group, gen_batch, steps: int,
     distributed_algo_abst.h:130-280). gen_batch(step) -> CSR batch.
     Returns list of per-step mean losses."""
     from ..ops import fm_ref
-    from ..ops._extension import has_hip_ops, require_hip_ops
+    from ..ops._extension import has_hip_ops, require_hip_ops, sort_ids
 
     worker = PSWorker(cfg, group, device=device)
     losses = []
@@ -378,7 +378,7 @@ def ps_train_fm(cfg: PSConfig, group, gen_batch, steps: int,
             gVl = torch.zeros(U, cfg.k, device=device)
             bitmap = torch.zeros((U + 63) // 64, dtype=torch.int64,
                                  device=device)
-            sorted_l, perm = torch.sort(fids_local)
+            sorted_l, perm = sort_ids(fids_local, U)
             ops.fm_sorted_apply(sorted_l, perm, gw, gv, gWl, gVl, bitmap)
         else:
             pred, sumVX = fm_ref.fm_forward_ref(row_ptr, fids_local, vals,

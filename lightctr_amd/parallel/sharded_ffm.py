@@ -18,7 +18,7 @@ import torch.distributed as dist
 
 from ..models.ffm import FFMHyper
 from ..ops import ffm_ref, fm_ref
-from ..ops._extension import require_hip_ops
+from ..ops._extension import require_hip_ops, sort_ids
 
 
 class ShardedFFMModel:
@@ -103,7 +103,7 @@ class ShardedFFMModel:
             gVl = torch.zeros(U, nf, K, device=self.device)
             scratch = torch.zeros((U + 63) // 64, dtype=torch.int64,
                                   device=self.device)
-            sorted_l, perm = torch.sort(fids_local)
+            sorted_l, perm = sort_ids(fids_local, U)
             row_of_entry = ops.row_index(row_ptr, fids.numel())
             ops.ffm_sorted_backward(sorted_l, perm, row_of_entry, row_ptr,
                                     fields, fids_local, vals, Vl, dpred,
@@ -121,7 +121,7 @@ class ShardedFFMModel:
         lidx32 = lidx.to(torch.int32)
         if self._use_hip:
             ops = require_hip_ops()
-            sorted_own, perm_own = torch.sort(lidx32)
+            sorted_own, perm_own = sort_ids(lidx32, self.F_local)
             ops.fm_sorted_apply(sorted_own, perm_own, gW_recv.contiguous(),
                                 gV_recv.contiguous(),
                                 self.gradW, self.gradV.view(self.F_local, -1),

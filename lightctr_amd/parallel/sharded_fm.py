@@ -22,7 +22,7 @@ import torch.distributed as dist
 
 from ..models.fm import FMHyper
 from ..ops import fm_ref
-from ..ops._extension import require_hip_ops
+from ..ops._extension import require_hip_ops, sort_ids
 
 
 class ShardedFMModel:
@@ -118,7 +118,7 @@ class ShardedFMModel:
             loss, dpred = ops.logloss_grad(pred, labels, scale)
             gw, gv = ops.fm_backward_emit(row_ptr, fids_local, vals, Vl,
                                           sumVX, dpred)
-            sorted_l, perm = torch.sort(fids_local)
+            sorted_l, perm = sort_ids(fids_local, U)
             gWl = torch.zeros(U, device=self.device)
             gVl = torch.zeros(U, h.k, device=self.device)
             scratch_bitmap = torch.zeros((U + 63) // 64, dtype=torch.int64,
@@ -141,7 +141,7 @@ class ShardedFMModel:
         lidx32 = lidx.to(torch.int32)
         if self._use_hip:
             ops = require_hip_ops()
-            sorted_own, perm_own = torch.sort(lidx32)
+            sorted_own, perm_own = sort_ids(lidx32, self.F_local)
             ops.fm_sorted_apply(sorted_own, perm_own, gW_recv.contiguous(),
                                 gV_recv.contiguous(), self.gradW, self.gradV,
                                 self.touched)
@@ -258,7 +258,7 @@ class ShardedFMModel:
             loss, dpred = ops.logloss_grad(pred, labels, scale)
             gw, gv = ops.fm_backward_emit(row_ptr, fids_local, vals, Vl,
                                           sumVX, dpred)
-            sorted_l, perm = torch.sort(fids_local)
+            sorted_l, perm = sort_ids(fids_local, U)
             gWl = torch.zeros(U, device=self.device)
             gVl = torch.zeros(U, h.k, device=self.device)
             scratch = torch.zeros((U + 63) // 64, dtype=torch.int64,
@@ -286,7 +286,7 @@ class ShardedFMModel:
         lidx32 = lidx.to(torch.int32)
         if self._use_hip:
             ops = require_hip_ops()
-            sorted_own, perm_own = torch.sort(lidx32)
+            sorted_own, perm_own = sort_ids(lidx32, self.F_local)
             ops.fm_sorted_apply(sorted_own, perm_own, gW_recv.contiguous(),
                                 gV_recv.contiguous(), self.gradW, self.gradV,
                                 self.touched)

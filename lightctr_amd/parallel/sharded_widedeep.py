@@ -21,7 +21,7 @@ import torch.distributed as dist
 from ..models.mlp import MLP
 from ..models.wide_deep import WideDeepHyper
 from ..ops import fm_ref
-from ..ops._extension import require_hip_ops
+from ..ops._extension import require_hip_ops, sort_ids
 from .ring import _flatten_into, _unflatten_from
 
 
@@ -120,7 +120,7 @@ class ShardedWideDeepModel:
             gw, gv = ops.embed_backward_emit(row_ptr, vals,
                                              dDeep.contiguous(), dpred,
                                              h.num_fields, h.k)
-            sorted_l, perm = torch.sort(fids_local)
+            sorted_l, perm = sort_ids(fids_local, U)
             gWl = torch.zeros(U, device=self.device)
             gEl = torch.zeros(U, h.k, device=self.device)
             scratch = torch.zeros((U + 63) // 64, dtype=torch.int64,
@@ -164,7 +164,7 @@ class ShardedWideDeepModel:
         lidx32 = lidx.to(torch.int32)
         if self._use_hip:
             ops = require_hip_ops()
-            sorted_own, perm_own = torch.sort(lidx32)
+            sorted_own, perm_own = sort_ids(lidx32, self.F_local)
             ops.fm_sorted_apply(sorted_own, perm_own, gW_recv.contiguous(),
                                 gE_recv.contiguous(), self.gradW, self.gradE,
                                 self.touched)

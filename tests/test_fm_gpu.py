@@ -262,3 +262,22 @@ def test_fm_step_fuzz_parity(seed):
     assert torch.allclose(gpu.W.cpu(), cpu.W, atol=2e-4, rtol=1e-3), \
         (F, K, B, (gpu.W.cpu() - cpu.W).abs().max())
     assert torch.allclose(gpu.V.cpu(), cpu.V, atol=2e-4, rtol=1e-3)
+
+
+def test_radix_sort_index_parity():
+    """Bit-range rocPRIM sort == torch.sort on the id distributions the
+    backward actually sees (power-law fids, localized ids, tiny arrays)."""
+    from lightctr_amd.ops._extension import sort_ids
+
+    torch.manual_seed(7)
+    for upper, n in [(1 << 24, 2_500_000), (1000, 4096), (7, 64), (1, 1)]:
+        fids = torch.randint(0, upper, (n,), dtype=torch.int32,
+                             device="cuda")
+        s, p = sort_ids(fids, upper)
+        ref_s, ref_p = torch.sort(fids)
+        assert torch.equal(s, ref_s)
+        assert p.dtype == torch.int64
+        # perm must be a valid permutation mapping fids -> sorted order
+        assert torch.equal(fids[p], s)
+        assert torch.equal(torch.sort(p).values,
+                           torch.arange(n, device="cuda"))
