@@ -90,7 +90,8 @@ void fm_backward(at::Tensor row_ptr, at::Tensor fids, at::Tensor vals,
 
 std::vector<at::Tensor> fm_backward_emit(at::Tensor row_ptr, at::Tensor fids,
                                          at::Tensor vals, at::Tensor V,
-                                         at::Tensor sumVX, at::Tensor dpred) {
+                                         at::Tensor sumVX, at::Tensor dpred,
+                                         c10::optional<at::Tensor> pos) {
   check_cuda_i32(row_ptr, "row_ptr");
   check_cuda_i32(fids, "fids");
   const int B = (int)row_ptr.numel() - 1;
@@ -98,21 +99,44 @@ std::vector<at::Tensor> fm_backward_emit(at::Tensor row_ptr, at::Tensor fids,
   const auto nnz = fids.numel();
   auto gw = at::empty({nnz}, V.options());
   auto gv = at::empty({nnz, K}, V.options());
+  const int* pos_ptr = nullptr;
+  if (pos.has_value()) {
+    check_cuda_i32(*pos, "pos");
+    CHK(pos->numel() == nnz, "pos must have one slot per entry");
+    pos_ptr = pos->data_ptr<int>();
+  }
   lightctr::fm_backward_emit_launch(
       row_ptr.data_ptr<int>(), fids.data_ptr<int>(), vals.data_ptr<float>(),
       V.data_ptr<float>(), sumVX.data_ptr<float>(), dpred.data_ptr<float>(),
-      gw.data_ptr<float>(), gv.data_ptr<float>(), B, K, cur_stream());
+      gw.data_ptr<float>(), gv.data_ptr<float>(), B, K, pos_ptr,
+      cur_stream());
   return {gw, gv};
 }
 
-void fm_sorted_apply(at::Tensor sorted_fids, at::Tensor perm, at::Tensor gw,
-                     at::Tensor gv, at::Tensor gradW, at::Tensor gradV,
-                     at::Tensor touched) {
+// inverse permutation (write slots for scatter-emit): inv[perm[i]] = i
+at::Tensor inv_perm_i32(at::Tensor perm) {
+  CHK(perm.is_cuda() && perm.scalar_type() == at::kLong &&
+          perm.is_contiguous(),
+      "perm must be contiguous cuda int64");
+  const int n = (int)perm.numel();
+  auto inv = at::empty({n}, perm.options().dtype(at::kInt));
+  lightctr::inv_perm_launch(perm.data_ptr<long>(), inv.data_ptr<int>(), n,
+                            cur_stream());
+  return inv;
+}
+
+void fm_sorted_apply(at::Tensor sorted_fids, c10::optional<at::Tensor> perm,
+                     at::Tensor gw, at::Tensor gv, at::Tensor gradW,
+                     at::Tensor gradV, at::Tensor touched) {
   check_cuda_i32(sorted_fids, "sorted_fids");
-  CHK(perm.scalar_type() == at::kLong, "perm must be int64");
+  const long* perm_ptr = nullptr;
+  if (perm.has_value()) {
+    CHK(perm->scalar_type() == at::kLong, "perm must be int64");
+    perm_ptr = perm->data_ptr<long>();
+  }
   const int K = (int)gradV.size(1);
   lightctr::fm_sorted_apply_launch(
-      sorted_fids.data_ptr<int>(), perm.data_ptr<long>(),
+      sorted_fids.data_ptr<int>(), perm_ptr,
       gw.data_ptr<float>(), gv.data_ptr<float>(), gradW.data_ptr<float>(),
       gradV.data_ptr<float>(), (unsigned long long*)touched.data_ptr(),
       (int)sorted_fids.numel(), K, 0, nullptr, nullptr, nullptr, nullptr,
@@ -122,7 +146,8 @@ void fm_sorted_apply(at::Tensor sorted_fids, at::Tensor perm, at::Tensor gw,
 // Fused variant: interior feature segments get their optimizer update
 // applied during the segment reduction; only boundary-spanning features go
 // through the slab + bitmap (+ the follow-up sparse apply).
-void fm_sorted_apply_fused(at::Tensor sorted_fids, at::Tensor perm,
+void fm_sorted_apply_fused(at::Tensor sorted_fids,
+                           c10::optional<at::Tensor> perm,
                            at::Tensor gw, at::Tensor gv, at::Tensor gradW,
                            at::Tensor gradV, at::Tensor touched,
                            at::Tensor W, at::Tensor V, at::Tensor nW,
@@ -130,11 +155,15 @@ void fm_sorted_apply_fused(at::Tensor sorted_fids, at::Tensor perm,
                            c10::optional<at::Tensor> zV, int64_t opt_mode,
                            double p0, double p1, double p2, double p3) {
   check_cuda_i32(sorted_fids, "sorted_fids");
-  CHK(perm.scalar_type() == at::kLong, "perm must be int64");
+  const long* perm_ptr = nullptr;
+  if (perm.has_value()) {
+    CHK(perm->scalar_type() == at::kLong, "perm must be int64");
+    perm_ptr = perm->data_ptr<long>();
+  }
   CHK(opt_mode == 1 || opt_mode == 2, "opt_mode 1=adagrad 2=ftrl");
   const int K = (int)gradV.size(1);
   lightctr::fm_sorted_apply_launch(
-      sorted_fids.data_ptr<int>(), perm.data_ptr<long>(),
+      sorted_fids.data_ptr<int>(), perm_ptr,
       gw.data_ptr<float>(), gv.data_ptr<float>(), gradW.data_ptr<float>(),
       gradV.data_ptr<float>(), (unsigned long long*)touched.data_ptr(),
       (int)sorted_fids.numel(), K, (int)opt_mode, V.data_ptr<float>(),
@@ -692,9 +721,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("logloss_grad", &logloss_grad, "stable logloss + dpred");
   m.def("fm_backward", &fm_backward, "FM fused backward scatter");
   m.def("fm_backward_emit", &fm_backward_emit,
-        "FM backward phase 1: per-entry grads (no atomics)");
+        "FM backward phase 1: per-entry grads (no atomics); pos=write slots",
+        py::arg("row_ptr"), py::arg("fids"), py::arg("vals"), py::arg("V"),
+        py::arg("sumVX"), py::arg("dpred"), py::arg("pos") = py::none());
   m.def("fm_sorted_apply", &fm_sorted_apply,
-        "FM backward phase 2: segment-reduce sorted grads into slabs");
+        "FM backward phase 2: segment-reduce sorted grads into slabs "
+        "(perm=None -> grads already in sorted order)",
+        py::arg("sorted_fids"), py::arg("perm"), py::arg("gw"),
+        py::arg("gv"), py::arg("gradW"), py::arg("gradV"),
+        py::arg("touched"));
   m.def("fm_sorted_apply_fused", &fm_sorted_apply_fused,
         "segment-reduce + fused optimizer for interior segments");
   m.def("ffm_forward", &ffm_forward, "FFM fused pairwise forward");
@@ -724,6 +759,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_bf16_full", &gemm_bf16_full,
         "MFMA bf16 GEMM returning (C fp32, C bf16)");
   m.def("act_backward", &act_backward, "dZ = dY*act'(Y), + bf16 mirror");
+  m.def("inv_perm_i32", &inv_perm_i32, "inverse permutation (int32)");
   m.def("radix_sort_index", &radix_sort_index,
         "bit-range radix sort -> (sorted, perm)");
   m.def("colsum", &colsum, "bias gradient column sum");

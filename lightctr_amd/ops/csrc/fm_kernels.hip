@@ -149,12 +149,19 @@ __global__ void fm_backward_kernel(
 // (measured 2.76 ms/step vs ~0.1 ms total for everything else; see
 // profiles/r01_fm_atomic_backward.txt).
 // ---------------------------------------------------------------------------
+// `pos` (optional): per-entry write slot. When given, entry j's gradient
+// lands at gv[pos[j]] / gw[pos[j]] — i.e. directly in SORTED order (pos =
+// inverse of the sort permutation). That moves the randomness of the
+// sorted pipeline from the apply kernel's dependent 64 B GATHERS (latency
+// bound, measured 285 us/step) onto this kernel's independent STORES
+// (fire-and-forget through L2), and lets the apply read gv sequentially.
 template <int K>
 __global__ void fm_backward_emit_kernel(
     const int* __restrict__ row_ptr, const int* __restrict__ fids,
     const float* __restrict__ vals, const float* __restrict__ V,
     const float* __restrict__ sumVX, const float* __restrict__ dpred,
-    float* __restrict__ gw, float* __restrict__ gv, int B) {
+    float* __restrict__ gw, float* __restrict__ gv, int B,
+    const int* __restrict__ pos) {
   constexpr int G = LCTR_WAVE / K;
   const int lane = threadIdx.x & (LCTR_WAVE - 1);
   const int row = blockIdx.x * (blockDim.x / LCTR_WAVE) + (threadIdx.x >> 6);
@@ -167,9 +174,18 @@ __global__ void fm_backward_emit_kernel(
   for (int j = beg + g; j < end; j += G) {
     const int fid = fids[j];
     const float x = vals[j];
-    gv[(size_t)j * K + k] = d * (sv - V[(size_t)fid * K + k] * x) * x;
-    if (k == 0) gw[j] = d * x;
+    const int slot = pos ? pos[j] : j;
+    gv[(size_t)slot * K + k] = d * (sv - V[(size_t)fid * K + k] * x) * x;
+    if (k == 0) gw[slot] = d * x;
   }
+}
+
+// inv[perm[i]] = i : build the emit kernel's write-slot array from the
+// sort permutation (sequential reads of perm, scattered int32 stores).
+__global__ void inv_perm_kernel(const long* __restrict__ perm,
+                                int* __restrict__ inv, int n) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) inv[perm[i]] = i;
 }
 
 // Sorted backward, phase 2: segment-reduce sorted per-entry grads into the
@@ -265,7 +281,9 @@ __global__ void fm_sorted_apply_kernel(
     for (int u = 0; u < 8; ++u) {
       if (u < nvalid) {
         f[u] = sorted_fids[e + u];
-        const long p = perm[e + u];
+        // perm == nullptr: gw/gv were emitted directly in sorted order
+        // (scatter-emit path) — sequential reads, no gather
+        const long p = perm ? perm[e + u] : (long)(e + u);
         v[u] = gv[(size_t)p * K + k];
         vw[u] = (k == 0) ? gw[p] : 0.f;
       }
@@ -429,14 +447,22 @@ void fm_backward_launch(const int* row_ptr, const int* fids, const float* vals,
 void fm_backward_emit_launch(const int* row_ptr, const int* fids,
                              const float* vals, const float* V,
                              const float* sumVX, const float* dpred, float* gw,
-                             float* gv, int B, int K, hipStream_t stream) {
+                             float* gv, int B, int K, const int* pos,
+                             hipStream_t stream) {
   if (B <= 0) return;
   const int wpb = waves_per_block();
   dim3 block(wpb * LCTR_WAVE);
   dim3 grid((B + wpb - 1) / wpb);
   DISPATCH_K(K, hipLaunchKernelGGL((fm_backward_emit_kernel<KC>), grid, block,
                                    0, stream, row_ptr, fids, vals, V, sumVX,
-                                   dpred, gw, gv, B));
+                                   dpred, gw, gv, B, pos));
+}
+
+void inv_perm_launch(const long* perm, int* inv, int n, hipStream_t stream) {
+  if (n <= 0) return;
+  dim3 block(256);
+  dim3 grid((n + 255) / 256);
+  hipLaunchKernelGGL(inv_perm_kernel, grid, block, 0, stream, perm, inv, n);
 }
 
 void fm_sorted_apply_launch(const int* sorted_fids, const long* perm,

@@ -21,7 +21,9 @@ void fm_backward_launch(const int* row_ptr, const int* fids, const float* vals,
 void fm_backward_emit_launch(const int* row_ptr, const int* fids,
                              const float* vals, const float* V,
                              const float* sumVX, const float* dpred, float* gw,
-                             float* gv, int B, int K, ihipStream_t* stream);
+                             float* gv, int B, int K, const int* pos,
+                             ihipStream_t* stream);
+void inv_perm_launch(const long* perm, int* inv, int n, ihipStream_t* stream);
 void fm_sorted_apply_launch(const int* sorted_fids, const long* perm,
                             const float* gw, const float* gv, float* gradW,
                             float* gradV, unsigned long long* touched, int nnz,
