@@ -114,27 +114,26 @@ class FMModel:
                 # sequential reads (see profiles/r01_fm_atomic_backward.txt
                 # and fm_kernels.hip for the evolution)
                 sorted_fids, perm = sort_ids(fids, self.h.num_features)
-                pos = ops.inv_perm_i32(perm)
                 gw, gv = ops.fm_backward_emit(row_ptr, fids, vals, self.V,
-                                              sumVX, dpred, pos)
+                                              sumVX, dpred)
                 if self.fused_apply and not (
                         self.h.optimizer == "ftrl"
                         and self.h.ftrl_v == "adagrad"):
                     if self.h.optimizer == "ftrl":
                         ops.fm_sorted_apply_fused(
-                            sorted_fids, None, gw, gv, self.gradW,
+                            sorted_fids, perm, gw, gv, self.gradW,
                             self.gradV, self.touched, self.W, self.V,
                             self.nW, self.nV, self.zW, self.zV, 2,
                             self.h.ftrl_alpha, self.h.ftrl_beta,
                             self.h.ftrl_l1, self.h.ftrl_l2)
                     else:
                         ops.fm_sorted_apply_fused(
-                            sorted_fids, None, gw, gv, self.gradW,
+                            sorted_fids, perm, gw, gv, self.gradW,
                             self.gradV, self.touched, self.W, self.V,
                             self.nW, self.nV, None, None, 1, self.h.lr,
                             self.h.eps, self.h.l2, 0.0)
                 else:
-                    ops.fm_sorted_apply(sorted_fids, None, gw, gv,
+                    ops.fm_sorted_apply(sorted_fids, perm, gw, gv,
                                         self.gradW, self.gradV,
                                         self.touched)
             else:

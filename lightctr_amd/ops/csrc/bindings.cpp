@@ -127,7 +127,7 @@ at::Tensor inv_perm_i32(at::Tensor perm) {
 
 void fm_sorted_apply(at::Tensor sorted_fids, c10::optional<at::Tensor> perm,
                      at::Tensor gw, at::Tensor gv, at::Tensor gradW,
-                     at::Tensor gradV, at::Tensor touched) {
+                     at::Tensor gradV, at::Tensor touched, int64_t chunk) {
   check_cuda_i32(sorted_fids, "sorted_fids");
   const long* perm_ptr = nullptr;
   if (perm.has_value()) {
@@ -140,7 +140,7 @@ void fm_sorted_apply(at::Tensor sorted_fids, c10::optional<at::Tensor> perm,
       gw.data_ptr<float>(), gv.data_ptr<float>(), gradW.data_ptr<float>(),
       gradV.data_ptr<float>(), (unsigned long long*)touched.data_ptr(),
       (int)sorted_fids.numel(), K, 0, nullptr, nullptr, nullptr, nullptr,
-      nullptr, nullptr, 0.f, 0.f, 0.f, 0.f, cur_stream());
+      nullptr, nullptr, 0.f, 0.f, 0.f, 0.f, (int)chunk, cur_stream());
 }
 
 // Fused variant: interior feature segments get their optimizer update
@@ -171,7 +171,7 @@ void fm_sorted_apply_fused(at::Tensor sorted_fids,
       zW.has_value() ? zW->data_ptr<float>() : nullptr,
       nV.data_ptr<float>(),
       zV.has_value() ? zV->data_ptr<float>() : nullptr, (float)p0,
-      (float)p1, (float)p2, (float)p3, cur_stream());
+      (float)p1, (float)p2, (float)p3, 384, cur_stream());
 }
 
 // ---- FFM ----
@@ -729,7 +729,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "(perm=None -> grads already in sorted order)",
         py::arg("sorted_fids"), py::arg("perm"), py::arg("gw"),
         py::arg("gv"), py::arg("gradW"), py::arg("gradV"),
-        py::arg("touched"));
+        py::arg("touched"), py::arg("chunk") = 384);
   m.def("fm_sorted_apply_fused", &fm_sorted_apply_fused,
         "segment-reduce + fused optimizer for interior segments");
   m.def("ffm_forward", &ffm_forward, "FFM fused pairwise forward");

@@ -372,15 +372,14 @@ def ps_train_fm(cfg: PSConfig, group, gen_batch, steps: int,
             pred, sumVX = ops.fm_forward(row_ptr, fids_local, vals, Wl, Vl)
             loss, dpred = ops.logloss_grad(pred, labels, scale)
             U = uniq.numel()
-            sorted_l, perm = sort_ids(fids_local, U)
-            pos = ops.inv_perm_i32(perm)
             gw, gv = ops.fm_backward_emit(row_ptr, fids_local, vals, Vl,
-                                          sumVX, dpred, pos)
+                                          sumVX, dpred)
+            sorted_l, perm = sort_ids(fids_local, U)
             gWl = torch.zeros(U, device=device)
             gVl = torch.zeros(U, cfg.k, device=device)
             bitmap = torch.zeros((U + 63) // 64, dtype=torch.int64,
                                  device=device)
-            ops.fm_sorted_apply(sorted_l, None, gw, gv, gWl, gVl, bitmap)
+            ops.fm_sorted_apply(sorted_l, perm, gw, gv, gWl, gVl, bitmap)
         else:
             pred, sumVX = fm_ref.fm_forward_ref(row_ptr, fids_local, vals,
                                                 Wl, Vl)

@@ -116,15 +116,14 @@ class ShardedFMModel:
             ops = require_hip_ops()
             pred, sumVX = ops.fm_forward(row_ptr, fids_local, vals, Wl, Vl)
             loss, dpred = ops.logloss_grad(pred, labels, scale)
-            sorted_l, perm = sort_ids(fids_local, U)
-            pos = ops.inv_perm_i32(perm)
             gw, gv = ops.fm_backward_emit(row_ptr, fids_local, vals, Vl,
-                                          sumVX, dpred, pos)
+                                          sumVX, dpred)
+            sorted_l, perm = sort_ids(fids_local, U)
             gWl = torch.zeros(U, device=self.device)
             gVl = torch.zeros(U, h.k, device=self.device)
             scratch_bitmap = torch.zeros((U + 63) // 64, dtype=torch.int64,
                                          device=self.device)
-            ops.fm_sorted_apply(sorted_l, None, gw, gv, gWl, gVl,
+            ops.fm_sorted_apply(sorted_l, perm, gw, gv, gWl, gVl,
                                 scratch_bitmap)
         else:
             pred, sumVX = fm_ref.fm_forward_ref(row_ptr, fids_local, vals,
@@ -257,16 +256,14 @@ class ShardedFMModel:
             ops = require_hip_ops()
             pred, sumVX = ops.fm_forward(row_ptr, fids_local, vals, Wl, Vl)
             loss, dpred = ops.logloss_grad(pred, labels, scale)
-            sorted_l, perm = sort_ids(fids_local, U)
-            pos = ops.inv_perm_i32(perm)
             gw, gv = ops.fm_backward_emit(row_ptr, fids_local, vals, Vl,
-                                          sumVX, dpred, pos)
+                                          sumVX, dpred)
+            sorted_l, perm = sort_ids(fids_local, U)
             gWl = torch.zeros(U, device=self.device)
             gVl = torch.zeros(U, h.k, device=self.device)
             scratch = torch.zeros((U + 63) // 64, dtype=torch.int64,
                                   device=self.device)
-            ops.fm_sorted_apply(sorted_l, None, gw, gv, gWl, gVl,
-                                scratch)
+            ops.fm_sorted_apply(sorted_l, perm, gw, gv, gWl, gVl, scratch)
         else:
             pred, sumVX = fm_ref.fm_forward_ref(row_ptr, fids_local, vals,
                                                 Wl, Vl)

@@ -281,3 +281,47 @@ def test_radix_sort_index_parity():
         assert torch.equal(fids[p], s)
         assert torch.equal(torch.sort(p).values,
                            torch.arange(n, device="cuda"))
+
+
+def test_scatter_emit_matches_gather_apply():
+    """Scatter-emit (pos=inverse perm, apply with perm=None) produces the
+    same slab gradients as the default emit+gather pipeline."""
+    from lightctr_amd.ops._extension import require_hip_ops, sort_ids
+
+    ops = require_hip_ops()
+    torch.manual_seed(3)
+    F, K, B = 5000, 16, 512
+    row_ptr, fids, vals, labels = make_random_csr(B=B, F_total=F,
+                                                  binary_vals=False, seed=3)
+    row_ptr, fids = row_ptr.cuda(), fids.cuda()
+    vals, labels = vals.cuda(), labels.cuda()
+    V = torch.randn(F, K, device="cuda") * 0.05
+    W = torch.randn(F, device="cuda") * 0.05
+    pred, sumVX = ops.fm_forward(row_ptr, fids, vals, W, V)
+    _, dpred = ops.logloss_grad(pred, labels, 1.0 / B)
+    sorted_fids, perm = sort_ids(fids, F)
+    nnz = fids.numel()
+
+    def run(scatter):
+        gradW = torch.zeros(F, device="cuda")
+        gradV = torch.zeros(F, K, device="cuda")
+        touched = torch.zeros((F + 63) // 64, dtype=torch.int64,
+                              device="cuda")
+        if scatter:
+            pos = ops.inv_perm_i32(perm)
+            gw, gv = ops.fm_backward_emit(row_ptr, fids, vals, V, sumVX,
+                                          dpred, pos)
+            ops.fm_sorted_apply(sorted_fids, None, gw, gv, gradW, gradV,
+                                touched)
+        else:
+            gw, gv = ops.fm_backward_emit(row_ptr, fids, vals, V, sumVX,
+                                          dpred)
+            ops.fm_sorted_apply(sorted_fids, perm, gw, gv, gradW, gradV,
+                                touched)
+        return gradW, gradV, touched
+
+    gW1, gV1, t1 = run(False)
+    gW2, gV2, t2 = run(True)
+    assert torch.allclose(gW1, gW2, atol=1e-5)
+    assert torch.allclose(gV1, gV2, atol=1e-5)
+    assert torch.equal(t1, t2)

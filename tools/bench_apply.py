@@ -1,0 +1,79 @@
+"""Micro-bench of the FM sorted-backward pipeline stages (GPU).
+
+Times each stage and the {gather-apply, scatter-emit} x chunk-size matrix
+to locate the real bound of fm_sorted_apply (hypothesis: hot-feature
+atomic flushes, one per (subgroup, run), not the gv gathers — larger
+chunks divide the flush count per hot fid).
+"""
+import sys
+import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
+import torch
+
+from lightctr_amd.data.synthetic import SyntheticCriteo
+from lightctr_amd.models.fm import FMHyper, FMModel
+from lightctr_amd.ops._extension import require_hip_ops, sort_ids
+
+
+def t(fn, reps=20):
+    for _ in range(3):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(reps):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / reps * 1e6  # us
+
+
+def main():
+    ops = require_hip_ops()
+    h = FMHyper(num_features=1 << 24, k=16, optimizer="ftrl")
+    m = FMModel(h, device="cuda")
+    gen = SyntheticCriteo(num_features=h.num_features, seed=5,
+                          device="cuda")
+    row_ptr, _fields, fids, vals, labels = gen.batch(65536)
+    nnz = fids.numel()
+    B = row_ptr.numel() - 1
+
+    pred, sumVX = ops.fm_forward(row_ptr, fids, vals, m.W, m.V)
+    loss, dpred = ops.logloss_grad(pred, labels, 1.0 / B)
+    print(f"nnz={nnz} uniq={torch.unique(fids).numel()}")
+
+    print(f"fwd        {t(lambda: ops.fm_forward(row_ptr, fids, vals, m.W, m.V)):8.1f} us")
+    print(f"sort       {t(lambda: sort_ids(fids, h.num_features)):8.1f} us")
+    sorted_fids, perm = sort_ids(fids, h.num_features)
+    print(f"inv_perm   {t(lambda: ops.inv_perm_i32(perm)):8.1f} us")
+    pos = ops.inv_perm_i32(perm)
+    print(f"emit seq   {t(lambda: ops.fm_backward_emit(row_ptr, fids, vals, m.V, sumVX, dpred)):8.1f} us")
+    print(f"emit scat  {t(lambda: ops.fm_backward_emit(row_ptr, fids, vals, m.V, sumVX, dpred, pos)):8.1f} us")
+
+    gw_s, gv_s = ops.fm_backward_emit(row_ptr, fids, vals, m.V, sumVX, dpred)
+    gw_p, gv_p = ops.fm_backward_emit(row_ptr, fids, vals, m.V, sumVX, dpred,
+                                      pos)
+    for chunk in (128, 256, 384, 512, 768, 1024):
+        a = t(lambda: ops.fm_sorted_apply(sorted_fids, perm, gw_s, gv_s,
+                                          m.gradW, m.gradV, m.touched,
+                                          chunk))
+        b = t(lambda: ops.fm_sorted_apply(sorted_fids, None, gw_p, gv_p,
+                                          m.gradW, m.gradV, m.touched,
+                                          chunk))
+        print(f"apply chunk={chunk:5d}  gather {a:8.1f} us   seq {b:8.1f} us")
+
+    # correctness cross-check: gather vs scatter paths agree
+    m.gradW.zero_(); m.gradV.zero_(); m.touched.zero_()
+    ops.fm_sorted_apply(sorted_fids, perm, gw_s, gv_s, m.gradW, m.gradV,
+                        m.touched, 256)
+    gW1, gV1 = m.gradW.clone(), m.gradV.clone()
+    m.gradW.zero_(); m.gradV.zero_(); m.touched.zero_()
+    ops.fm_sorted_apply(sorted_fids, None, gw_p, gv_p, m.gradW, m.gradV,
+                        m.touched, 2048)
+    dmax = (gV1 - m.gradV).abs().max().item()
+    print(f"gather-vs-scatter gradV maxdiff {dmax:.3e}")
+
+
+if __name__ == "__main__":
+    main()
