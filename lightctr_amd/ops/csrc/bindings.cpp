@@ -729,7 +729,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "(perm=None -> grads already in sorted order)",
         py::arg("sorted_fids"), py::arg("perm"), py::arg("gw"),
         py::arg("gv"), py::arg("gradW"), py::arg("gradV"),
-        py::arg("touched"), py::arg("chunk") = 384);
+        py::arg("touched"), py::arg("chunk") = 0);
   m.def("fm_sorted_apply_fused", &fm_sorted_apply_fused,
         "segment-reduce + fused optimizer for interior segments");
   m.def("ffm_forward", &ffm_forward, "FFM fused pairwise forward");

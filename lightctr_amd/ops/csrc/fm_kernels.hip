@@ -316,6 +316,118 @@ __global__ void fm_sorted_apply_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// Sorted backward, phase 2, segmented-scan variant (K <= 16): one LANE per
+// sorted entry, the entry's whole [K] gradient row + gw held in registers.
+// A wave covers 64 consecutive sorted entries; run sums are produced by a
+// 6-step intra-wave SEGMENTED inclusive scan (shfl_up with head-flag
+// propagation) instead of the walk kernel's serial per-subgroup loop —
+// the measured bound of the walk was exactly that serial run-detection
+// chain + flush divergence (tools/bench_apply.py), not the gv gathers.
+// Segment tails write their run's total: a plain 64 B row store when the
+// segment's head is in-wave (exclusive owner — slabs are zeroed by the
+// optimizer pass), atomicAdd only for the <=2 wave-spanning segments.
+// ---------------------------------------------------------------------------
+template <int K>
+__global__ void fm_segscan_apply_kernel(
+    const int* __restrict__ sorted_fids, const long* __restrict__ perm,
+    const float* __restrict__ gw, const float* __restrict__ gv,
+    float* __restrict__ gradW, float* __restrict__ gradV,
+    unsigned long long* __restrict__ touched, int nnz) {
+  static_assert(K <= 16, "register-vector variant holds the row per lane");
+  const int lane = threadIdx.x & (LCTR_WAVE - 1);
+  const int wave = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  const long base = (long)wave * LCTR_WAVE;
+  if (base >= nnz) return;
+  const long e = base + lane;
+  const bool valid = e < nnz;
+
+  const int fid = valid ? sorted_fids[e] : -1;
+  // original head flag: first entry of a run (lane 0 checks its global
+  // predecessor — one extra cached read per wave)
+  bool head0 = false;
+  if (valid) head0 = (e == 0) || (sorted_fids[e - 1] != fid);
+
+  float v[K + 1];  // [0..K) = gv row, [K] = gw
+  if (valid) {
+    const long p = perm ? perm[e] : e;
+    const float4* src = (const float4*)&gv[(size_t)p * K];
+#pragma unroll
+    for (int q = 0; q < K / 4; ++q) {
+      const float4 t = src[q];
+      v[q * 4 + 0] = t.x;
+      v[q * 4 + 1] = t.y;
+      v[q * 4 + 2] = t.z;
+      v[q * 4 + 3] = t.w;
+    }
+    v[K] = gw[p];
+  } else {
+#pragma unroll
+    for (int i = 0; i <= K; ++i) v[i] = 0.f;
+  }
+
+  // segmented inclusive scan: f = "a head lies within my covered window";
+  // add the incoming prefix only while my window is still head-free.
+  unsigned f = head0 ? 1u : 0u;
+#pragma unroll
+  for (int s = 1; s < LCTR_WAVE; s <<= 1) {
+    const unsigned tf = __shfl_up(f, s);
+    const bool take = (lane >= s) && (f == 0u);
+    float tv[K + 1];
+#pragma unroll
+    for (int i = 0; i <= K; ++i) tv[i] = __shfl_up(v[i], s);
+    if (take) {
+#pragma unroll
+      for (int i = 0; i <= K; ++i) v[i] += tv[i];
+    }
+    if (lane >= s) f |= tf;
+  }
+
+  // tail = last entry of a run: the next entry starts a new run (its
+  // original head flag), or the stream ends here.
+  const bool nh = __shfl_down((int)head0, 1);
+  bool tail = false;
+  if (valid) {
+    if (e + 1 >= nnz) {
+      tail = true;
+    } else if (lane == LCTR_WAVE - 1) {
+      tail = sorted_fids[e + 1] != fid;
+    } else {
+      tail = nh;
+    }
+  }
+
+  if (tail) {
+    if (f) {
+      // head in-wave -> this lane holds the run's TOTAL and owns it
+      float4* dst = (float4*)&gradV[(size_t)fid * K];
+#pragma unroll
+      for (int q = 0; q < K / 4; ++q) {
+        float4 t;
+        t.x = v[q * 4 + 0];
+        t.y = v[q * 4 + 1];
+        t.z = v[q * 4 + 2];
+        t.w = v[q * 4 + 3];
+        dst[q] = t;
+      }
+      gradW[fid] = v[K];
+    } else {
+#pragma unroll
+      for (int i = 0; i < K; ++i)
+        atomicAdd(&gradV[(size_t)fid * K + i], v[i]);
+      atomicAdd(&gradW[fid], v[K]);
+    }
+    atomicOr(&touched[fid >> 6], 1ull << (fid & 63));
+  } else if (valid && lane == LCTR_WAVE - 1) {
+    // run continues into the next wave: flush this wave's partial sum
+#pragma unroll
+    for (int i = 0; i < K; ++i)
+      atomicAdd(&gradV[(size_t)fid * K + i], v[i]);
+    atomicAdd(&gradW[fid], v[K]);
+    atomicOr(&touched[fid >> 6], 1ull << (fid & 63));
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Bitmap -> unique-fid list compaction. One thread per 64-feature word;
 // clears the word as it goes so the bitmap is reusable next step.
 // ---------------------------------------------------------------------------
@@ -483,6 +595,35 @@ void fm_sorted_apply_launch(const int* sorted_fids, const long* perm,
                             float p0, float p1, float p2, float p3,
                             int chunk, hipStream_t stream) {
   if (nnz <= 0) return;
+  // chunk 0 = auto (walk kernel, chunk 384 — measured best), -1 = the
+  // segmented-scan kernel (correct, measured 275 vs 218 us: the 6-step
+  // 17-register shfl_up chain costs more than the serial walk saves;
+  // kept selectable + parity-tested as a documented negative result)
+  const bool can_scan =
+      (K == 4 || K == 8 || K == 16) && opt_mode == 0;
+  if (chunk == -1 && can_scan) {
+    const int wpb = waves_per_block();
+    const int nwaves = (nnz + LCTR_WAVE - 1) / LCTR_WAVE;
+    dim3 block(wpb * LCTR_WAVE);
+    dim3 grid((nwaves + wpb - 1) / wpb);
+    switch (K) {
+      case 4:
+        hipLaunchKernelGGL((fm_segscan_apply_kernel<4>), grid, block, 0,
+                           stream, sorted_fids, perm, gw, gv, gradW, gradV,
+                           touched, nnz);
+        return;
+      case 8:
+        hipLaunchKernelGGL((fm_segscan_apply_kernel<8>), grid, block, 0,
+                           stream, sorted_fids, perm, gw, gv, gradW, gradV,
+                           touched, nnz);
+        return;
+      default:
+        hipLaunchKernelGGL((fm_segscan_apply_kernel<16>), grid, block, 0,
+                           stream, sorted_fids, perm, gw, gv, gradW, gradV,
+                           touched, nnz);
+        return;
+    }
+  }
   if (chunk <= 0) chunk = 384;  // measured optimum, tools/bench_apply.py
   const int wpb = waves_per_block();
   const int nwaves = (nnz + chunk - 1) / chunk;

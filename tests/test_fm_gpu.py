@@ -325,3 +325,36 @@ def test_scatter_emit_matches_gather_apply():
     assert torch.allclose(gW1, gW2, atol=1e-5)
     assert torch.allclose(gV1, gV2, atol=1e-5)
     assert torch.equal(t1, t2)
+
+
+def test_segscan_apply_matches_walk():
+    """chunk=-1 selects the segmented-scan apply; slabs must match the
+    default walk kernel bit-for-bit up to fp32 reduction order."""
+    from lightctr_amd.ops._extension import require_hip_ops, sort_ids
+
+    ops = require_hip_ops()
+    torch.manual_seed(11)
+    F, K, B = 3000, 16, 777
+    row_ptr, fids, vals, labels = make_random_csr(B=B, F_total=F,
+                                                  binary_vals=False, seed=11)
+    row_ptr, fids = row_ptr.cuda(), fids.cuda()
+    vals, labels = vals.cuda(), labels.cuda()
+    V = torch.randn(F, K, device="cuda") * 0.05
+    W = torch.randn(F, device="cuda") * 0.05
+    pred, sumVX = ops.fm_forward(row_ptr, fids, vals, W, V)
+    _, dpred = ops.logloss_grad(pred, labels, 1.0 / B)
+    gw, gv = ops.fm_backward_emit(row_ptr, fids, vals, V, sumVX, dpred)
+    sorted_fids, perm = sort_ids(fids, F)
+
+    out = []
+    for chunk in (0, -1):
+        gradW = torch.zeros(F, device="cuda")
+        gradV = torch.zeros(F, K, device="cuda")
+        touched = torch.zeros((F + 63) // 64, dtype=torch.int64,
+                              device="cuda")
+        ops.fm_sorted_apply(sorted_fids, perm, gw, gv, gradW, gradV,
+                            touched, chunk)
+        out.append((gradW, gradV, touched))
+    assert torch.allclose(out[0][0], out[1][0], atol=1e-5)
+    assert torch.allclose(out[0][1], out[1][1], atol=1e-5)
+    assert torch.equal(out[0][2], out[1][2])
