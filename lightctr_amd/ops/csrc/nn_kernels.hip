@@ -52,8 +52,19 @@ __global__ void colsum_kernel(const float* __restrict__ dZ,
   const int chunk = (M + gridDim.y - 1) / gridDim.y;
   const int m0 = blockIdx.y * chunk;
   const int m1 = min(M, m0 + chunk);
-  float s = 0.f;
-  for (int m = m0; m < m1; ++m) s += dZ[(size_t)m * N + col];
+  // 4 independent partials keep 4 loads in flight per lane (the single-
+  // accumulator loop left one outstanding load per wave: 59 us for a
+  // 67 MB pass ~ 14x off the HBM floor; profiles/prof_wd)
+  float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
+  int m = m0;
+  for (; m + 3 < m1; m += 4) {
+    s0 += dZ[(size_t)m * N + col];
+    s1 += dZ[(size_t)(m + 1) * N + col];
+    s2 += dZ[(size_t)(m + 2) * N + col];
+    s3 += dZ[(size_t)(m + 3) * N + col];
+  }
+  for (; m < m1; ++m) s0 += dZ[(size_t)m * N + col];
+  const float s = (s0 + s1) + (s2 + s3);
   if (s != 0.f) atomicAdd(&db[col], s);
 }
 
@@ -140,7 +151,7 @@ void colsum_launch(const float* dZ, float* db, int M, int N,
     return;
   }
   const int xtiles = (N + 255) / 256;
-  const int ychunks = max(1, min(1024 / xtiles, (M + 255) / 256));
+  const int ychunks = max(1, min(2048 / xtiles, (M + 127) / 128));
   dim3 grid(xtiles * 256 / 256, ychunks);
   grid.x = xtiles;
   hipLaunchKernelGGL(colsum_kernel, grid, block, 0, stream, dZ, db, M, N);
