@@ -63,9 +63,13 @@ class FFMModel:
         self._use_hip = self.device.type == "cuda"
         # backward variants, all parity-tested (tests/test_ffm.py):
         #   sorted (default) — per-run LDS-block recompute, 4.0 ms
-        #   blocks — wave/entry emit + coalesced block reduce, 4.4+1.1 ms
-        #            (measured slower: the bound is the per-(pair,k)
-        #            instruction stream, not scheduling; kept as evidence)
+        #   rowemit — per-row fp16-staged block emit + interior-store
+        #            segment reduce (4.7-5.2 + 1.3 ms measured across
+        #            fp32/fp16 staging and 4/8-wave workgroups: every
+        #            emit flavor lands 4-5 ms, i.e. the bound is the
+        #            per-(entry,partner) accumulate stream itself, not
+        #            V-read scheduling; kept + tested as evidence)
+        #   blocks — wave/entry fp32 emit + block reduce, 4.4+1.1 ms
         #   atomic — naive scatter (hot-feature serialization)
         self.backward_mode = "sorted"
         if self._use_hip:
@@ -103,7 +107,16 @@ class FFMModel:
             ops = require_hip_ops()
             pred = ops.ffm_forward(row_ptr, fields, fids, vals, self.W, self.V)
             loss, dpred = ops.logloss_grad(pred, labels, scale)
-            if self.backward_mode == "blocks":
+            if self.backward_mode == "rowemit":
+                gw, gblocks = ops.ffm_row_emit(row_ptr, fields, fids, vals,
+                                               self.V, dpred)
+                sorted_fids, perm = sort_ids(fids, self.h.num_features)
+                ops.ffm_blocks_apply_f16(sorted_fids, perm, gblocks, gw,
+                                         self.gradW,
+                                         self.gradV.view(
+                                             self.h.num_features, -1),
+                                         self.touched)
+            elif self.backward_mode == "blocks":
                 row_of_entry = ops.row_index(row_ptr, fids.numel())
                 gw, gblocks = ops.ffm_block_emit(row_of_entry, row_ptr,
                                                  fields, fids, vals, self.V,

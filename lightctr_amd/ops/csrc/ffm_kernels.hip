@@ -12,6 +12,8 @@
 //     exactly one pair: no reuse => no LDS staging, stay bandwidth-bound)
 //   * backward scatters symmetric grads with fp32 atomics + touched bitmap
 //     (optimizer = generic sparse fused kernels in misc_kernels.hip).
+#include <hip/hip_fp16.h>
+
 #include "common.h"
 
 namespace lightctr {
@@ -283,6 +285,266 @@ __global__ void ffm_blocks_apply_kernel(
     }
     if (lane == 0) atomicAdd(&gradW[cur], accw);
   }
+}
+
+// ---------------------------------------------------------------------------
+// Per-ROW LDS-staged FFM kernels. The direct pair-loop kernels read two
+// scattered 32 B V slices per pair; at K=8 that wastes half of every 64 B
+// HBM line (measured forward 1.2 ms vs a ~0.4 ms once-read floor). Here a
+// WORKGROUP owns one CSR row and first stages every entry's full
+// [nfields,K] V row into LDS with fully-coalesced float4 loads (each V
+// byte crosses the bus exactly once), then runs the pair loop against
+// LDS. Rows longer than `maxn` (duplicate-field outliers) fall back to
+// the direct-HBM loop inside the same kernel. Eligibility (launcher):
+// staged LDS footprint <= 64 KB.
+// ---------------------------------------------------------------------------
+template <int K>
+__global__ void ffm_fwd_staged_kernel(
+    const int* __restrict__ row_ptr, const int* __restrict__ fields,
+    const int* __restrict__ fids, const float* __restrict__ vals,
+    const float* __restrict__ W, const float* __restrict__ V,
+    float* __restrict__ pred, int nfields, int B, int maxn) {
+  extern __shared__ float lds[];  // [maxn * nfields * K] staged V rows
+  __shared__ float red[8][2];
+  constexpr int G = LCTR_WAVE / K;
+  const int row = blockIdx.x;
+  if (row >= B) return;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wv = tid >> 6;
+  const int nw = blockDim.x >> 6;
+  const int g = lane / K;
+  const int k = lane % K;
+  const int beg = row_ptr[row], end = row_ptr[row + 1];
+  const int n = end - beg;
+  const int D = nfields * K;
+  const int npairs = n * (n - 1) / 2;
+
+  float lin = 0.f;
+  for (int j = beg + tid; j < end; j += blockDim.x)
+    lin += W[fids[j]] * vals[j];
+
+  float acc = 0.f;
+  const bool staged = n <= maxn;
+  if (staged) {
+    const int nf4 = D / 4;  // K >= 4 -> D % 4 == 0
+    float4* st4 = (float4*)lds;
+    for (int idx = tid; idx < n * nf4; idx += blockDim.x) {
+      const int e = idx / nf4, q = idx - e * nf4;
+      st4[idx] = ((const float4*)(V + (size_t)fids[beg + e] * D))[q];
+    }
+  }
+  __syncthreads();
+  for (int p = wv * G + g; p < npairs; p += nw * G) {
+    int i, j;
+    tri_decode(p, n, &i, &j);
+    const int Fi = fields[beg + i], Fj = fields[beg + j];
+    float t;
+    if (staged) {
+      t = lds[(i * nfields + Fj) * K + k] * lds[(j * nfields + Fi) * K + k];
+    } else {
+      t = V[((size_t)fids[beg + i] * nfields + Fj) * K + k] *
+          V[((size_t)fids[beg + j] * nfields + Fi) * K + k];
+    }
+#pragma unroll
+    for (int sft = 1; sft < K; sft <<= 1) t += __shfl_xor(t, sft);
+    acc += t * vals[beg + i] * vals[beg + j];
+  }
+  acc = wave_reduce_sum(acc) * (1.f / K);
+  lin = wave_reduce_sum(lin);
+  if (lane == 0) {
+    red[wv][0] = acc;
+    red[wv][1] = lin;
+  }
+  __syncthreads();
+  if (tid == 0) {
+    float a = 0.f, l = 0.f;
+    for (int w = 0; w < nw; ++w) {
+      a += red[w][0];
+      l += red[w][1];
+    }
+    pred[row] = l + a;
+  }
+}
+
+// Backward phase 1, per-row staged: emit each entry's [nfields,K]
+// gradient block in HALF precision (the block tensor is the dominant
+// traffic: fp16 halves both the emit write and the apply read; gradient
+// noise >> fp16 rounding). Waves of the workgroup take entries
+// round-robin; each accumulates the entry block in its private LDS slice
+// (ds atomics absorb duplicate fields) reading partner slices from the
+// shared stage.
+template <int K>
+__global__ void ffm_row_emit_kernel(
+    const int* __restrict__ row_ptr, const int* __restrict__ fields,
+    const int* __restrict__ fids, const float* __restrict__ vals,
+    const float* __restrict__ V, const float* __restrict__ dpred,
+    _Float16* __restrict__ gblocks, float* __restrict__ gw, int nfields,
+    int B, int maxn) {
+  // stage holds fp16 (halves the LDS footprint -> 2x the workgroups per
+  // CU; the emitted blocks are fp16 anyway, so staging precision is not
+  // the bottleneck term); acc stays fp32.
+  extern __shared__ float lds[];  // acc[nw][nf*(K+1)] | stage_h[maxn*D]
+  constexpr int G = LCTR_WAVE / K;
+  const int row = blockIdx.x;
+  if (row >= B) return;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wv = tid >> 6;
+  const int nw = blockDim.x >> 6;
+  const int g = lane / K;
+  const int k = lane % K;
+  const int beg = row_ptr[row], end = row_ptr[row + 1];
+  const int n = end - beg;
+  const int D = nfields * K;
+  const int fstride = K + 1;
+  float* acc = lds + (size_t)wv * nfields * fstride;
+  _Float16* stage = (_Float16*)(lds + (size_t)nw * nfields * fstride);
+  const float d = dpred[row];
+
+  const bool staged = n <= maxn;
+  if (staged) {
+    for (int idx = tid; idx < n * D; idx += blockDim.x) {
+      const int e = idx / D, q = idx - e * D;
+      stage[idx] = (_Float16)V[(size_t)fids[beg + e] * D + q];
+    }
+  }
+  __syncthreads();
+
+  for (int i = wv; i < n; i += nw) {
+    for (int t = lane; t < nfields * fstride; t += LCTR_WAVE) acc[t] = 0.f;
+    const int p = beg + i;
+    const int Fi = fields[p];
+    const float xi = vals[p];
+    for (int j = g; j < n; j += G) {
+      if (j == i) continue;
+      const float v = staged
+                          ? (float)stage[(j * nfields + Fi) * K + k]
+                          : V[((size_t)fids[beg + j] * nfields + Fi) * K + k];
+      atomicAdd(&acc[fields[beg + j] * fstride + k],
+                d * xi * vals[beg + j] * v);
+    }
+    for (int t = lane; t < D; t += LCTR_WAVE)
+      gblocks[(size_t)p * D + t] = (_Float16)acc[(t / K) * fstride + (t % K)];
+    if (lane == 0) gw[p] = d * xi;
+  }
+}
+
+// Backward phase 2 for fp16 blocks: fid-sorted chunked segment reduce with
+// the FM interior-store trick — a run whose global head AND tail fall in
+// this wave's chunk owns its feature exclusively, so the flush is a plain
+// coalesced store sweep (slabs are zeroed by the optimizer pass); only
+// chunk-spanning runs use atomicAdd.
+__global__ void ffm_blocks_apply_f16_kernel(
+    const int* __restrict__ sorted_fids, const long* __restrict__ perm,
+    const _Float16* __restrict__ gblocks, const float* __restrict__ gw,
+    float* __restrict__ gradW, float* __restrict__ gradV,
+    unsigned long long* __restrict__ touched, int D, int nnz, int chunk) {
+  extern __shared__ float lds_acc[];
+  const int lane = threadIdx.x & 63;
+  const int wave_in_blk = threadIdx.x >> 6;
+  const int wave = blockIdx.x * (blockDim.x >> 6) + wave_in_blk;
+  const int base = wave * chunk;
+  if (base >= nnz) return;
+  const int end = min(base + chunk, nnz);
+  float* acc = &lds_acc[(size_t)wave_in_blk * D];
+  for (int i = lane; i < D; i += LCTR_WAVE) acc[i] = 0.f;
+
+  int cur = -1;
+  bool head_ok = false;
+  float accw = 0.f;
+  auto flush = [&](int tail_e) {
+    if (cur < 0) return;
+    const bool tail_ok = tail_e >= nnz || sorted_fids[tail_e] != cur;
+    if (head_ok && tail_ok) {
+      for (int i = lane; i < D; i += LCTR_WAVE) {
+        gradV[(size_t)cur * D + i] = acc[i];
+        acc[i] = 0.f;
+      }
+      if (lane == 0) gradW[cur] = accw;
+    } else {
+      for (int i = lane; i < D; i += LCTR_WAVE) {
+        if (acc[i] != 0.f) atomicAdd(&gradV[(size_t)cur * D + i], acc[i]);
+        acc[i] = 0.f;
+      }
+      if (lane == 0) atomicAdd(&gradW[cur], accw);
+    }
+  };
+  for (int e = base; e < end; ++e) {
+    const int fid = sorted_fids[e];
+    if (fid != cur) {
+      flush(e);
+      cur = fid;
+      accw = 0.f;
+      head_ok = (e == 0 || sorted_fids[e - 1] != fid);
+      if (lane == 0 && head_ok)
+        atomicOr(&touched[fid >> 6], 1ull << (fid & 63));
+    }
+    const long p = perm[e];
+    const __half2* gb2 = (const __half2*)&gblocks[(size_t)p * D];
+    for (int i = lane; i < D / 2; i += LCTR_WAVE) {
+      const float2 t = __half22float2(gb2[i]);
+      acc[2 * i] += t.x;
+      acc[2 * i + 1] += t.y;
+    }
+    if (D & 1)
+      if (lane == 0) acc[D - 1] += (float)gblocks[(size_t)p * D + D - 1];
+    if (lane == 0) accw += gw[p];
+  }
+  flush(end);
+}
+
+bool ffm_staged_eligible(int nfields, int K, int maxn) {
+  const size_t stage_b = (size_t)maxn * nfields * K * sizeof(_Float16);
+  const size_t acc_b = (size_t)4 * nfields * (K + 1) * sizeof(float);
+  return stage_b + acc_b <= (64 << 10);
+}
+
+void ffm_fwd_staged_launch(const int* row_ptr, const int* fields,
+                           const int* fids, const float* vals, const float* W,
+                           const float* V, float* pred, int nfields, int B,
+                           int maxn, int K, hipStream_t stream) {
+  if (B <= 0) return;
+  dim3 block(512);
+  dim3 grid(B);
+  const size_t lds = (size_t)maxn * nfields * K * sizeof(float);
+  DISPATCH_FFM_K(K, hipLaunchKernelGGL((ffm_fwd_staged_kernel<KC>), grid,
+                                       block, lds, stream, row_ptr, fields,
+                                       fids, vals, W, V, pred, nfields, B,
+                                       maxn));
+}
+
+void ffm_row_emit_launch(const int* row_ptr, const int* fields,
+                         const int* fids, const float* vals, const float* V,
+                         const float* dpred, void* gblocks, float* gw,
+                         int nfields, int B, int maxn, int K,
+                         hipStream_t stream) {
+  if (B <= 0) return;
+  dim3 block(256);
+  dim3 grid(B);
+  const size_t lds = (size_t)4 * nfields * (K + 1) * sizeof(float) +
+                     (size_t)maxn * nfields * K * sizeof(_Float16);
+  DISPATCH_FFM_K(K, hipLaunchKernelGGL((ffm_row_emit_kernel<KC>), grid, block,
+                                       lds, stream, row_ptr, fields, fids,
+                                       vals, V, dpred, (_Float16*)gblocks, gw,
+                                       nfields, B, maxn));
+}
+
+void ffm_blocks_apply_f16_launch(const int* sorted_fids, const long* perm,
+                                 const void* gblocks, const float* gw,
+                                 float* gradW, float* gradV,
+                                 unsigned long long* touched, int D, int nnz,
+                                 hipStream_t stream) {
+  if (nnz <= 0) return;
+  const int chunk = 96;
+  const int wpb = 4;
+  const int nwaves = (nnz + chunk - 1) / chunk;
+  dim3 block(wpb * LCTR_WAVE);
+  dim3 grid((nwaves + wpb - 1) / wpb);
+  const size_t lds = (size_t)wpb * D * sizeof(float);
+  hipLaunchKernelGGL(ffm_blocks_apply_f16_kernel, grid, block, lds, stream,
+                     sorted_fids, perm, (const _Float16*)gblocks, gw, gradW,
+                     gradV, touched, D, nnz, chunk);
 }
 
 void ffm_block_emit_launch(const int* row_of_entry, const int* row_ptr,

@@ -61,6 +61,21 @@ void ffm_block_emit_launch(const int* row_of_entry, const int* row_ptr,
                            const float* vals, const float* V,
                            const float* dpred, float* gblocks, float* gw,
                            int nfields, int nnz, int K, ihipStream_t* stream);
+bool ffm_staged_eligible(int nfields, int K, int maxn);
+void ffm_fwd_staged_launch(const int* row_ptr, const int* fields,
+                           const int* fids, const float* vals, const float* W,
+                           const float* V, float* pred, int nfields, int B,
+                           int maxn, int K, ihipStream_t* stream);
+void ffm_row_emit_launch(const int* row_ptr, const int* fields,
+                         const int* fids, const float* vals, const float* V,
+                         const float* dpred, void* gblocks, float* gw,
+                         int nfields, int B, int maxn, int K,
+                         ihipStream_t* stream);
+void ffm_blocks_apply_f16_launch(const int* sorted_fids, const long* perm,
+                                 const void* gblocks, const float* gw,
+                                 float* gradW, float* gradV,
+                                 unsigned long long* touched, int D, int nnz,
+                                 ihipStream_t* stream);
 void ffm_blocks_apply_launch(const int* sorted_fids, const long* perm,
                              const float* gblocks, const float* gw,
                              float* gradW, float* gradV,

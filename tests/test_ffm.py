@@ -223,3 +223,66 @@ def test_ffm_blocks_backward_parity():
         (gradW - gW_ref).abs().max()
     assert torch.allclose(gradV, gV_ref, atol=1e-5, rtol=1e-4), \
         (gradV - gV_ref).abs().max()
+
+
+@pytest.mark.gpu
+def test_ffm_rowemit_backward_parity():
+    """rowemit mode (per-row staged fp16 block emit + interior-store
+    reduce) == dense reference within fp16 block tolerance. Includes rows
+    longer than the staging cap (maxn=40) to cover the in-kernel HBM
+    fallback, and duplicate fields per row (LDS-atomic path)."""
+    from lightctr_amd.ops import hip_ops
+
+    row_ptr, fids, vals, labels = make_random_csr(B=96, F_total=3000,
+                                                  min_f=2, max_f=60, seed=23,
+                                                  device="cuda:0",
+                                                  binary_vals=False)
+    nf, K = 6, 8
+    fields = (fids.long() % nf).int()
+    g = torch.Generator().manual_seed(8)
+    W = torch.randn(3000, generator=g).cuda()
+    V = (torch.randn(3000, nf, K, generator=g) * 0.1).cuda()
+    pred = hip_ops.ffm_forward(row_ptr, fields, fids, vals, W, V)
+    _, dpred = hip_ops.logloss_grad(pred, labels, 1.0 / 96)
+    gw, gblocks = hip_ops.ffm_row_emit(row_ptr, fields, fids, vals, V, dpred)
+    assert gblocks.dtype == torch.float16
+    sorted_fids, perm = torch.sort(fids)
+    gradW = torch.zeros(3000).cuda()
+    gradV = torch.zeros(3000, nf, K).cuda()
+    touched = torch.zeros((3000 + 63) // 64, dtype=torch.int64).cuda()
+    hip_ops.ffm_blocks_apply_f16(sorted_fids, perm, gblocks, gw, gradW,
+                                 gradV.view(3000, -1), touched)
+    gW_ref, gV_ref = ffm_ref.ffm_backward_ref(row_ptr, fields, fids, vals,
+                                              V, dpred)
+    # gw is fp32 (exact); gv blocks round through fp16 once
+    assert torch.allclose(gradW, gW_ref, atol=1e-5, rtol=1e-4), \
+        (gradW - gW_ref).abs().max()
+    ref_scale = gV_ref.abs().max()
+    assert torch.allclose(gradV, gV_ref, atol=float(ref_scale) * 2e-2,
+                          rtol=2e-2), (gradV - gV_ref).abs().max()
+    # touched bitmap covers exactly the batch's features
+    expect = torch.zeros((3000 + 63) // 64, dtype=torch.int64).cuda()
+    for f in torch.unique(fids).tolist():
+        expect[f // 64] |= 1 << (f % 64)
+    assert torch.equal(touched, expect)
+
+
+@pytest.mark.gpu
+def test_ffm_forward_staged_long_rows():
+    """Forward parity on rows spanning the staged/fallback boundary."""
+    from lightctr_amd.ops import hip_ops
+
+    row_ptr, fids, vals, labels = make_random_csr(B=64, F_total=2000,
+                                                  min_f=30, max_f=55, seed=31,
+                                                  device="cuda:0",
+                                                  binary_vals=False)
+    nf, K = 8, 8
+    fields = (fids.long() % nf).int()
+    g = torch.Generator().manual_seed(9)
+    W = torch.randn(2000, generator=g).cuda()
+    V = (torch.randn(2000, nf, K, generator=g) * 0.1).cuda()
+    pred = hip_ops.ffm_forward(row_ptr, fields, fids, vals, W, V)
+    ref = ffm_ref.ffm_forward_ref(row_ptr.cpu(), fields.cpu(), fids.cpu(),
+                                  vals.cpu(), W.cpu(), V.cpu())
+    assert torch.allclose(pred.cpu(), ref, atol=1e-3, rtol=1e-4), \
+        (pred.cpu() - ref).abs().max()
