@@ -300,6 +300,24 @@ def cmd_predict(args) -> int:
         model = FFMModel(FFMHyper(num_features=F,
                                   num_fields=ds.num_fields, k=args.k),
                          device=dev)
+    elif args.model == "nfm":
+        from .models.nfm import NFMHyper, NFMModel
+
+        model = NFMModel(NFMHyper(num_features=F, k=args.k,
+                                  optimizer=args.optimizer), device=dev)
+    elif args.model == "widedeep":
+        from .models.wide_deep import WideDeepHyper, WideDeepModel
+
+        model = WideDeepModel(WideDeepHyper(num_features=F,
+                                            num_fields=ds.num_fields,
+                                            k=args.k,
+                                            optimizer=args.optimizer),
+                              device=dev)
+    elif args.model == "lr":
+        from .models.lr import LRHyper, LRModel
+
+        model = LRModel(LRHyper(num_features=F, optimizer=args.optimizer),
+                        device=dev)
     else:
         print(f"unknown model {args.model}", file=sys.stderr)
         return 2

@@ -156,3 +156,23 @@ def test_cli_train_and_predict(tmp_path):
                "256", "--features", "8192", "--k", "8", "--device", "cpu",
                "--load", model_path])
     assert rc == 0
+
+
+def test_cli_predict_all_sparse_models(tmp_path):
+    """Every sparse model family round-trips train --save -> predict
+    --load --dump (the reference's FM_Predict covers its whole zoo)."""
+    from lightctr_amd.cli import main
+
+    for m in ("ffm", "nfm", "widedeep", "lr"):
+        model_path = str(tmp_path / f"{m}.pt")
+        dump_path = str(tmp_path / f"{m}.scores")
+        rc = main(["train", "--model", m, "--data", "synthetic", "--rows",
+                   "256", "--features", "4096", "--k", "8", "--epochs", "1",
+                   "--batch", "128", "--device", "cpu", "--save", model_path])
+        assert rc == 0, m
+        rc = main(["predict", "--model", m, "--data", "synthetic", "--rows",
+                   "128", "--features", "4096", "--k", "8", "--device",
+                   "cpu", "--load", model_path, "--dump", dump_path])
+        assert rc == 0, m
+        with open(dump_path) as f:
+            assert len(f.readlines()) == 128, m
