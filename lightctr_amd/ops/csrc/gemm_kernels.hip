@@ -21,6 +21,7 @@
 //   * fused epilogue: optional bias add + activation (relu/sigmoid) +
 //     optional bf16 mirror of C for the next layer's input.
 #include "common.h"
+#include "launchers.h"
 
 namespace lightctr {
 
@@ -225,6 +226,21 @@ void gemm_bf16_launch(const void* A, const void* Bst, const float* bias,
   // kept (correct, race-screened) as the documented experiment.
   if (gemm256_eligible(M, N, K, transA, transB)) {
     gemm256_bf16_launch(A, Bst, bias, C, Cbf, M, N, K, act, stream);
+    return;
+  }
+  // degenerate shapes (the MLP's 1-wide output layer): direct kernels
+  // instead of 128x128 tiles at ~99% waste (see nn_kernels.hip)
+  if (N == 1 && transA == 0) {
+    gemv_n1_launch(A, Bst, bias, C, Cbf, M, K, act, stream);
+    return;
+  }
+  if (M == 1 && transA == 1 && transB == 1 && bias == nullptr &&
+      act == 0 && Cbf == nullptr) {
+    wrowsum_m1_launch(A, Bst, C, K, N, stream);
+    return;
+  }
+  if (K == 1 && transA == 0) {
+    outer_k1_launch(A, Bst, bias, C, Cbf, M, N, act, stream);
     return;
   }
   dim3 block(256);

@@ -121,6 +121,14 @@ void gemm256_bf16_launch(const void* A, const void* Bst, const float* bias,
                          ihipStream_t* stream);
 
 // --- nn_kernels.hip ---
+void gemv_n1_launch(const void* A, const void* b, const float* bias,
+                    float* C, void* Cbf, int M, int K, int act,
+                    ihipStream_t* stream);
+void wrowsum_m1_launch(const void* a, const void* Bst, float* C, int K,
+                       int N, ihipStream_t* stream);
+void outer_k1_launch(const void* A, const void* Bst, const float* bias,
+                     float* C, void* Cbf, long M, long N, int act,
+                     ihipStream_t* stream);
 void act_backward_launch(const float* dY, const float* Y, float* dZ,
                          void* dZbf, long n, int act, ihipStream_t* stream);
 void colsum_launch(const float* dZ, float* db, int M, int N,

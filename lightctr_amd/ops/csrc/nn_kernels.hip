@@ -188,4 +188,112 @@ void dense_adagrad_launch(float* W, const float* grad, float* n_t, void* Wbf,
                      l2);
 }
 
+// ---------------------------------------------------------------------------
+// Degenerate-GEMM fast paths. The MLP's last layer (out_dim 1) pushes
+// three shapes through the 128x128 tile kernel at ~99% tile waste:
+// forward [M,1,K] (a GEMV), wgrad [1,N,K] (a weighted row-sum), and
+// dgrad [M,N,1] (an outer product). Each is bandwidth-trivial when
+// expressed directly. Routed from gemm_bf16_launch (gemm_kernels.hip).
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ float gemm_act(float v, int act) {
+  if (act == 1) return fmaxf(v, 0.f);
+  if (act == 2) return sigmoidf_clamped(v);
+  return v;
+}
+
+// C[m,0] = sum_k A[m,k]*b[k] : one wave per row, lanes stride K.
+__global__ void gemv_n1_kernel(const __bf16* __restrict__ A,
+                               const __bf16* __restrict__ b,
+                               const float* __restrict__ bias,
+                               float* __restrict__ C,
+                               __bf16* __restrict__ Cbf, int M, int K,
+                               int act) {
+  const int lane = threadIdx.x & 63;
+  const int row = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  if (row >= M) return;
+  float s = 0.f;
+  for (int k = lane; k < K; k += 64)
+    s += (float)A[(size_t)row * K + k] * (float)b[k];
+  s = wave_reduce_sum(s);
+  if (lane == 0) {
+    const float v = gemm_act(s + (bias ? bias[0] : 0.f), act);
+    C[row] = v;
+    if (Cbf) Cbf[row] = (__bf16)v;
+  }
+}
+
+// C[0,n] = sum_k a[k]*Bst[k,n] (wgrad for a 1-wide layer): column-tiled
+// weighted sum, 4-deep unrolled partials, atomic per (block,column).
+// C must be zeroed by the launcher; no bias/act/Cbf (wgrad has none).
+__global__ void wrowsum_m1_kernel(const __bf16* __restrict__ a,
+                                  const __bf16* __restrict__ Bst,
+                                  float* __restrict__ C, int K, int N) {
+  const int col = blockIdx.x * blockDim.x + threadIdx.x;
+  if (col >= N) return;
+  const int chunk = (K + gridDim.y - 1) / gridDim.y;
+  const int k0 = blockIdx.y * chunk;
+  const int k1 = min(K, k0 + chunk);
+  float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
+  int k = k0;
+  for (; k + 3 < k1; k += 4) {
+    s0 += (float)a[k] * (float)Bst[(size_t)k * N + col];
+    s1 += (float)a[k + 1] * (float)Bst[(size_t)(k + 1) * N + col];
+    s2 += (float)a[k + 2] * (float)Bst[(size_t)(k + 2) * N + col];
+    s3 += (float)a[k + 3] * (float)Bst[(size_t)(k + 3) * N + col];
+  }
+  for (; k < k1; ++k) s0 += (float)a[k] * (float)Bst[(size_t)k * N + col];
+  const float s = (s0 + s1) + (s2 + s3);
+  if (s != 0.f) atomicAdd(&C[col], s);
+}
+
+// C[m,n] = A[m,0]*Bst[n,0] (+bias[n], act): outer product, K == 1.
+__global__ void outer_k1_kernel(const __bf16* __restrict__ A,
+                                const __bf16* __restrict__ Bst,
+                                const float* __restrict__ bias,
+                                float* __restrict__ C,
+                                __bf16* __restrict__ Cbf, long M, long N,
+                                int act) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= M * N) return;
+  const long m = i / N, n = i - m * N;
+  const float v = gemm_act((float)A[m] * (float)Bst[n] +
+                               (bias ? bias[n] : 0.f),
+                           act);
+  C[i] = v;
+  if (Cbf) Cbf[i] = (__bf16)v;
+}
+
+void gemv_n1_launch(const void* A, const void* b, const float* bias,
+                    float* C, void* Cbf, int M, int K, int act,
+                    hipStream_t stream) {
+  const int wpb = 4;
+  dim3 block(wpb * 64);
+  dim3 grid((M + wpb - 1) / wpb);
+  hipLaunchKernelGGL(gemv_n1_kernel, grid, block, 0, stream,
+                     (const __bf16*)A, (const __bf16*)b, bias, C,
+                     (__bf16*)Cbf, M, K, act);
+}
+
+void wrowsum_m1_launch(const void* a, const void* Bst, float* C, int K,
+                       int N, hipStream_t stream) {
+  LCTR_CHECK_HIP(hipMemsetAsync(C, 0, (size_t)N * sizeof(float), stream));
+  const int xtiles = (N + 255) / 256;
+  const int ychunks = max(1, min(2048 / xtiles, (K + 127) / 128));
+  dim3 block(256);
+  dim3 grid(xtiles, ychunks);
+  hipLaunchKernelGGL(wrowsum_m1_kernel, grid, block, 0, stream,
+                     (const __bf16*)a, (const __bf16*)Bst, C, K, N);
+}
+
+void outer_k1_launch(const void* A, const void* Bst, const float* bias,
+                     float* C, void* Cbf, long M, long N, int act,
+                     hipStream_t stream) {
+  const long total = M * N;
+  dim3 block(256);
+  dim3 grid((unsigned)((total + 255) / 256));
+  hipLaunchKernelGGL(outer_k1_kernel, grid, block, 0, stream,
+                     (const __bf16*)A, (const __bf16*)Bst, bias, C,
+                     (__bf16*)Cbf, M, N, act);
+}
+
 }  // namespace lightctr

@@ -245,3 +245,37 @@ def test_gemm_v2_race_screen():
         C = hip_ops.gemm_bf16(A, Bst, None, M, N, K, 0, 0, 0, False)
         assert torch.allclose(C, ref, atol=2e-2 * K ** 0.5, rtol=1e-2), \
             (it, (C - ref).abs().max())
+
+
+@pytest.mark.gpu
+def test_gemm_degenerate_shapes():
+    """The 1-wide-layer fast paths (gemv N=1, weighted rowsum M=1,
+    outer product K=1) match the fp32 reference in the exact call
+    patterns the MLP uses."""
+    from lightctr_amd.ops import hip_ops
+
+    g = torch.Generator().manual_seed(99)
+    B, In = 4096, 128
+    # forward: C[B,1] = X[B,In] @ w[1,In]^T + bias, sigmoid epilogue
+    X = (torch.randn(B, In, generator=g) * 0.5).to(torch.bfloat16).cuda()
+    w = (torch.randn(1, In, generator=g) * 0.5).to(torch.bfloat16).cuda()
+    bias = torch.randn(1, generator=g).cuda()
+    C, Cbf = hip_ops.gemm_bf16_full(X, w, bias, B, 1, In, 0, 0, 2)
+    z = X.float() @ w.float().t() + bias
+    ref = torch.sigmoid(torch.clamp(z, -16, 16))
+    assert torch.allclose(C, ref, atol=3e-2, rtol=1e-2), (C - ref).abs().max()
+    assert torch.allclose(Cbf.float(), ref, atol=5e-2), "bf16 mirror"
+
+    # wgrad: dW[1,In] = dZ[B,1]^T @ X[B,In]  (transA=1, transB=1, no bias)
+    dZ = (torch.randn(B, 1, generator=g) * 0.1).to(torch.bfloat16).cuda()
+    dW = hip_ops.gemm_bf16(dZ, X, None, 1, In, B, 1, 1, 0, False)
+    ref = dZ.float().t() @ X.float()
+    assert torch.allclose(dW, ref, atol=2e-1, rtol=1e-2), \
+        (dW - ref).abs().max()
+
+    # dgrad: dX[B,In] = dZ[B,1] @ Wt[In,1]^T  (K=1 outer product)
+    Wt = (torch.randn(In, 1, generator=g) * 0.5).to(torch.bfloat16).cuda()
+    dX = hip_ops.gemm_bf16(dZ, Wt, None, B, In, 1, 0, 0, 0, False)
+    ref = dZ.float() @ Wt.float().t()
+    assert torch.allclose(dX, ref, atol=1e-3, rtol=1e-2), \
+        (dX - ref).abs().max()
