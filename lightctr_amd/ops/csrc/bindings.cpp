@@ -186,15 +186,31 @@ at::Tensor ffm_forward(at::Tensor row_ptr, at::Tensor fields, at::Tensor fids,
   const int nfields = (int)V.size(1);
   const int K = (int)V.size(2);
   auto pred = at::empty({B}, W.options());
-  // note: a per-row LDS-staged forward (ffm_fwd_staged_launch) measured
-  // 1611 vs 1200 us here — LDS capacity caps occupancy below what the
-  // direct stream needs; kept in ffm_kernels.hip as the documented
-  // experiment, direct kernel stays the route.
-  lightctr::ffm_forward_launch(row_ptr.data_ptr<int>(),
-                               fields.data_ptr<int>(), fids.data_ptr<int>(),
-                               vals.data_ptr<float>(), W.data_ptr<float>(),
-                               V.data_ptr<float>(), pred.data_ptr<float>(),
-                               nfields, B, K, cur_stream());
+  // lane-per-pair layout: 458 vs 1174 us for the (pair-group, k) layout
+  // at B=65536/nf=39/K=8 (tools/ab_ffm_fwd.py; bit-identical output).
+  // The group kernel and a per-row LDS-staged variant (1611 us — LDS
+  // capacity caps occupancy) stay in ffm_kernels.hip as documented
+  // experiments.
+  lightctr::ffm_forward_pp_launch(
+      row_ptr.data_ptr<int>(), fields.data_ptr<int>(), fids.data_ptr<int>(),
+      vals.data_ptr<float>(), W.data_ptr<float>(), V.data_ptr<float>(),
+      pred.data_ptr<float>(), nfields, B, K, cur_stream());
+  return pred;
+}
+
+// lane-per-pair forward variant (A/B against the group layout)
+at::Tensor ffm_forward_pp(at::Tensor row_ptr, at::Tensor fields,
+                          at::Tensor fids, at::Tensor vals, at::Tensor W,
+                          at::Tensor V) {
+  check_cuda_i32(row_ptr, "row_ptr");
+  const int B = (int)row_ptr.numel() - 1;
+  const int nfields = (int)V.size(1);
+  const int K = (int)V.size(2);
+  auto pred = at::empty({B}, W.options());
+  lightctr::ffm_forward_pp_launch(
+      row_ptr.data_ptr<int>(), fields.data_ptr<int>(), fids.data_ptr<int>(),
+      vals.data_ptr<float>(), W.data_ptr<float>(), V.data_ptr<float>(),
+      pred.data_ptr<float>(), nfields, B, K, cur_stream());
   return pred;
 }
 
@@ -777,6 +793,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fm_sorted_apply_fused", &fm_sorted_apply_fused,
         "segment-reduce + fused optimizer for interior segments");
   m.def("ffm_forward", &ffm_forward, "FFM pairwise forward (LDS-staged)");
+  m.def("ffm_forward_pp", &ffm_forward_pp, "lane-per-pair FFM forward");
   m.def("ffm_row_emit", &ffm_row_emit,
         "FFM per-row staged fp16 block emit -> (gw, gblocks)");
   m.def("ffm_blocks_apply_f16", &ffm_blocks_apply_f16,

@@ -61,6 +61,10 @@ void ffm_block_emit_launch(const int* row_of_entry, const int* row_ptr,
                            const float* vals, const float* V,
                            const float* dpred, float* gblocks, float* gw,
                            int nfields, int nnz, int K, ihipStream_t* stream);
+void ffm_forward_pp_launch(const int* row_ptr, const int* fields,
+                           const int* fids, const float* vals,
+                           const float* W, const float* V, float* pred,
+                           int nfields, int B, int K, ihipStream_t* stream);
 bool ffm_staged_eligible(int nfields, int K, int maxn);
 void ffm_fwd_staged_launch(const int* row_ptr, const int* fields,
                            const int* fids, const float* vals, const float* W,
