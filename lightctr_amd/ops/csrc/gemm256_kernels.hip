@@ -12,6 +12,8 @@
 //
 // Eligibility: TRANSA=TRANSB=0, M%256==0, N%256==0, K%64==0, K%8==0 —
 // dispatched from gemm_bf16_launch; everything else takes the 128^2 path.
+#include <cstdlib>
+
 #include "common.h"
 
 namespace lightctr {
@@ -65,7 +67,10 @@ __device__ __forceinline__ float act_apply256(float v, int act) {
   return v;
 }
 
-template <bool TAIL>
+// BURST: issue all four of tile t+1's half-tile stages at phase 0 so the
+// tile-end vmcnt(0) drain waits on loads ~3 phases old instead of on the
+// half-tile just issued in phase 3 (A/B via LCTR_GEMM_BURST=1).
+template <bool TAIL, bool BURST>
 __global__ __launch_bounds__(512, 1) void gemm256_bf16_kernel(
     const __bf16* __restrict__ A, const __bf16* __restrict__ Bst,
     const float* __restrict__ bias, float* __restrict__ C,
@@ -120,19 +125,29 @@ __global__ __launch_bounds__(512, 1) void gemm256_bf16_kernel(
     // 4 phases: quadrant p = fm in [2p, 2p+2), all fn, full K-tile
 #pragma unroll
     for (int p = 0; p < 4; ++p) {
-      if (more) {  // prefetch half-tile p of tile t+1
+      if (more) {  // prefetch half-tile(s) of tile t+1
         const bool next_full = !TAIL || (k0 + G256_BK <= K);
         if (next_full) {
-          if (p == 0)
+          if (BURST) {
+            if (p == 0) {
+              stage_half_512(A + (long)M0 * K + k0, K, nxtA);
+              stage_half_512(A + (long)(M0 + 128) * K + k0, K,
+                             nxtA + 128 * G256_BK);
+              stage_half_512(Bst + (long)N0 * K + k0, K, nxtB);
+              stage_half_512(Bst + (long)(N0 + 128) * K + k0, K,
+                             nxtB + 128 * G256_BK);
+            }
+          } else if (p == 0) {
             stage_half_512(A + (long)M0 * K + k0, K, nxtA);
-          else if (p == 1)
+          } else if (p == 1) {
             stage_half_512(A + (long)(M0 + 128) * K + k0, K,
                            nxtA + 128 * G256_BK);
-          else if (p == 2)
+          } else if (p == 2) {
             stage_half_512(Bst + (long)N0 * K + k0, K, nxtB);
-          else
+          } else {
             stage_half_512(Bst + (long)(N0 + 128) * K + k0, K,
                            nxtB + 128 * G256_BK);
+          }
         } else {  // K tail: zero-filled scalar staging covers all 256
                   // rows per call — phases 0 (A) and 2 (B) only
           if (p == 0)
@@ -198,14 +213,28 @@ void gemm256_bf16_launch(const void* A, const void* Bst, const float* bias,
                          hipStream_t stream) {
   dim3 block(512);
   dim3 grid(N / 256, M / 256);
+  static const bool burst = [] {
+    const char* e = getenv("LCTR_GEMM_BURST");
+    return e && e[0] == '1';
+  }();
   if (K % G256_BK == 0) {
-    hipLaunchKernelGGL((gemm256_bf16_kernel<false>), grid, block, 0, stream,
-                       (const __bf16*)A, (const __bf16*)Bst, bias, C,
-                       (__bf16*)Cbf, M, N, K, act);
+    if (burst)
+      hipLaunchKernelGGL((gemm256_bf16_kernel<false, true>), grid, block, 0,
+                         stream, (const __bf16*)A, (const __bf16*)Bst, bias,
+                         C, (__bf16*)Cbf, M, N, K, act);
+    else
+      hipLaunchKernelGGL((gemm256_bf16_kernel<false, false>), grid, block, 0,
+                         stream, (const __bf16*)A, (const __bf16*)Bst, bias,
+                         C, (__bf16*)Cbf, M, N, K, act);
   } else {
-    hipLaunchKernelGGL((gemm256_bf16_kernel<true>), grid, block, 0, stream,
-                       (const __bf16*)A, (const __bf16*)Bst, bias, C,
-                       (__bf16*)Cbf, M, N, K, act);
+    if (burst)
+      hipLaunchKernelGGL((gemm256_bf16_kernel<true, true>), grid, block, 0,
+                         stream, (const __bf16*)A, (const __bf16*)Bst, bias,
+                         C, (__bf16*)Cbf, M, N, K, act);
+    else
+      hipLaunchKernelGGL((gemm256_bf16_kernel<true, false>), grid, block, 0,
+                         stream, (const __bf16*)A, (const __bf16*)Bst, bias,
+                         C, (__bf16*)Cbf, M, N, K, act);
   }
 }
 
