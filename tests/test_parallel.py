@@ -526,3 +526,60 @@ def test_sharded_fm_pipelined():
     assert ok_eq, f"pipelined(no overlap) != sync: {sl} vs {pl}"
     assert ok_first, f"SSP-1 first step differs: {sl[0]} vs {ol[0]}"
     assert ok_finite, f"SSP-1 losses not finite: {ol}"
+
+
+def _sharded_fm_w3_worker(rank, port, q):
+    try:
+        import torch.distributed as dist
+
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        dist.init_process_group("gloo", rank=rank, world_size=3)
+        from lightctr_amd.models.fm import FMHyper
+        from lightctr_amd.parallel.sharded_fm import ShardedFMModel
+        from conftest import make_random_csr
+
+        # odd world + F not divisible by world: exercises shard rounding
+        h = FMHyper(num_features=1001, k=8, optimizer="adagrad", seed=7)
+        m = ShardedFMModel(h, device="cpu")
+        for step in range(2):
+            rp, fi, v, lb = make_random_csr(B=48, F_total=1001,
+                                            seed=step * 3 + rank,
+                                            binary_vals=False)
+            loss = m.train_step(rp, fi, v, lb)
+            assert torch.isfinite(loss).all()
+        # sharded checkpoint round-trip: save, perturb, load, compare
+        import tempfile
+
+        d = tempfile.mkdtemp()
+        prefix = os.path.join(d, "ck")
+        m.save(prefix)
+        W0, V0 = m.W.clone(), m.V.clone()
+        m.W.add_(1.0)
+        m.V.mul_(0.5)
+        m.load(prefix)
+        ok = (torch.equal(m.W, W0) and torch.equal(m.V, V0))
+        # one more step after resume must still work
+        rp, fi, v, lb = make_random_csr(B=48, F_total=1001, seed=99 + rank,
+                                        binary_vals=False)
+        loss = m.train_step(rp, fi, v, lb)
+        ok = ok and bool(torch.isfinite(loss).all())
+        if rank == 0:
+            q.put(("result", ok))
+        dist.destroy_process_group()
+    except Exception:  # pragma: no cover
+        import traceback
+
+        q.put(("error", rank, traceback.format_exc()))
+        raise
+
+
+def test_sharded_fm_world3_and_checkpoint():
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    mp.start_processes(_sharded_fm_w3_worker, args=(29541, q), nprocs=3,
+                       join=True, start_method="spawn")
+    assert not q.empty()
+    msg = q.get()
+    assert msg[0] == "result", f"worker error: {msg}"
+    assert msg[1], "world-3 sharded train/checkpoint failed"
