@@ -206,3 +206,51 @@ def test_bundled_datasets_train():
     X, y = load_dense_csv(os.path.join(root, "data", "train_dense.csv"),
                           scale=1.0)
     assert X.shape == (400, 256) and y.numel() == 400
+
+
+def test_libffm_native_python_parser_parity(tmp_path):
+    """The native C++ parser and the Python fallback agree on a fuzzed
+    file with blank lines, exponent floats and a max_rows cut."""
+    import random
+
+    import torch
+
+    from lightctr_amd.data import libffm as lf
+    from lightctr_amd.ops._extension import has_hip_ops
+
+    if not has_hip_ops():
+        import pytest
+
+        pytest.skip("native extension not built")
+
+    rnd = random.Random(5)
+    path = tmp_path / "fuzz.csv"
+    with open(path, "w") as f:
+        for i in range(60):
+            if i % 13 == 7:
+                f.write("\n")  # blank line must be skipped
+            n = rnd.randint(1, 9)
+            toks = [str(rnd.randint(0, 1))]
+            for _ in range(n):
+                toks.append(f"{rnd.randint(0, 30)}:{rnd.randint(0, 5000)}:"
+                            f"{rnd.uniform(-2, 2):.4g}")
+            f.write(" ".join(toks) + ("  \n" if i % 5 == 0 else "\n"))
+
+    for max_rows in (None, 17):
+        nat = lf.load_libffm(str(path), max_rows=max_rows)  # native path
+
+        # force the python fallback: temporarily hide the extension
+        from lightctr_amd.ops import _extension as ext
+
+        orig = ext.has_hip_ops
+        ext.has_hip_ops = lambda: False
+        try:
+            py = lf.load_libffm(str(path), max_rows=max_rows)
+        finally:
+            ext.has_hip_ops = orig
+
+        assert torch.equal(nat.row_ptr, py.row_ptr), max_rows
+        assert torch.equal(nat.fields, py.fields)
+        assert torch.equal(nat.fids, py.fids)
+        assert torch.allclose(nat.vals, py.vals, atol=1e-6)
+        assert torch.equal(nat.labels, py.labels)
