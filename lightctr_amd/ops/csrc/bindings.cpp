@@ -31,6 +31,15 @@ ihipStream_t* cur_stream() {
   return (ihipStream_t*)at::cuda::getCurrentCUDAStream().stream();
 }
 
+// sort permutations are int32 end-to-end (half the radix payload of
+// torch.sort's int64 indices); accept int64 for compatibility (e.g.
+// torch.sort callers in tests) by converting.
+at::Tensor perm32(const at::Tensor& perm) {
+  if (perm.scalar_type() == at::kInt) return perm;
+  CHK(perm.scalar_type() == at::kLong, "perm must be int32 or int64");
+  return perm.to(at::kInt);
+}
+
 // ---- FM ----
 
 std::vector<at::Tensor> fm_forward(at::Tensor row_ptr, at::Tensor fids,
@@ -114,13 +123,12 @@ std::vector<at::Tensor> fm_backward_emit(at::Tensor row_ptr, at::Tensor fids,
 }
 
 // inverse permutation (write slots for scatter-emit): inv[perm[i]] = i
-at::Tensor inv_perm_i32(at::Tensor perm) {
-  CHK(perm.is_cuda() && perm.scalar_type() == at::kLong &&
-          perm.is_contiguous(),
-      "perm must be contiguous cuda int64");
+at::Tensor inv_perm_i32(at::Tensor perm_in) {
+  auto perm = perm32(perm_in).contiguous();
+  CHK(perm.is_cuda(), "perm must be on GPU");
   const int n = (int)perm.numel();
-  auto inv = at::empty({n}, perm.options().dtype(at::kInt));
-  lightctr::inv_perm_launch(perm.data_ptr<long>(), inv.data_ptr<int>(), n,
+  auto inv = at::empty({n}, perm.options());
+  lightctr::inv_perm_launch(perm.data_ptr<int>(), inv.data_ptr<int>(), n,
                             cur_stream());
   return inv;
 }
@@ -129,10 +137,11 @@ void fm_sorted_apply(at::Tensor sorted_fids, c10::optional<at::Tensor> perm,
                      at::Tensor gw, at::Tensor gv, at::Tensor gradW,
                      at::Tensor gradV, at::Tensor touched, int64_t chunk) {
   check_cuda_i32(sorted_fids, "sorted_fids");
-  const long* perm_ptr = nullptr;
+  const int* perm_ptr = nullptr;
+  at::Tensor perm_i;
   if (perm.has_value()) {
-    CHK(perm->scalar_type() == at::kLong, "perm must be int64");
-    perm_ptr = perm->data_ptr<long>();
+    perm_i = perm32(*perm).contiguous();
+    perm_ptr = perm_i.data_ptr<int>();
   }
   const int K = (int)gradV.size(1);
   lightctr::fm_sorted_apply_launch(
@@ -155,10 +164,11 @@ void fm_sorted_apply_fused(at::Tensor sorted_fids,
                            c10::optional<at::Tensor> zV, int64_t opt_mode,
                            double p0, double p1, double p2, double p3) {
   check_cuda_i32(sorted_fids, "sorted_fids");
-  const long* perm_ptr = nullptr;
+  const int* perm_ptr = nullptr;
+  at::Tensor perm_i;
   if (perm.has_value()) {
-    CHK(perm->scalar_type() == at::kLong, "perm must be int64");
-    perm_ptr = perm->data_ptr<long>();
+    perm_i = perm32(*perm).contiguous();
+    perm_ptr = perm_i.data_ptr<int>();
   }
   CHK(opt_mode == 1 || opt_mode == 2, "opt_mode 1=adagrad 2=ftrl");
   const int K = (int)gradV.size(1);
@@ -244,11 +254,11 @@ void ffm_blocks_apply_f16(at::Tensor sorted_fids, at::Tensor perm,
                           at::Tensor gblocks, at::Tensor gw, at::Tensor gradW,
                           at::Tensor gradV, at::Tensor touched) {
   check_cuda_i32(sorted_fids, "sorted_fids");
-  CHK(perm.scalar_type() == at::kLong, "perm must be int64");
+  auto perm_i = perm32(perm).contiguous();
   CHK(gblocks.scalar_type() == at::kHalf, "gblocks must be fp16");
   const int D = (int)gradV.size(1);
   lightctr::ffm_blocks_apply_f16_launch(
-      sorted_fids.data_ptr<int>(), perm.data_ptr<long>(), gblocks.data_ptr(),
+      sorted_fids.data_ptr<int>(), perm_i.data_ptr<int>(), gblocks.data_ptr(),
       gw.data_ptr<float>(), gradW.data_ptr<float>(), gradV.data_ptr<float>(),
       (unsigned long long*)touched.data_ptr(), D, (int)sorted_fids.numel(),
       cur_stream());
@@ -274,12 +284,12 @@ void ffm_sorted_backward(at::Tensor sorted_fids, at::Tensor perm,
                          at::Tensor V, at::Tensor dpred, at::Tensor gradW,
                          at::Tensor gradV, at::Tensor touched) {
   check_cuda_i32(sorted_fids, "sorted_fids");
-  CHK(perm.scalar_type() == at::kLong, "perm must be int64");
+  auto perm_i = perm32(perm).contiguous();
   check_cuda_i32(row_of_entry, "row_of_entry");
   const int nfields = (int)V.size(1);
   const int K = (int)V.size(2);
   lightctr::ffm_sorted_backward_launch(
-      sorted_fids.data_ptr<int>(), perm.data_ptr<long>(),
+      sorted_fids.data_ptr<int>(), perm_i.data_ptr<int>(),
       row_of_entry.data_ptr<int>(), row_ptr.data_ptr<int>(),
       fields.data_ptr<int>(), fids.data_ptr<int>(), vals.data_ptr<float>(),
       V.data_ptr<float>(), dpred.data_ptr<float>(), gradW.data_ptr<float>(),
@@ -311,10 +321,10 @@ void ffm_blocks_apply(at::Tensor sorted_fids, at::Tensor perm,
                       at::Tensor gblocks, at::Tensor gw, at::Tensor gradW,
                       at::Tensor gradV, at::Tensor touched) {
   check_cuda_i32(sorted_fids, "sorted_fids");
-  CHK(perm.scalar_type() == at::kLong, "perm must be int64");
+  auto perm_i = perm32(perm).contiguous();
   const int D = (int)gblocks.size(1);
   lightctr::ffm_blocks_apply_launch(
-      sorted_fids.data_ptr<int>(), perm.data_ptr<long>(),
+      sorted_fids.data_ptr<int>(), perm_i.data_ptr<int>(),
       gblocks.data_ptr<float>(), gw.data_ptr<float>(),
       gradW.data_ptr<float>(), gradV.data_ptr<float>(),
       (unsigned long long*)touched.data_ptr(), D,
@@ -483,15 +493,15 @@ std::vector<at::Tensor> radix_sort_index(at::Tensor keys, int64_t end_bit) {
   CHK(end_bit >= 1 && end_bit <= 32, "end_bit in [1,32]");
   const int n = (int)keys.numel();
   auto sorted = at::empty_like(keys);
-  auto perm = at::empty({n}, keys.options().dtype(at::kLong));
-  auto iota = at::empty({n}, keys.options().dtype(at::kLong));
-  lightctr::iota_i64_launch(iota.data_ptr<long>(), n, cur_stream());
+  auto perm = at::empty({n}, keys.options());
+  auto iota = at::empty({n}, keys.options());
+  lightctr::iota_i32_launch(iota.data_ptr<int>(), n, cur_stream());
   const unsigned long bytes =
       lightctr::radix_sort_pairs_i32_temp_bytes(n, (int)end_bit);
   auto temp = at::empty({(long)bytes}, keys.options().dtype(at::kByte));
   lightctr::radix_sort_pairs_i32_launch(
       temp.data_ptr(), bytes, keys.data_ptr<int>(), sorted.data_ptr<int>(),
-      iota.data_ptr<long>(), perm.data_ptr<long>(), n, (int)end_bit,
+      iota.data_ptr<int>(), perm.data_ptr<int>(), n, (int)end_bit,
       cur_stream());
   return {sorted, perm};
 }

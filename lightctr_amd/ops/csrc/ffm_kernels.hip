@@ -170,7 +170,7 @@ __global__ void ffm_backward_kernel(
 // ---------------------------------------------------------------------------
 template <int K>
 __global__ void ffm_sorted_backward_kernel(
-    const int* __restrict__ sorted_fids, const long* __restrict__ perm,
+    const int* __restrict__ sorted_fids, const int* __restrict__ perm,
     const int* __restrict__ row_of_entry, const int* __restrict__ row_ptr,
     const int* __restrict__ fields, const int* __restrict__ fids,
     const float* __restrict__ vals, const float* __restrict__ V,
@@ -235,7 +235,7 @@ __global__ void ffm_sorted_backward_kernel(
         atomicOr(&touched[fid >> 6], 1ull << (fid & 63));
       }
     }
-    const int p = (int)perm[e];
+    const int p = perm[e];
     const int r = row_of_entry[p];
     const int Fi = fields[p];
     const float xi = vals[p];
@@ -311,7 +311,7 @@ __global__ void ffm_block_emit_kernel(
 
 // Phase 2: segment-reduce fid-sorted blocks into the grad slabs.
 __global__ void ffm_blocks_apply_kernel(
-    const int* __restrict__ sorted_fids, const long* __restrict__ perm,
+    const int* __restrict__ sorted_fids, const int* __restrict__ perm,
     const float* __restrict__ gblocks, const float* __restrict__ gw,
     float* __restrict__ gradW, float* __restrict__ gradV,
     unsigned long long* __restrict__ touched, int D, int nnz, int chunk) {
@@ -345,7 +345,7 @@ __global__ void ffm_blocks_apply_kernel(
         atomicOr(&touched[fid >> 6], 1ull << (fid & 63));
       }
     }
-    const long p = perm[e];
+    const long p = (long)perm[e];
     for (int i = lane; i < D; i += LCTR_WAVE)
       acc[i] += gblocks[(size_t)p * D + i];
     if (lane == 0) accw += gw[p];
@@ -507,7 +507,7 @@ __global__ void ffm_row_emit_kernel(
 // coalesced store sweep (slabs are zeroed by the optimizer pass); only
 // chunk-spanning runs use atomicAdd.
 __global__ void ffm_blocks_apply_f16_kernel(
-    const int* __restrict__ sorted_fids, const long* __restrict__ perm,
+    const int* __restrict__ sorted_fids, const int* __restrict__ perm,
     const _Float16* __restrict__ gblocks, const float* __restrict__ gw,
     float* __restrict__ gradW, float* __restrict__ gradV,
     unsigned long long* __restrict__ touched, int D, int nnz, int chunk) {
@@ -551,7 +551,7 @@ __global__ void ffm_blocks_apply_f16_kernel(
       if (lane == 0 && head_ok)
         atomicOr(&touched[fid >> 6], 1ull << (fid & 63));
     }
-    const long p = perm[e];
+    const long p = (long)perm[e];
     const __half2* gb2 = (const __half2*)&gblocks[(size_t)p * D];
     for (int i = lane; i < D / 2; i += LCTR_WAVE) {
       const float2 t = __half22float2(gb2[i]);
@@ -601,7 +601,7 @@ void ffm_row_emit_launch(const int* row_ptr, const int* fields,
                                        nfields, B, maxn));
 }
 
-void ffm_blocks_apply_f16_launch(const int* sorted_fids, const long* perm,
+void ffm_blocks_apply_f16_launch(const int* sorted_fids, const int* perm,
                                  const void* gblocks, const float* gw,
                                  float* gradW, float* gradV,
                                  unsigned long long* touched, int D, int nnz,
@@ -647,7 +647,7 @@ void ffm_block_emit_launch(const int* row_of_entry, const int* row_ptr,
                                        gblocks, gw, nfields, nnz));
 }
 
-void ffm_blocks_apply_launch(const int* sorted_fids, const long* perm,
+void ffm_blocks_apply_launch(const int* sorted_fids, const int* perm,
                              const float* gblocks, const float* gw,
                              float* gradW, float* gradV,
                              unsigned long long* touched, int D, int nnz,
@@ -674,7 +674,7 @@ __global__ void row_index_kernel(const int* __restrict__ row_ptr,
     row_idx[j] = row;
 }
 
-void ffm_sorted_backward_launch(const int* sorted_fids, const long* perm,
+void ffm_sorted_backward_launch(const int* sorted_fids, const int* perm,
                                 const int* row_of_entry, const int* row_ptr,
                                 const int* fields, const int* fids,
                                 const float* vals, const float* V,

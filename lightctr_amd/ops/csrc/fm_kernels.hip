@@ -182,7 +182,7 @@ __global__ void fm_backward_emit_kernel(
 
 // inv[perm[i]] = i : build the emit kernel's write-slot array from the
 // sort permutation (sequential reads of perm, scattered int32 stores).
-__global__ void inv_perm_kernel(const long* __restrict__ perm,
+__global__ void inv_perm_kernel(const int* __restrict__ perm,
                                 int* __restrict__ inv, int n) {
   const int i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i < n) inv[perm[i]] = i;
@@ -212,7 +212,7 @@ struct FmOptArgs {
 
 template <int K>
 __global__ void fm_sorted_apply_kernel(
-    const int* __restrict__ sorted_fids, const long* __restrict__ perm,
+    const int* __restrict__ sorted_fids, const int* __restrict__ perm,
     const float* __restrict__ gw, const float* __restrict__ gv,
     float* __restrict__ gradW, float* __restrict__ gradV,
     unsigned long long* __restrict__ touched, int nnz, int chunk,
@@ -293,7 +293,7 @@ __global__ void fm_sorted_apply_kernel(
         f[u] = sorted_fids[e + u];
         // perm == nullptr: gw/gv were emitted directly in sorted order
         // (scatter-emit path) — sequential reads, no gather
-        const long p = perm ? perm[e + u] : (long)(e + u);
+        const long p = perm ? (long)perm[e + u] : (long)(e + u);
         v[u] = gv[(size_t)p * K + k];
         vw[u] = (k == 0) ? gw[p] : 0.f;
       }
@@ -329,7 +329,7 @@ __global__ void fm_sorted_apply_kernel(
 // ---------------------------------------------------------------------------
 template <int K>
 __global__ void fm_segscan_apply_kernel(
-    const int* __restrict__ sorted_fids, const long* __restrict__ perm,
+    const int* __restrict__ sorted_fids, const int* __restrict__ perm,
     const float* __restrict__ gw, const float* __restrict__ gv,
     float* __restrict__ gradW, float* __restrict__ gradV,
     unsigned long long* __restrict__ touched, int nnz) {
@@ -349,7 +349,7 @@ __global__ void fm_segscan_apply_kernel(
 
   float v[K + 1];  // [0..K) = gv row, [K] = gw
   if (valid) {
-    const long p = perm ? perm[e] : e;
+    const long p = perm ? (long)perm[e] : e;
     const float4* src = (const float4*)&gv[(size_t)p * K];
 #pragma unroll
     for (int q = 0; q < K / 4; ++q) {
@@ -580,14 +580,14 @@ void fm_backward_emit_launch(const int* row_ptr, const int* fids,
                                    dpred, gw, gv, B, pos));
 }
 
-void inv_perm_launch(const long* perm, int* inv, int n, hipStream_t stream) {
+void inv_perm_launch(const int* perm, int* inv, int n, hipStream_t stream) {
   if (n <= 0) return;
   dim3 block(256);
   dim3 grid((n + 255) / 256);
   hipLaunchKernelGGL(inv_perm_kernel, grid, block, 0, stream, perm, inv, n);
 }
 
-void fm_sorted_apply_launch(const int* sorted_fids, const long* perm,
+void fm_sorted_apply_launch(const int* sorted_fids, const int* perm,
                             const float* gw, const float* gv, float* gradW,
                             float* gradV, unsigned long long* touched, int nnz,
                             int K, int opt_mode, float* V, float* W,

@@ -23,8 +23,8 @@ void fm_backward_emit_launch(const int* row_ptr, const int* fids,
                              const float* sumVX, const float* dpred, float* gw,
                              float* gv, int B, int K, const int* pos,
                              ihipStream_t* stream);
-void inv_perm_launch(const long* perm, int* inv, int n, ihipStream_t* stream);
-void fm_sorted_apply_launch(const int* sorted_fids, const long* perm,
+void inv_perm_launch(const int* perm, int* inv, int n, ihipStream_t* stream);
+void fm_sorted_apply_launch(const int* sorted_fids, const int* perm,
                             const float* gw, const float* gv, float* gradW,
                             float* gradV, unsigned long long* touched, int nnz,
                             int K, int opt_mode, float* V, float* W,
@@ -46,7 +46,7 @@ void ffm_backward_launch(const int* row_ptr, const int* fields,
                          unsigned long long* touched, int nfields, int B,
                          int K, ihipStream_t* stream);
 
-void ffm_sorted_backward_launch(const int* sorted_fids, const long* perm,
+void ffm_sorted_backward_launch(const int* sorted_fids, const int* perm,
                                 const int* row_of_entry, const int* row_ptr,
                                 const int* fields, const int* fids,
                                 const float* vals, const float* V,
@@ -75,12 +75,12 @@ void ffm_row_emit_launch(const int* row_ptr, const int* fields,
                          const float* dpred, void* gblocks, float* gw,
                          int nfields, int B, int maxn, int K,
                          ihipStream_t* stream);
-void ffm_blocks_apply_f16_launch(const int* sorted_fids, const long* perm,
+void ffm_blocks_apply_f16_launch(const int* sorted_fids, const int* perm,
                                  const void* gblocks, const float* gw,
                                  float* gradW, float* gradV,
                                  unsigned long long* touched, int D, int nnz,
                                  ihipStream_t* stream);
-void ffm_blocks_apply_launch(const int* sorted_fids, const long* perm,
+void ffm_blocks_apply_launch(const int* sorted_fids, const int* perm,
                              const float* gblocks, const float* gw,
                              float* gradW, float* gradV,
                              unsigned long long* touched, int D, int nnz,
@@ -184,11 +184,11 @@ void lowbit_decode_launch(const unsigned int* words, float* x, float lo,
                           float hi, int bits, long n, ihipStream_t* stream);
 
 // --- sort_kernels.hip ---
-void iota_i64_launch(long* out, int n, ihipStream_t* stream);
+void iota_i32_launch(int* out, int n, ihipStream_t* stream);
 unsigned long radix_sort_pairs_i32_temp_bytes(int n, int end_bit);
 void radix_sort_pairs_i32_launch(void* temp, unsigned long temp_bytes,
                                  const int* keys_in, int* keys_out,
-                                 const long* vals_in, long* vals_out, int n,
+                                 const int* vals_in, int* vals_out, int n,
                                  int end_bit, ihipStream_t* stream);
 
 // --- gbm_kernels.hip ---

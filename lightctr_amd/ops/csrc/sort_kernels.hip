@@ -10,42 +10,41 @@
 // segment-reduce does not even require (any order within an equal-key
 // run accumulates the same), so correctness is a strict superset.
 //
-// Values are int64 so (sorted, perm) is a bit-for-bit drop-in for
-// torch.sort(fids) at the fm_sorted_apply call sites (perm is long*
-// there). Host-side wrappers live here because rocPRIM instantiates
-// device kernels (must be compiled by hipcc); tensor plumbing stays in
-// bindings.cpp.
+// Values are int32 (nnz < 2^31): half the payload torch.sort's int64
+// indices move through the 2-3 radix passes. Host-side wrappers live
+// here because rocPRIM instantiates device kernels (must be compiled by
+// hipcc); tensor plumbing stays in bindings.cpp.
 #include <rocprim/device/device_radix_sort.hpp>
 
 #include "common.h"
 
 namespace lightctr {
 
-__global__ void iota_i64_kernel(long* __restrict__ out, int n) {
+__global__ void iota_i32_kernel(int* __restrict__ out, int n) {
   int i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i < n) out[i] = i;
 }
 
-void iota_i64_launch(long* out, int n, hipStream_t stream) {
+void iota_i32_launch(int* out, int n, hipStream_t stream) {
   if (n <= 0) return;
   int threads = 256;
   int blocks = (n + threads - 1) / threads;
-  hipLaunchKernelGGL(iota_i64_kernel, dim3(blocks), dim3(threads), 0, stream,
+  hipLaunchKernelGGL(iota_i32_kernel, dim3(blocks), dim3(threads), 0, stream,
                      out, n);
 }
 
 size_t radix_sort_pairs_i32_temp_bytes(int n, int end_bit) {
   size_t bytes = 0;
   (void)rocprim::radix_sort_pairs(nullptr, bytes, (const int*)nullptr,
-                                  (int*)nullptr, (const long*)nullptr,
-                                  (long*)nullptr, (size_t)n, 0u,
+                                  (int*)nullptr, (const int*)nullptr,
+                                  (int*)nullptr, (size_t)n, 0u,
                                   (unsigned)end_bit, (hipStream_t)0, false);
   return bytes;
 }
 
 void radix_sort_pairs_i32_launch(void* temp, size_t temp_bytes,
                                  const int* keys_in, int* keys_out,
-                                 const long* vals_in, long* vals_out, int n,
+                                 const int* vals_in, int* vals_out, int n,
                                  int end_bit, hipStream_t stream) {
   LCTR_CHECK_HIP(rocprim::radix_sort_pairs(temp, temp_bytes, keys_in,
                                            keys_out, vals_in, vals_out,
