@@ -50,10 +50,13 @@ def build_model(args, device, world):
     if args.model == "nfm":
         from lightctr_amd.models.nfm import NFMHyper, NFMModel
 
-        assert world == 1
-        return NFMModel(NFMHyper(num_features=args.features, k=args.k,
-                                 hidden=(64,), seed=1234),
-                        device=device), "sparse"
+        hyper = NFMHyper(num_features=args.features, k=args.k,
+                         hidden=(64,), seed=1234)
+        if world > 1:
+            from lightctr_amd.parallel.sharded_nfm import ShardedNFMModel
+
+            return ShardedNFMModel(hyper, device=device), "sparse"
+        return NFMModel(hyper, device=device), "sparse"
     if args.model == "widedeep":
         from lightctr_amd.models.wide_deep import (WideDeepHyper,
                                                    WideDeepModel)

@@ -583,3 +583,74 @@ def test_sharded_fm_world3_and_checkpoint():
     msg = q.get()
     assert msg[0] == "result", f"worker error: {msg}"
     assert msg[1], "world-3 sharded train/checkpoint failed"
+
+
+def _sharded_nfm_worker(rank, port, q):
+    try:
+        dist = _init(rank, port)
+        import torch as t
+        from lightctr_amd.models.nfm import NFMHyper, NFMModel
+        from lightctr_amd.parallel.sharded_nfm import ShardedNFMModel
+        from conftest import make_random_csr
+
+        F, K = 1200, 8
+        h = NFMHyper(num_features=F, k=K, optimizer="adagrad", hidden=(16,),
+                     seed=23, mlp_optimizer="adagrad", mlp_lr=0.01,
+                     dropout=0.0)
+        sharded = ShardedNFMModel(h, device="cpu")
+        batches = [make_random_csr(B=32, F_total=F, seed=60 + 3 * s + rank,
+                                   binary_vals=False) for s in range(2)]
+        for rp, fi, v, lb in batches:
+            loss = sharded.train_step(rp, fi, v, lb)
+            assert t.isfinite(loss).all()
+        gathered_W = [t.zeros_like(sharded.W) for _ in range(WORLD)]
+        gathered_V = [t.zeros_like(sharded.V) for _ in range(WORLD)]
+        dist.all_gather(gathered_W, sharded.W)
+        dist.all_gather(gathered_V, sharded.V)
+        if rank == 0:
+            W_full = t.zeros(F)
+            V_full = t.zeros(F, K)
+            for r in range(WORLD):
+                idx = t.arange(r, F, WORLD)
+                W_full[idx] = gathered_W[r][: idx.numel()]
+                V_full[idx] = gathered_V[r][: idx.numel()]
+            single = NFMModel(h, device="cpu")
+            for r in range(WORLD):
+                g = t.Generator().manual_seed(h.seed + 17 * r)
+                Fl = (F + WORLD - 1) // WORLD
+                Vr = t.randn(Fl, K, generator=g) * h.init_sigma
+                idx = t.arange(r, F, WORLD)
+                single.V[idx] = Vr[: idx.numel()]
+                single.W[idx] = 0.0
+            for s in range(2):
+                bs = [make_random_csr(B=32, F_total=F, seed=60 + 3 * s + r,
+                                      binary_vals=False)
+                      for r in range(WORLD)]
+                rp, fi, v, lb = bs[0]
+                for r in range(1, WORLD):
+                    rp = t.cat([rp[:-1], bs[r][0] + rp[-1]])
+                    fi = t.cat([fi, bs[r][1]])
+                    v = t.cat([v, bs[r][2]])
+                    lb = t.cat([lb, bs[r][3]])
+                single.train_step(rp, fi, v, lb)
+            ok_v = t.allclose(V_full, single.V, atol=1e-5)
+            ok_w = t.allclose(W_full, single.W, atol=1e-5)
+            ok_mlp = t.allclose(sharded.mlp.layers[0].W,
+                                single.mlp.layers[0].W, atol=1e-5)
+            q.put(("result", bool(ok_v and ok_w and ok_mlp),
+                   float((V_full - single.V).abs().max()),
+                   float((sharded.mlp.layers[0].W
+                          - single.mlp.layers[0].W).abs().max())))
+        dist.destroy_process_group()
+    except Exception:  # pragma: no cover
+        import traceback
+
+        q.put(("error", rank, traceback.format_exc()))
+        raise
+
+
+def test_sharded_nfm_matches_single():
+    """Sharded NFM (sharded W/V + SUM-allreduced bi-interaction MLP) ==
+    single model on the union batch."""
+    ok, dv, dmlp = _run_spawn(_sharded_nfm_worker, 29542)
+    assert ok, f"maxdiff V={dv} mlp={dmlp}"
