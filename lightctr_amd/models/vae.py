@@ -13,7 +13,7 @@ from dataclasses import dataclass
 
 import torch
 
-from .mlp import MLP, DenseLayer
+from .mlp import MLP
 
 
 @dataclass

@@ -355,7 +355,7 @@ def ps_train_fm(cfg: PSConfig, group, gen_batch, steps: int,
     distributed_algo_abst.h:130-280). gen_batch(step) -> CSR batch.
     Returns list of per-step mean losses."""
     from ..ops import fm_ref
-    from ..ops._extension import has_hip_ops, require_hip_ops, sort_ids
+    from ..ops._extension import require_hip_ops, sort_ids
 
     worker = PSWorker(cfg, group, device=device)
     losses = []
