@@ -28,9 +28,28 @@ __global__ void embed_gather_kernel(const int* __restrict__ row_ptr,
   const int g = lane / K;
   const int k = lane % K;
   const int beg = row_ptr[row], end = row_ptr[row + 1];
-  for (int j = beg + g; j < end; j += G) {
+  // 4-deep pipelined gathers (latency-bound random E rows; same pattern
+  // as fm_forward_kernel)
+  const int lim = min(end, beg + nf);
+  int j = beg + g;
+  for (; j + 3 * G < lim; j += 4 * G) {
+    int f[4];
+    float x[4], e[4];
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      f[u] = fids[j + u * G];
+      x[u] = vals[j + u * G];
+    }
+#pragma unroll
+    for (int u = 0; u < 4; ++u) e[u] = E[(size_t)f[u] * K + k];
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const int pos = j + u * G - beg;
+      out_bf[((size_t)row * nf + pos) * K + k] = (__bf16)(e[u] * x[u]);
+    }
+  }
+  for (; j < lim; j += G) {
     const int pos = j - beg;
-    if (pos >= nf) break;
     const float v = E[(size_t)fids[j] * K + k] * vals[j];
     out_bf[((size_t)row * nf + pos) * K + k] = (__bf16)v;
   }
@@ -82,7 +101,29 @@ __global__ void nfm_forward_kernel(const int* __restrict__ row_ptr,
   const int k = lane % K;
   const int beg = row_ptr[row], end = row_ptr[row + 1];
   float sVX = 0.f, sV2X2 = 0.f, lin = 0.f;
-  for (int j = beg + g; j < end; j += G) {
+  int j = beg + g;
+  for (; j + 3 * G < end; j += 4 * G) {
+    int f[4];
+    float x[4], v[4], w[4];
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      f[u] = fids[j + u * G];
+      x[u] = vals[j + u * G];
+    }
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      v[u] = V[(size_t)f[u] * K + k];
+      w[u] = (k == 0) ? W[f[u]] : 0.f;
+    }
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const float vx = v[u] * x[u];
+      sVX += vx;
+      sV2X2 += vx * vx;
+      lin += w[u] * x[u];
+    }
+  }
+  for (; j < end; j += G) {
     const int fid = fids[j];
     const float x = vals[j];
     const float vx = V[(size_t)fid * K + k] * x;
@@ -123,7 +164,24 @@ __global__ void nfm_backward_emit_kernel(
   const float dw = dwide[row];
   const float sv = sumVX[(size_t)row * K + k];
   const int beg = row_ptr[row], end = row_ptr[row + 1];
-  for (int j = beg + g; j < end; j += G) {
+  int j = beg + g;
+  for (; j + 3 * G < end; j += 4 * G) {
+    int f[4];
+    float x[4], v[4];
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      f[u] = fids[j + u * G];
+      x[u] = vals[j + u * G];
+    }
+#pragma unroll
+    for (int u = 0; u < 4; ++u) v[u] = V[(size_t)f[u] * K + k];
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      gv[(size_t)(j + u * G) * K + k] = dv * (sv - v[u] * x[u]) * x[u];
+      if (k == 0) gw[j + u * G] = dw * x[u];
+    }
+  }
+  for (; j < end; j += G) {
     const int fid = fids[j];
     const float x = vals[j];
     gv[(size_t)j * K + k] = dv * (sv - V[(size_t)fid * K + k] * x) * x;
