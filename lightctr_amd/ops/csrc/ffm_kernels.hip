@@ -99,7 +99,37 @@ __global__ void ffm_forward_pp_kernel(const int* __restrict__ row_ptr,
 
   float acc = 0.f;
   const int npairs = n * (n - 1) / 2;
-  for (int p = lane; p < npairs; p += LCTR_WAVE) {
+  // 2-deep pair pipeline: both pairs' slices issued before either dot is
+  // consumed (2x the independent bytes in flight per lane)
+  int p = lane;
+  for (; p + LCTR_WAVE < npairs; p += 2 * LCTR_WAVE) {
+    int i0, j0, i1, j1;
+    tri_decode(p, n, &i0, &j0);
+    tri_decode(p + LCTR_WAVE, n, &i1, &j1);
+    const float4* va0 =
+        (const float4*)&V[((size_t)fids[beg + i0] * nfields +
+                           fields[beg + j0]) * K];
+    const float4* vb0 =
+        (const float4*)&V[((size_t)fids[beg + j0] * nfields +
+                           fields[beg + i0]) * K];
+    const float4* va1 =
+        (const float4*)&V[((size_t)fids[beg + i1] * nfields +
+                           fields[beg + j1]) * K];
+    const float4* vb1 =
+        (const float4*)&V[((size_t)fids[beg + j1] * nfields +
+                           fields[beg + i1]) * K];
+    float t0 = 0.f, t1 = 0.f;
+#pragma unroll
+    for (int q = 0; q < K / 4; ++q) {
+      const float4 a0 = va0[q], b0 = vb0[q];
+      const float4 a1 = va1[q], b1 = vb1[q];
+      t0 += a0.x * b0.x + a0.y * b0.y + a0.z * b0.z + a0.w * b0.w;
+      t1 += a1.x * b1.x + a1.y * b1.y + a1.z * b1.z + a1.w * b1.w;
+    }
+    acc += t0 * vals[beg + i0] * vals[beg + j0];
+    acc += t1 * vals[beg + i1] * vals[beg + j1];
+  }
+  for (; p < npairs; p += LCTR_WAVE) {
     int i, j;
     tri_decode(p, n, &i, &j);
     const float4* va =
