@@ -228,6 +228,10 @@ void gemm_bf16_launch(const void* A, const void* Bst, const float* bias,
     gemm256_bf16_launch(A, Bst, bias, C, Cbf, M, N, K, act, stream);
     return;
   }
+  if (gemm256n128_eligible(M, N, K, transA, transB)) {
+    gemm256n128_bf16_launch(A, Bst, bias, C, Cbf, M, N, K, act, stream);
+    return;
+  }
   // degenerate shapes (the MLP's 1-wide output layer): direct kernels
   // instead of 128x128 tiles at ~99% waste (see nn_kernels.hip)
   if (N == 1 && transA == 0) {

@@ -80,7 +80,10 @@ def test_widedeep_cpu_convergence():
                                    (1000, 250, 100), (64, 1, 256),
                                    (513, 129, 65), (512, 256, 128),
                                    (384, 128, 624), (256, 256, 64),
-                                   (512, 512, 192)])
+                                   (512, 512, 192),
+                                   # 256x128-tile path (N%256!=0, K%64==0)
+                                   (512, 384, 128), (256, 624, 256),
+                                   (512, 128, 256)])
 def test_gemm_bf16_parity(shape):
     """MFMA GEMM vs torch bf16 matmul (fp32 accumulate) incl. odd tails.
 
