@@ -243,6 +243,11 @@ void gemm_bf16_launch(const void* A, const void* Bst, const float* bias,
     wrowsum_m1_launch(A, Bst, C, K, N, stream);
     return;
   }
+  if (transA == 1 && transB == 1 && (long)M * N <= 4096 && K >= 8192 &&
+      bias == nullptr && act == 0 && Cbf == nullptr) {
+    small_wgrad_launch(A, Bst, C, M, N, K, stream);
+    return;
+  }
   if (K == 1 && transA == 0) {
     outer_k1_launch(A, Bst, bias, C, Cbf, M, N, act, stream);
     return;

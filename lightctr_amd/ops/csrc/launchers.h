@@ -129,6 +129,8 @@ void gemm256_bf16_launch(const void* A, const void* Bst, const float* bias,
                          ihipStream_t* stream);
 
 // --- nn_kernels.hip ---
+void small_wgrad_launch(const void* Ast, const void* Bst, float* C, int M,
+                        int N, int K, ihipStream_t* stream);
 void gemv_n1_launch(const void* A, const void* b, const float* bias,
                     float* C, void* Cbf, int M, int K, int act,
                     ihipStream_t* stream);

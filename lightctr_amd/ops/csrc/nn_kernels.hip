@@ -263,6 +263,48 @@ __global__ void outer_k1_kernel(const __bf16* __restrict__ A,
   if (Cbf) Cbf[i] = (__bf16)v;
 }
 
+// C[M,N] = sum_k Ast[k,m]*Bst[k,n] for TINY outputs (M*N <= 4096) and
+// deep K (the small MLP's wgrad, e.g. NFM [64,16,65536]): each thread
+// owns ceil(M*N/256) outputs in registers, blocks split K, one atomic
+// per (block, output) at the end. The tile kernels waste >90% of their
+// 128x128 output on these shapes.
+__global__ void small_wgrad_kernel(const __bf16* __restrict__ Ast,
+                                   const __bf16* __restrict__ Bst,
+                                   float* __restrict__ C, int M, int N,
+                                   long K, int kchunk) {
+  const int tid = threadIdx.x;
+  const int MN = M * N;
+  const long kbeg = (long)blockIdx.x * kchunk;
+  const long kend = min(K, kbeg + kchunk);
+  float acc[16];
+  int m[16], n[16];
+  int no = 0;
+  for (int o = tid; o < MN && no < 16; o += 256, ++no) {
+    m[no] = o / N;
+    n[no] = o - m[no] * N;
+    acc[no] = 0.f;
+  }
+  for (long k = kbeg; k < kend; ++k) {
+    const __bf16* ar = Ast + k * M;
+    const __bf16* br = Bst + k * N;
+    for (int i = 0; i < no; ++i)
+      acc[i] += (float)ar[m[i]] * (float)br[n[i]];
+  }
+  for (int i = 0; i < no; ++i)
+    if (acc[i] != 0.f) atomicAdd(&C[(size_t)m[i] * N + n[i]], acc[i]);
+}
+
+void small_wgrad_launch(const void* Ast, const void* Bst, float* C, int M,
+                        int N, int K, hipStream_t stream) {
+  LCTR_CHECK_HIP(
+      hipMemsetAsync(C, 0, (size_t)M * N * sizeof(float), stream));
+  const int blocks = 512;
+  const int kchunk = (K + blocks - 1) / blocks;
+  hipLaunchKernelGGL(small_wgrad_kernel, dim3(blocks), dim3(256), 0,
+                     stream, (const __bf16*)Ast, (const __bf16*)Bst, C, M,
+                     N, (long)K, kchunk);
+}
+
 void gemv_n1_launch(const void* A, const void* b, const float* bias,
                     float* C, void* Cbf, int M, int K, int act,
                     hipStream_t stream) {

@@ -276,6 +276,14 @@ def test_gemm_degenerate_shapes():
     assert torch.allclose(dW, ref, atol=2e-1, rtol=1e-2), \
         (dW - ref).abs().max()
 
+    # small-output wgrad: dW[64,16] = dZ[B,64]^T @ X[B,16], deep K
+    B2 = 16384
+    dZ2 = (torch.randn(B2, 64, generator=g) * 0.1).to(torch.bfloat16).cuda()
+    X2 = (torch.randn(B2, 16, generator=g) * 0.5).to(torch.bfloat16).cuda()
+    dW2 = hip_ops.gemm_bf16(dZ2, X2, None, 64, 16, B2, 1, 1, 0, False)
+    ref2 = dZ2.float().t() @ X2.float()
+    assert torch.allclose(dW2, ref2, atol=5e-1, rtol=1e-2),         (dW2 - ref2).abs().max()
+
     # dgrad: dX[B,In] = dZ[B,1] @ Wt[In,1]^T  (K=1 outer product)
     Wt = (torch.randn(In, 1, generator=g) * 0.5).to(torch.bfloat16).cuda()
     dX = hip_ops.gemm_bf16(dZ, Wt, None, B, In, 1, 0, 0, 0, False)
