@@ -276,10 +276,10 @@ __global__ void small_wgrad_kernel(const __bf16* __restrict__ Ast,
   const int MN = M * N;
   const long kbeg = (long)blockIdx.x * kchunk;
   const long kend = min(K, kbeg + kchunk);
-  float acc[16];
-  int m[16], n[16];
+  float acc[4];
+  int m[4], n[4];
   int no = 0;
-  for (int o = tid; o < MN && no < 16; o += 256, ++no) {
+  for (int o = tid; o < MN && no < 4; o += 1024, ++no) {
     m[no] = o / N;
     n[no] = o - m[no] * N;
     acc[no] = 0.f;
@@ -300,7 +300,7 @@ void small_wgrad_launch(const void* Ast, const void* Bst, float* C, int M,
       hipMemsetAsync(C, 0, (size_t)M * N * sizeof(float), stream));
   const int blocks = 512;
   const int kchunk = (K + blocks - 1) / blocks;
-  hipLaunchKernelGGL(small_wgrad_kernel, dim3(blocks), dim3(256), 0,
+  hipLaunchKernelGGL(small_wgrad_kernel, dim3(blocks), dim3(1024), 0,
                      stream, (const __bf16*)Ast, (const __bf16*)Bst, C, M,
                      N, (long)K, kchunk);
 }
