@@ -108,11 +108,12 @@ class FMModel:
             pred, sumVX = ops.fm_forward(row_ptr, fids, vals, self.W, self.V)
             loss, dpred = ops.logloss_grad(pred, labels, scale)
             if self.backward_mode == "sorted":
-                # contention-free backward: radix-sort the fids, then EMIT
-                # per-entry grads directly into sorted slots (scatter-emit:
-                # randomness rides the store side), then segment-reduce with
-                # sequential reads (see profiles/r01_fm_atomic_backward.txt
-                # and fm_kernels.hip for the evolution)
+                # contention-free backward: emit per-entry grads
+                # (coalesced), bit-range radix-sort the fids, then
+                # segment-reduce into the slabs with interior-store
+                # flushes (see profiles/ and fm_kernels.hip for the
+                # measured evolution incl. the reverted scatter-emit and
+                # segmented-scan variants)
                 sorted_fids, perm = sort_ids(fids, self.h.num_features)
                 gw, gv = ops.fm_backward_emit(row_ptr, fids, vals, self.V,
                                               sumVX, dpred)
