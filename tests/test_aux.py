@@ -254,3 +254,27 @@ def test_libffm_native_python_parser_parity(tmp_path):
         assert torch.equal(nat.fids, py.fids)
         assert torch.allclose(nat.vals, py.vals, atol=1e-6)
         assert torch.equal(nat.labels, py.labels)
+
+
+def test_tools_cpu_smokes():
+    """The measurement tools (train-to-AUC harness, GBM bench) run on CPU
+    with tiny configs — keeps them from rotting between GPU rounds."""
+    import subprocess
+    import sys
+    import os
+
+    root = os.path.join(os.path.dirname(__file__), "..")
+    r = subprocess.run(
+        [sys.executable, os.path.join(root, "tools", "train_auc.py"),
+         "--model", "fm", "--steps", "2", "--batch", "512", "--features",
+         "20000", "--eval-rows", "1024"],
+        capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-500:]
+    assert "held-out AUC" in r.stdout
+
+    r = subprocess.run(
+        [sys.executable, os.path.join(root, "tools", "bench_gbm.py"),
+         "--rows", "2000", "--cols", "16", "--rounds", "3", "--depth", "3"],
+        capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-500:]
+    assert "rounds/s" in r.stdout
