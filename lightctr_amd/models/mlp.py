@@ -26,13 +26,18 @@ _ACTS = {"none": 0, "relu": 1, "sigmoid": 2}
 class DenseLayer:
     def __init__(self, in_dim: int, out_dim: int, act: str = "relu",
                  optimizer: str = "adam", lr: float = 1e-3, l2: float = 0.0,
-                 dropout: float = 0.0, seed: int = 0, device: str = "cpu"):
+                 dropout: float = 0.0, seed: int = 0, device: str = "cpu",
+                 clip: float = 0.0):
         self.in_dim, self.out_dim = in_dim, out_dim
         self.act = act
         self.act_id = _ACTS[act]
         self.optimizer = optimizer
         self.lr, self.l2 = lr, l2
         self.dropout = dropout
+        # elementwise delta clipping to +-clip before the grads are formed
+        # (the reference FC/LSTM error_clip_threshold=15 semantics,
+        # fullyconnLayer.h:130, matrix.h:152-159); 0 disables
+        self.clip = clip
         self.device = torch.device(device)
         g = torch.Generator().manual_seed(seed)
         # He init like the reference FC layer's scaled Gaussian
@@ -77,6 +82,8 @@ class DenseLayer:
         ops = require_hip_ops()
         if self.dropout > 0.0:
             dy = dy * self._mask
+        if self.clip > 0.0:
+            dy = torch.clamp(dy, -self.clip, self.clip)
         dZ, dZbf = ops.act_backward(dy.contiguous(), self._y, self.act_id)
         self._db = ops.colsum(dZ)
         B = dZ.shape[0]
@@ -108,6 +115,8 @@ class DenseLayer:
     def backward_cpu(self, dy: torch.Tensor):
         if self.dropout > 0.0:
             dy = dy * self._mask
+        if self.clip > 0.0:
+            dy = torch.clamp(dy, -self.clip, self.clip)
         if self.act == "relu":
             dZ = dy * (self._y > 0).float()
         elif self.act == "sigmoid":
@@ -185,12 +194,14 @@ class MLP:
     """Stack of DenseLayers with a scalar head (last layer act='none')."""
 
     def __init__(self, dims, acts=None, optimizer="adam", lr=1e-3,
-                 l2: float = 0.0, dropout: float = 0.0, seed=0, device="cpu"):
+                 l2: float = 0.0, dropout: float = 0.0, seed=0, device="cpu",
+                 clip: float = 0.0):
         n = len(dims) - 1
         acts = acts or (["relu"] * (n - 1) + ["none"])
         self.layers = [
             DenseLayer(dims[i], dims[i + 1], acts[i], optimizer, lr, l2,
-                       dropout if i < n - 1 else 0.0, seed + i, device)
+                       dropout if i < n - 1 else 0.0, seed + i, device,
+                       clip=clip)
             for i in range(n)
         ]
         self._gpu = self.layers[0]._gpu

@@ -36,7 +36,11 @@ class _Adagrad:
 class LSTMUnit:
     """Batched 4-gate LSTM with stored sequence history + exact BPTT."""
 
-    def __init__(self, in_dim: int, hidden: int, seed=0, device="cpu"):
+    def __init__(self, in_dim: int, hidden: int, seed=0, device="cpu",
+                 clip: float = 15.0):
+        # elementwise delta clipping at +-clip during BPTT (the reference
+        # LSTM's error_clip_threshold = 15, lstm_unit.h:56,177-179)
+        self.clip = clip
         self.D, self.H = in_dim, hidden
         g = torch.Generator().manual_seed(seed)
         s = (1.0 / (in_dim + hidden)) ** 0.5
@@ -85,6 +89,8 @@ class LSTMUnit:
             z, i, f, g, o, c, tc = self._cache[t]
             c_prev = self._cache[t - 1][5] if t > 0 else torch.zeros_like(c)
             dh = dhs[:, t, :] + dh_next
+            if self.clip > 0.0:
+                dh = torch.clamp(dh, -self.clip, self.clip)
             do = dh * tc
             dc = dh * o * (1 - tc * tc) + dc_next
             di = dc * g

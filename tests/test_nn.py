@@ -290,3 +290,28 @@ def test_gemm_degenerate_shapes():
     ref = dZ.float() @ Wt.float().t()
     assert torch.allclose(dX, ref, atol=1e-3, rtol=1e-2), \
         (dX - ref).abs().max()
+
+
+def test_gradient_clipping():
+    """DenseLayer clips backprop deltas elementwise at +-clip before
+    forming weight grads (the reference FC error_clip_threshold
+    semantics); matches a manual clamp oracle."""
+    from lightctr_amd.models.mlp import DenseLayer
+
+    torch.manual_seed(0)
+    la = DenseLayer(8, 4, act="none", optimizer="adagrad", lr=0.1,
+                    clip=1.0, device="cpu")
+    x = torch.randn(16, 8)
+    la.forward(x, train=True)
+    dy = torch.randn(16, 4) * 100.0  # exploding deltas
+    la.backward(dy)
+    ref_dW = torch.clamp(dy, -1.0, 1.0).t() @ x
+    assert torch.allclose(la._dW, ref_dW, atol=1e-5)
+    assert la._dW.abs().max() <= 16 * 8  # bounded by clip * batch
+
+    # clip=0 disables
+    la2 = DenseLayer(8, 4, act="none", optimizer="adagrad", lr=0.1,
+                     device="cpu")
+    la2.forward(x, train=True)
+    la2.backward(dy)
+    assert torch.allclose(la2._dW, dy.t() @ x, atol=1e-4)
