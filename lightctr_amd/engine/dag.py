@@ -47,8 +47,23 @@ class Node:
         with self._lock:
             if self._fwd_done:
                 return self.value
-            futures = [_EXECUTOR.submit(i.forward_run) for i in self.inputs]
-            vals = [f.result() for f in futures]
+            vals = []
+            if self.inputs:
+                # Run the FIRST input on this thread (graph depth then
+                # consumes the caller's stack, not pool workers) and pool
+                # only the siblings; a pooled task that has not started is
+                # cancelled and run inline. Every blocked result() is
+                # therefore on a RUNNING task, so a saturated pool can
+                # never deadlock the flow (the naive submit-all-and-wait
+                # version deadlocks once depth or width exceeds the pool).
+                futures = [_EXECUTOR.submit(i.forward_run)
+                           for i in self.inputs[1:]]
+                vals.append(self.inputs[0].forward_run())
+                for f, node in zip(futures, self.inputs[1:]):
+                    if f.cancel():
+                        vals.append(node.forward_run())
+                    else:
+                        vals.append(f.result())
             self.value = self.forward_compute(*vals)
             self._fwd_done = True
             return self.value

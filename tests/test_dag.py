@@ -86,3 +86,33 @@ def test_dag_softmax_backward_matches_autograd():
     term.accumulate_grad(dy)
     term.backward_run()
     assert torch.allclose(src.grad, xt.grad, atol=1e-6)
+
+
+def test_deep_and_wide_graph_no_deadlock():
+    """Graphs deeper/wider than the executor pool must complete (the
+    naive submit-all-and-wait scheduling deadlocked at depth > workers;
+    forward_run now runs its first input inline and work-steals queued
+    siblings)."""
+    import torch
+
+    from lightctr_amd.engine.dag import AddOp, AggregateNode, SourceNode
+
+    # chain of 60 adds (depth >> 8 workers)
+    node = SourceNode(torch.ones(2))
+    one = SourceNode(torch.ones(2))
+    for _ in range(60):
+        node = AddOp(node, one)
+    out = node.forward_run()
+    assert torch.allclose(out, torch.full((2,), 61.0))
+
+    # wide fan-in of 24 branches, each a depth-4 chain
+    branches = []
+    for b in range(24):
+        n = SourceNode(torch.full((2,), float(b)))
+        for _ in range(4):
+            n = AddOp(n, one)
+        branches.append(n)
+    agg = AggregateNode(*branches)
+    out = agg.forward_run()
+    expect = sum(float(b) + 4.0 for b in range(24))
+    assert torch.allclose(out, torch.full((2,), expect))
