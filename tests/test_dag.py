@@ -116,3 +116,8 @@ def test_deep_and_wide_graph_no_deadlock():
     out = agg.forward_run()
     expect = sum(float(b) + 4.0 for b in range(24))
     assert torch.allclose(out, torch.full((2,), expect))
+
+    # backward through the deep chain completes and reaches the source
+    node.accumulate_grad(torch.ones(2))
+    node.backward_run()
+    assert hasattr(node, "grad")
