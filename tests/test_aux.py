@@ -278,3 +278,26 @@ def test_tools_cpu_smokes():
         capture_output=True, text=True, timeout=300)
     assert r.returncode == 0, r.stderr[-500:]
     assert "rounds/s" in r.stdout
+
+
+def test_proc_file_split_tool(tmp_path):
+    """The per-rank data sharder splits a libffm file into N shards that
+    cover every row exactly once (the reference's offline splitter)."""
+    import os
+    import subprocess
+    import sys
+
+    src = tmp_path / "d.csv"
+    lines = [f"{i % 2} 0:{i}:1\n" for i in range(103)]
+    src.write_text("".join(lines))
+    root = os.path.join(os.path.dirname(__file__), "..")
+    r = subprocess.run(
+        [sys.executable, os.path.join(root, "tools", "proc_file_split.py"),
+         str(src), "4"], capture_output=True, text=True, timeout=60)
+    assert r.returncode == 0, r.stderr
+    got = []
+    for i in range(1, 5):
+        p = tmp_path / f"d_{i}.csv"
+        assert p.exists()
+        got += p.read_text().splitlines(keepends=True)
+    assert sorted(got) == sorted(lines)
