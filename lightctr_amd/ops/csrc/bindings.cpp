@@ -397,7 +397,8 @@ at::Tensor bitmap_compact(at::Tensor bitmap, at::Tensor out_fids,
   lightctr::bitmap_compact_launch((unsigned long long*)bitmap.data_ptr(),
                                   (int)bitmap.numel(),
                                   out_fids.data_ptr<int>(),
-                                  out_count.data_ptr<int>(), cur_stream());
+                                  out_count.data_ptr<int>(),
+                                  (int)out_fids.numel(), cur_stream());
   return out_count;
 }
 

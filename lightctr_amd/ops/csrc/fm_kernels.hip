@@ -479,7 +479,7 @@ __global__ void fm_segscan_apply_kernel(
 // ---------------------------------------------------------------------------
 __global__ void bitmap_compact_kernel(unsigned long long* __restrict__ bitmap,
                                       int nwords, int* __restrict__ out_fids,
-                                      int* __restrict__ out_count) {
+                                      int* __restrict__ out_count, int cap) {
   // BLOCK-aggregated append (1024 threads = 16 waves): waves prefix-scan
   // locally, wave totals are scanned in LDS, and only thread 0 touches
   // the global counter — one atomic per 65536 features instead of per
@@ -516,7 +516,8 @@ __global__ void bitmap_compact_kernel(unsigned long long* __restrict__ bitmap,
   while (m) {
     const int b = __ffsll((long long)m) - 1;
     m &= m - 1;
-    out_fids[off++] = w * 64 + b;
+    if (off < cap) out_fids[off] = w * 64 + b;  // clamp: never overrun uniq
+    ++off;
   }
 }
 
@@ -698,11 +699,12 @@ void fm_sorted_apply_launch(const int* sorted_fids, const int* perm,
 }
 
 void bitmap_compact_launch(unsigned long long* bitmap, int nwords,
-                           int* out_fids, int* out_count, hipStream_t stream) {
+                           int* out_fids, int* out_count, int cap,
+                           hipStream_t stream) {
   dim3 block(1024);
   dim3 grid((nwords + 1023) / 1024);
   hipLaunchKernelGGL(bitmap_compact_kernel, grid, block, 0, stream, bitmap,
-                     nwords, out_fids, out_count);
+                     nwords, out_fids, out_count, cap);
 }
 
 void fm_adagrad_apply_launch(const int* uniq, const int* count, float* W,

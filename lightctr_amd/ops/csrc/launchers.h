@@ -32,7 +32,7 @@ void fm_sorted_apply_launch(const int* sorted_fids, const int* perm,
                             float p0, float p1, float p2, float p3,
                             int chunk, ihipStream_t* stream);
 void bitmap_compact_launch(unsigned long long* bitmap, int nwords,
-                           int* out_fids, int* out_count,
+                           int* out_fids, int* out_count, int cap,
                            ihipStream_t* stream);
 
 // --- ffm_kernels.hip ---

@@ -107,13 +107,14 @@ class ShmHashTable:
             % self.slots
 
     def put(self, key: int, value) -> bool:
-        key = int(key) | 1  # 0 = empty marker; force nonzero
+        # 0 is the empty-slot marker; shift by 1 (injective) instead of
+        # OR-ing bit 0, which aliased every even key k with k+1.
+        key = int(key) + 1
+        assert 0 < key <= 0xFFFFFFFFFFFFFFFF, "key out of u64 range"
         for t in range(self.n_tables):
             s = self._slot(key, t)
             k = int(self._mm_keys[s])
             if k == 0 or k == key:
-                kw = np.frombuffer(self._mm, dtype=np.uint64, offset=16,
-                                   count=len(self._mm_keys))
                 # numpy view is read-only from mmap buffer in some paths;
                 # write through the mmap directly
                 off = 16 + s * 8
@@ -125,7 +126,7 @@ class ShmHashTable:
         return False  # all sub-tables collided
 
     def get(self, key: int):
-        key = int(key) | 1
+        key = int(key) + 1
         for t in range(self.n_tables):
             s = self._slot(key, t)
             if int(self._mm_keys[s]) == key:

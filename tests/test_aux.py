@@ -120,6 +120,14 @@ def test_shm_hashtable(tmp_path):
         v = t.get(k * 977 + 3)
         assert v is not None and abs(v[0] - k) < 1e-6
     assert t.get(123456789) is None
+    # adjacent even/odd keys must not alias (round-1 advisor finding:
+    # `key | 1` mapped 2 and 3 to the same slot key)
+    assert t.put(2, np.array([20.0, 0.0], dtype=np.float32))
+    assert t.put(3, np.array([30.0, 0.0], dtype=np.float32))
+    assert t.get(2)[0] == 20.0 and t.get(3)[0] == 30.0
+    # key 0 is representable (maps to stored 1, still nonzero)
+    assert t.put(0, np.array([7.0, 0.0], dtype=np.float32))
+    assert t.get(0)[0] == 7.0
     t.close()
     # reopen persists
     t2 = ShmHashTable(p, slots_per_table=64, n_tables=4, value_dim=2)
