@@ -132,7 +132,7 @@ def test_shm_hashtable(tmp_path):
     # reopen persists
     t2 = ShmHashTable(p, slots_per_table=64, n_tables=4, value_dim=2)
     v = t2.get(3)
-    assert v is not None and v[0] == 0
+    assert v is not None and v[0] == 30.0
     t2.close()
 
 
