@@ -471,10 +471,22 @@ __global__ void ffm_fwd_staged_kernel(
 // Backward phase 1, per-row staged: emit each entry's [nfields,K]
 // gradient block in HALF precision (the block tensor is the dominant
 // traffic: fp16 halves both the emit write and the apply read; gradient
-// noise >> fp16 rounding). Waves of the workgroup take entries
-// round-robin; each accumulates the entry block in its private LDS slice
-// (ds atomics absorb duplicate fields) reading partner slices from the
-// shared stage.
+// noise >> fp16 rounding).
+//
+// Round-2 rewrite (docs/STATUS.md lever 1, ISA-audit driven): the round-1
+// version compiled to fully scalar staging (one global_load_dword +
+// ds_write_b16 per element), scalar ds_read_b16 partner reads behind LDS
+// atomics, and 2-byte global_store_short emits. This version:
+//   * stages with float4 loads packed to half2 pairs -> one ds_write_b64
+//     per 4 elements (4x fewer instructions, full 16 B coalescing)
+//   * builds a field -> entry map for the row; on the (overwhelmingly
+//     common) duplicate-free row the emit needs NO accumulation at all:
+//     output quad (f, 4k) is  d*x_i*x_j * V_h[fid_j, F_i, 4k]  for the
+//     single partner j in field f — read with one ds_read_b64
+//   * stores the block as uint2 (dwordx2, 512 B/wave/inst) instead of
+//     global_store_short (128 B/wave/inst)
+// Rows with duplicate fields (or n > maxn) fall back to the round-1
+// atomic-accumulate loop, which stays bit-correct for that case.
 template <int K>
 __global__ void ffm_row_emit_kernel(
     const int* __restrict__ row_ptr, const int* __restrict__ fields,
@@ -482,10 +494,9 @@ __global__ void ffm_row_emit_kernel(
     const float* __restrict__ V, const float* __restrict__ dpred,
     _Float16* __restrict__ gblocks, float* __restrict__ gw, int nfields,
     int B, int maxn) {
-  // stage holds fp16 (halves the LDS footprint -> 2x the workgroups per
-  // CU; the emitted blocks are fp16 anyway, so staging precision is not
-  // the bottleneck term); acc stays fp32.
-  extern __shared__ float lds[];  // acc[nw][nf*(K+1)] | stage_h[maxn*D]
+  // LDS layout: acc[nw][nf*(K+1)] fp32 (dup fallback only) | stage_h
+  // [maxn*D] fp16 | vals_s[maxn] fp32 | fmap[nfields] i32 | dup flag
+  extern __shared__ float lds[];
   constexpr int G = LCTR_WAVE / K;
   const int row = blockIdx.x;
   if (row >= B) return;
@@ -501,17 +512,72 @@ __global__ void ffm_row_emit_kernel(
   const int fstride = K + 1;
   float* acc = lds + (size_t)wv * nfields * fstride;
   _Float16* stage = (_Float16*)(lds + (size_t)nw * nfields * fstride);
+  float* vals_s = (float*)(stage + (size_t)maxn * D);
+  int* fmap = (int*)(vals_s + maxn);
+  int* dupflag = fmap + nfields;
   const float d = dpred[row];
 
   const bool staged = n <= maxn;
   if (staged) {
-    for (int idx = tid; idx < n * D; idx += blockDim.x) {
-      const int e = idx / D, q = idx - e * D;
-      stage[idx] = (_Float16)V[(size_t)fids[beg + e] * D + q];
+    // vectorized stage: float4 V load -> 2x half2 -> one 8 B LDS store
+    const int nq = D / 4;  // K % 4 == 0 always (K in {4,8,16,32,64})
+    uint2* st2 = (uint2*)stage;
+    for (int idx = tid; idx < n * nq; idx += blockDim.x) {
+      const int e = idx / nq, q = idx - e * nq;
+      const float4 v4 = ((const float4*)(V + (size_t)fids[beg + e] * D))[q];
+      union {
+        __half2 h[2];
+        uint2 u;
+      } pk;
+      pk.h[0] = __floats2half2_rn(v4.x, v4.y);
+      pk.h[1] = __floats2half2_rn(v4.z, v4.w);
+      st2[idx] = pk.u;
     }
+  }
+  for (int f = tid; f < nfields; f += blockDim.x) fmap[f] = -1;
+  if (tid == 0) *dupflag = 0;
+  __syncthreads();
+  for (int e = tid; e < n; e += blockDim.x) {
+    vals_s[e] = vals[beg + e];
+    const int old = atomicExch(&fmap[fields[beg + e]], e);
+    if (old >= 0) atomicOr(dupflag, 1);
   }
   __syncthreads();
 
+  if (staged && !*dupflag) {
+    // fast path: every field has at most one entry -> direct quad emit
+    const int nq = D / 4;
+    for (int i = wv; i < n; i += nw) {
+      const int p = beg + i;
+      const int Fi = fields[p];
+      const float dxi = d * vals_s[i];
+      uint2* out2 = (uint2*)(gblocks + (size_t)p * D);
+      for (int u = lane; u < nq; u += LCTR_WAVE) {
+        const int f = (u * 4) / K;
+        const int j = fmap[f];
+        uint2 o = make_uint2(0u, 0u);
+        if (j >= 0 && j != i) {
+          union {
+            __half2 h[2];
+            uint2 u2;
+          } in, ov;
+          in.u2 = *(const uint2*)&stage[(size_t)(j * nfields + Fi) * K +
+                                        ((u * 4) & (K - 1))];
+          const float s = dxi * vals_s[j];
+          const float2 a = __half22float2(in.h[0]);
+          const float2 b = __half22float2(in.h[1]);
+          ov.h[0] = __floats2half2_rn(a.x * s, a.y * s);
+          ov.h[1] = __floats2half2_rn(b.x * s, b.y * s);
+          o = ov.u2;
+        }
+        out2[u] = o;
+      }
+      if (lane == 0) gw[p] = dxi;
+    }
+    return;
+  }
+
+  // fallback: duplicate fields in the row (or row longer than the stage)
   for (int i = wv; i < n; i += nw) {
     for (int t = lane; t < nfields * fstride; t += LCTR_WAVE) acc[t] = 0.f;
     const int p = beg + i;
@@ -582,23 +648,40 @@ __global__ void ffm_blocks_apply_f16_kernel(
         atomicOr(&touched[fid >> 6], 1ull << (fid & 63));
     }
     const long p = (long)perm[e];
-    const __half2* gb2 = (const __half2*)&gblocks[(size_t)p * D];
-    for (int i = lane; i < D / 2; i += LCTR_WAVE) {
-      const float2 t = __half22float2(gb2[i]);
-      acc[2 * i] += t.x;
-      acc[2 * i + 1] += t.y;
+    // quad loads (dwordx2): D % 4 == 0 whenever K % 4 == 0 (all dispatched
+    // K); generic tail kept for safety.
+    const uint2* gb4 = (const uint2*)&gblocks[(size_t)p * D];
+    const int nq = D / 4;
+    for (int i = lane; i < nq; i += LCTR_WAVE) {
+      union {
+        uint2 u2;
+        __half2 h[2];
+      } in;
+      in.u2 = gb4[i];
+      const float2 a = __half22float2(in.h[0]);
+      const float2 b = __half22float2(in.h[1]);
+      acc[4 * i] += a.x;
+      acc[4 * i + 1] += a.y;
+      acc[4 * i + 2] += b.x;
+      acc[4 * i + 3] += b.y;
     }
-    if (D & 1)
-      if (lane == 0) acc[D - 1] += (float)gblocks[(size_t)p * D + D - 1];
+    for (int i = 4 * nq + lane; i < D; i += LCTR_WAVE)
+      acc[i] += (float)gblocks[(size_t)p * D + i];
     if (lane == 0) accw += gw[p];
   }
   flush(end);
 }
 
+static size_t ffm_row_emit_lds_bytes(int nfields, int K, int maxn) {
+  // acc (4 waves, dup fallback) | fp16 stage | vals_s | fmap | dup flag
+  return (size_t)4 * nfields * (K + 1) * sizeof(float) +
+         (size_t)maxn * nfields * K * sizeof(_Float16) +
+         (size_t)maxn * sizeof(float) + (size_t)nfields * sizeof(int) +
+         sizeof(int);
+}
+
 bool ffm_staged_eligible(int nfields, int K, int maxn) {
-  const size_t stage_b = (size_t)maxn * nfields * K * sizeof(_Float16);
-  const size_t acc_b = (size_t)4 * nfields * (K + 1) * sizeof(float);
-  return stage_b + acc_b <= (64 << 10);
+  return ffm_row_emit_lds_bytes(nfields, K, maxn) <= (64 << 10);
 }
 
 void ffm_fwd_staged_launch(const int* row_ptr, const int* fields,
@@ -623,8 +706,7 @@ void ffm_row_emit_launch(const int* row_ptr, const int* fields,
   if (B <= 0) return;
   dim3 block(256);
   dim3 grid(B);
-  const size_t lds = (size_t)4 * nfields * (K + 1) * sizeof(float) +
-                     (size_t)maxn * nfields * K * sizeof(_Float16);
+  const size_t lds = ffm_row_emit_lds_bytes(nfields, K, maxn);
   DISPATCH_FFM_K(K, hipLaunchKernelGGL((ffm_row_emit_kernel<KC>), grid, block,
                                        lds, stream, row_ptr, fields, fids,
                                        vals, V, dpred, (_Float16*)gblocks, gw,

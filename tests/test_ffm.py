@@ -268,6 +268,42 @@ def test_ffm_rowemit_backward_parity():
 
 
 @pytest.mark.gpu
+def test_ffm_rowemit_fastpath_parity():
+    """Duplicate-free rows (one feature per field, the Criteo shape) take
+    the round-2 field-map direct-emit fast path in ffm_row_emit_kernel —
+    no LDS accumulation at all. Parity vs the dense reference."""
+    from lightctr_amd.ops import hip_ops
+
+    g = torch.Generator().manual_seed(77)
+    B, nf, K, F = 128, 13, 8, 4000
+    # one entry per field, fields 0..nf-1 in order, distinct per row
+    fields = torch.arange(nf, dtype=torch.int32).repeat(B).cuda()
+    fids = torch.randint(0, F, (B * nf,), generator=g,
+                         dtype=torch.int32).cuda()
+    vals = (torch.rand(B * nf, generator=g) + 0.5).cuda()
+    row_ptr = (torch.arange(B + 1, dtype=torch.int32) * nf).cuda()
+    labels = (torch.rand(B, generator=g) > 0.5).float().cuda()
+    W = torch.randn(F, generator=g).cuda()
+    V = (torch.randn(F, nf, K, generator=g) * 0.1).cuda()
+    pred = hip_ops.ffm_forward(row_ptr, fields, fids, vals, W, V)
+    _, dpred = hip_ops.logloss_grad(pred, labels, 1.0 / B)
+    gw, gblocks = hip_ops.ffm_row_emit(row_ptr, fields, fids, vals, V, dpred)
+    sorted_fids, perm = torch.sort(fids)
+    gradW = torch.zeros(F).cuda()
+    gradV = torch.zeros(F, nf, K).cuda()
+    touched = torch.zeros((F + 63) // 64, dtype=torch.int64).cuda()
+    hip_ops.ffm_blocks_apply_f16(sorted_fids, perm, gblocks, gw, gradW,
+                                 gradV.view(F, -1), touched)
+    gW_ref, gV_ref = ffm_ref.ffm_backward_ref(row_ptr, fields, fids, vals,
+                                              V, dpred)
+    assert torch.allclose(gradW, gW_ref, atol=1e-5, rtol=1e-4), \
+        (gradW - gW_ref).abs().max()
+    ref_scale = gV_ref.abs().max()
+    assert torch.allclose(gradV, gV_ref, atol=float(ref_scale) * 2e-2,
+                          rtol=2e-2), (gradV - gV_ref).abs().max()
+
+
+@pytest.mark.gpu
 def test_ffm_forward_staged_long_rows():
     """Forward parity on rows spanning the staged/fallback boundary."""
     from lightctr_amd.ops import hip_ops
