@@ -224,6 +224,18 @@ void gemm_bf16_launch(const void* A, const void* Bst, const float* bias,
   // 992/1133 at 4k/8k^3: one barrier per 32-K chunk costs more than the
   // barrier-spanning loads buy at this geometry. v1 stays preferred; v2
   // kept (correct, race-screened) as the documented experiment.
+  if (gemm256p8_eligible(M, N, K, transA, transB)) {
+    // 8-phase counted-vmcnt schedule (round 2). LCTR_GEMM_P8=0 falls back
+    // to the round-1 per-tile-drain kernel for A/B.
+    static const bool p8 = [] {
+      const char* e = getenv("LCTR_GEMM_P8");
+      return !(e && e[0] == '0');
+    }();
+    if (p8) {
+      gemm256p8_bf16_launch(A, Bst, bias, C, Cbf, M, N, K, act, stream);
+      return;
+    }
+  }
   if (gemm256_eligible(M, N, K, transA, transB)) {
     gemm256_bf16_launch(A, Bst, bias, C, Cbf, M, N, K, act, stream);
     return;

@@ -120,6 +120,10 @@ void gemm_bf16_launch(const void* A, const void* Bst, const float* bias,
                       int transB, int act, ihipStream_t* stream);
 
 bool gemm256_eligible(int M, int N, int K, int transA, int transB);
+bool gemm256p8_eligible(int M, int N, int K, int transA, int transB);
+void gemm256p8_bf16_launch(const void* A, const void* Bst, const float* bias,
+                           float* C, void* Cbf, int M, int N, int K, int act,
+                           ihipStream_t* stream);
 bool gemm256n128_eligible(int M, int N, int K, int transA, int transB);
 void gemm256n128_bf16_launch(const void* A, const void* Bst,
                              const float* bias, float* C, void* Cbf, int M,

@@ -83,7 +83,10 @@ def test_widedeep_cpu_convergence():
                                    (512, 512, 192),
                                    # 256x128-tile path (N%256!=0, K%64==0)
                                    (512, 384, 128), (256, 624, 256),
-                                   (512, 128, 256)])
+                                   (512, 128, 256),
+                                   # 256^2 8-phase path: NT=1/2/tail tiles
+                                   (256, 256, 88), (512, 256, 624),
+                                   (256, 512, 40)])
 def test_gemm_bf16_parity(shape):
     """MFMA GEMM vs torch bf16 matmul (fp32 accumulate) incl. odd tails.
 
@@ -248,6 +251,24 @@ def test_gemm_v2_race_screen():
         C = hip_ops.gemm_bf16(A, Bst, None, M, N, K, 0, 0, 0, False)
         assert torch.allclose(C, ref, atol=2e-2 * K ** 0.5, rtol=1e-2), \
             (it, (C - ref).abs().max())
+
+
+@pytest.mark.gpu
+def test_gemm_p8_race_screen():
+    """The 8-phase counted-vmcnt schedule is a new sync structure (guide
+    two-lane discipline): repeated runs at a multi-tile shape and a
+    tail shape must all match the fp32 reference."""
+    from lightctr_amd.ops import hip_ops
+
+    for (M, N, K) in [(1024, 1024, 1024), (512, 256, 624)]:
+        g = torch.Generator().manual_seed(41 + K)
+        A = (torch.randn(M, K, generator=g) * 0.4).to(torch.bfloat16).cuda()
+        Bst = (torch.randn(N, K, generator=g) * 0.4).to(torch.bfloat16).cuda()
+        ref = A.float() @ Bst.float().t()
+        for it in range(10):
+            C = hip_ops.gemm_bf16(A, Bst, None, M, N, K, 0, 0, 0, False)
+            assert torch.allclose(C, ref, atol=2e-2 * K ** 0.5, rtol=1e-2), \
+                (it, M, N, K, (C - ref).abs().max())
 
 
 @pytest.mark.gpu
