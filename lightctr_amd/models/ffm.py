@@ -62,16 +62,17 @@ class FFMModel:
         self.count = torch.zeros(1, dtype=torch.int32, device=self.device)
         self._use_hip = self.device.type == "cuda"
         # backward variants, all parity-tested (tests/test_ffm.py):
-        #   sorted (default) — per-run LDS-block recompute, 4.0 ms
-        #   rowemit — per-row fp16-staged block emit + interior-store
-        #            segment reduce (4.7-5.2 + 1.3 ms measured across
-        #            fp32/fp16 staging and 4/8-wave workgroups: every
-        #            emit flavor lands 4-5 ms, i.e. the bound is the
-        #            per-(entry,partner) accumulate stream itself, not
-        #            V-read scheduling; kept + tested as evidence)
-        #   blocks — wave/entry fp32 emit + block reduce, 4.4+1.1 ms
+        #   rowemit (default since round 2) — per-row staged block emit
+        #            with field-map direct quad stores + interior-store
+        #            segment reduce: 1.26 + 1.56 ms = 2.82 ms backward
+        #            (gpurun_out/ab_ffm_r2.txt) after the float4/half2
+        #            staging + dwordx2 store vectorization; the round-1
+        #            scalar emit measured 4.7-5.2 ms
+        #   sorted — per-run LDS-block recompute, 3.91 ms (random-line
+        #            bandwidth bound per the round-1 PMC/ISA analysis)
+        #   blocks — wave/entry fp32 emit + block reduce, 5.6 ms
         #   atomic — naive scatter (hot-feature serialization)
-        self.backward_mode = "sorted"
+        self.backward_mode = "rowemit"
         if self._use_hip:
             require_hip_ops()
 
