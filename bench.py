@@ -40,7 +40,7 @@ def build_model(args, device, world):
 
         hyper = FFMHyper(num_features=args.features, num_fields=39,
                          k=min(args.k, 8), optimizer=args.optimizer,
-                         seed=1234)
+                         seed=1234, dtype="bf16")
         if world > 1:
             from lightctr_amd.parallel.sharded_ffm import ShardedFFMModel
 
@@ -260,7 +260,8 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16" if args.model in ("nfm", "widedeep") else "fp32",
+            "dtype": "bf16" if args.model in ("ffm", "nfm", "widedeep")
+            else "fp32",
             "data": "synthetic",
             "config": {
                 "model": mdesc[args.model],

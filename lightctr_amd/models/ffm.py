@@ -34,6 +34,12 @@ class FFMHyper:
     ftrl_v: str = "adagrad"  # latent updater under ftrl (see models/fm.py)
     init_sigma: float = 0.01
     seed: int = 1234
+    # "bf16" = BASELINE config #3 precision: fp32 master weights +
+    # optimizer state, bf16 compute mirror of V read by the forward /
+    # backward kernels (refreshed in the fused optimizer for exactly the
+    # touched features). Exceeds the reference's fp16-wire-only contract
+    # (/root/reference/LightCTR/common/float16.h:65-154).
+    dtype: str = "fp32"  # fp32 | bf16
 
 
 class FFMModel:
@@ -61,6 +67,10 @@ class FFMModel:
         self.uniq = torch.zeros(cap, dtype=torch.int32, device=self.device)
         self.count = torch.zeros(1, dtype=torch.int32, device=self.device)
         self._use_hip = self.device.type == "cuda"
+        assert hyper.dtype in ("fp32", "bf16")
+        # bf16 compute mirror (fp32 master stays in self.V)
+        self.Vh = self.V.to(torch.bfloat16) if hyper.dtype == "bf16" \
+            else None
         # backward variants, all parity-tested (tests/test_ffm.py):
         #   rowemit (default since round 2) — per-row staged block emit
         #            with field-map direct quad stores + interior-store
@@ -76,10 +86,20 @@ class FFMModel:
         if self._use_hip:
             require_hip_ops()
 
+    @property
+    def _Vc(self):
+        """compute-side V: the bf16 mirror in bf16 mode, fp32 master
+        otherwise"""
+        return self.Vh if self.Vh is not None else self.V
+
     def forward(self, row_ptr, fields, fids, vals):
         if self._use_hip:
             ops = require_hip_ops()
-            return ops.ffm_forward(row_ptr, fields, fids, vals, self.W, self.V)
+            return ops.ffm_forward(row_ptr, fields, fids, vals, self.W,
+                                   self._Vc)
+        if self.Vh is not None:
+            return ffm_ref.ffm_forward_ref(row_ptr, fields, fids, vals,
+                                           self.W, self.Vh.float())
         return ffm_ref.ffm_forward_ref(row_ptr, fields, fids, vals, self.W,
                                        self.V)
 
@@ -95,28 +115,37 @@ class FFMModel:
                                   self.h.ftrl_beta, self.h.ftrl_l1,
                                   self.h.ftrl_l2,
                                   1 if self.h.ftrl_v == "adagrad" else 0,
-                                  self.h.lr, self.h.eps, self.h.l2)
+                                  self.h.lr, self.h.eps, self.h.l2,
+                                  Vh=self.Vh)
         else:
             ops.sparse_adagrad_apply(live, self.count, self.W, self.V,
                                      self.nW, self.nV, self.gradW, self.gradV,
-                                     self.h.lr, self.h.eps, self.h.l2)
+                                     self.h.lr, self.h.eps, self.h.l2,
+                                     Vh=self.Vh)
 
     def train_step(self, row_ptr, fields, fids, vals, labels) -> torch.Tensor:
         B = row_ptr.numel() - 1
         scale = 1.0 / B
         if self._use_hip:
             ops = require_hip_ops()
-            pred = ops.ffm_forward(row_ptr, fields, fids, vals, self.W, self.V)
+            pred = ops.ffm_forward(row_ptr, fields, fids, vals, self.W,
+                                   self._Vc)
             loss, dpred = ops.logloss_grad(pred, labels, scale)
             if self.backward_mode == "rowemit":
+                # power-of-two block scale keeps 1/B-scaled logloss grads
+                # (~1e-8 at B=65536) inside fp16 normal range; the apply
+                # divides it back out exactly
+                bscale = float(1 << min(24, max(0, B.bit_length() - 1)))
                 gw, gblocks = ops.ffm_row_emit(row_ptr, fields, fids, vals,
-                                               self.V, dpred)
+                                               self._Vc, dpred,
+                                               scale=bscale)
                 sorted_fids, perm = sort_ids(fids, self.h.num_features)
                 ops.ffm_blocks_apply_f16(sorted_fids, perm, gblocks, gw,
                                          self.gradW,
                                          self.gradV.view(
                                              self.h.num_features, -1),
-                                         self.touched)
+                                         self.touched,
+                                         inv_scale=1.0 / bscale)
             elif self.backward_mode == "blocks":
                 row_of_entry = ops.row_index(row_ptr, fids.numel())
                 gw, gblocks = ops.ffm_block_emit(row_of_entry, row_ptr,
@@ -186,6 +215,8 @@ class FFMModel:
         if self.h.optimizer == "ftrl" and "zW" in d:
             self.zW.copy_(d["zW"])
             self.zV.copy_(d["zV"])
+        if self.Vh is not None:
+            self.Vh.copy_(self.V.to(torch.bfloat16))
 
 
 class FFMTrainer:

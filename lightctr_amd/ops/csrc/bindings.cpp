@@ -191,7 +191,8 @@ at::Tensor ffm_forward(at::Tensor row_ptr, at::Tensor fields, at::Tensor fids,
   check_cuda_i32(row_ptr, "row_ptr");
   check_cuda_i32(fields, "fields");
   check_cuda_i32(fids, "fids");
-  check_cuda_f32(V, "V");
+  const bool v_bf16 = V.scalar_type() == at::kBFloat16;
+  CHK(v_bf16 || V.scalar_type() == at::kFloat, "V must be fp32 or bf16");
   const int B = (int)row_ptr.numel() - 1;
   const int nfields = (int)V.size(1);
   const int K = (int)V.size(2);
@@ -203,8 +204,8 @@ at::Tensor ffm_forward(at::Tensor row_ptr, at::Tensor fields, at::Tensor fids,
   // experiments.
   lightctr::ffm_forward_pp_launch(
       row_ptr.data_ptr<int>(), fields.data_ptr<int>(), fids.data_ptr<int>(),
-      vals.data_ptr<float>(), W.data_ptr<float>(), V.data_ptr<float>(),
-      pred.data_ptr<float>(), nfields, B, K, cur_stream());
+      vals.data_ptr<float>(), W.data_ptr<float>(), V.data_ptr(),
+      v_bf16 ? 1 : 0, pred.data_ptr<float>(), nfields, B, K, cur_stream());
   return pred;
 }
 
@@ -219,19 +220,22 @@ at::Tensor ffm_forward_pp(at::Tensor row_ptr, at::Tensor fields,
   auto pred = at::empty({B}, W.options());
   lightctr::ffm_forward_pp_launch(
       row_ptr.data_ptr<int>(), fields.data_ptr<int>(), fids.data_ptr<int>(),
-      vals.data_ptr<float>(), W.data_ptr<float>(), V.data_ptr<float>(),
-      pred.data_ptr<float>(), nfields, B, K, cur_stream());
+      vals.data_ptr<float>(), W.data_ptr<float>(), V.data_ptr(),
+      V.scalar_type() == at::kBFloat16 ? 1 : 0, pred.data_ptr<float>(),
+      nfields, B, K, cur_stream());
   return pred;
 }
 
 // per-row staged fp16 block emit -> (gw fp32, gblocks fp16 [nnz, nf*K])
 std::vector<at::Tensor> ffm_row_emit(at::Tensor row_ptr, at::Tensor fields,
                                      at::Tensor fids, at::Tensor vals,
-                                     at::Tensor V, at::Tensor dpred) {
+                                     at::Tensor V, at::Tensor dpred,
+                                     double scale) {
   check_cuda_i32(row_ptr, "row_ptr");
   check_cuda_i32(fields, "fields");
   check_cuda_i32(fids, "fids");
-  check_cuda_f32(V, "V");
+  const bool v_bf16 = V.scalar_type() == at::kBFloat16;
+  CHK(v_bf16 || V.scalar_type() == at::kFloat, "V must be fp32 or bf16");
   const int B = (int)row_ptr.numel() - 1;
   const int nfields = (int)V.size(1);
   const int K = (int)V.size(2);
@@ -239,20 +243,21 @@ std::vector<at::Tensor> ffm_row_emit(at::Tensor row_ptr, at::Tensor fields,
   CHK(lightctr::ffm_staged_eligible(nfields, K, maxn),
       "nfields*K too large for the staged row emit");
   const auto nnz = fids.numel();
-  auto gw = at::empty({nnz}, V.options());
+  auto gw = at::empty({nnz}, vals.options());
   auto gblocks = at::empty({nnz, (long)nfields * K},
-                           V.options().dtype(at::kHalf));
+                           vals.options().dtype(at::kHalf));
   lightctr::ffm_row_emit_launch(
       row_ptr.data_ptr<int>(), fields.data_ptr<int>(), fids.data_ptr<int>(),
-      vals.data_ptr<float>(), V.data_ptr<float>(), dpred.data_ptr<float>(),
-      gblocks.data_ptr(), gw.data_ptr<float>(), nfields, B, maxn, K,
-      cur_stream());
+      vals.data_ptr<float>(), V.data_ptr(), v_bf16 ? 1 : 0,
+      dpred.data_ptr<float>(), gblocks.data_ptr(), gw.data_ptr<float>(),
+      nfields, B, maxn, K, (float)scale, cur_stream());
   return {gw, gblocks};
 }
 
 void ffm_blocks_apply_f16(at::Tensor sorted_fids, at::Tensor perm,
                           at::Tensor gblocks, at::Tensor gw, at::Tensor gradW,
-                          at::Tensor gradV, at::Tensor touched) {
+                          at::Tensor gradV, at::Tensor touched,
+                          double inv_scale) {
   check_cuda_i32(sorted_fids, "sorted_fids");
   auto perm_i = perm32(perm).contiguous();
   CHK(gblocks.scalar_type() == at::kHalf, "gblocks must be fp16");
@@ -261,7 +266,7 @@ void ffm_blocks_apply_f16(at::Tensor sorted_fids, at::Tensor perm,
       sorted_fids.data_ptr<int>(), perm_i.data_ptr<int>(), gblocks.data_ptr(),
       gw.data_ptr<float>(), gradW.data_ptr<float>(), gradV.data_ptr<float>(),
       (unsigned long long*)touched.data_ptr(), D, (int)sorted_fids.numel(),
-      cur_stream());
+      (float)inv_scale, cur_stream());
 }
 
 void ffm_backward(at::Tensor row_ptr, at::Tensor fields, at::Tensor fids,
@@ -362,14 +367,17 @@ void ps_apply(at::Tensor lidx, at::Tensor gW, at::Tensor gV, at::Tensor W,
 void sparse_adagrad_apply(at::Tensor uniq, at::Tensor count, at::Tensor W,
                           at::Tensor V, at::Tensor nW, at::Tensor nV,
                           at::Tensor gradW, at::Tensor gradV, double lr,
-                          double eps, double l2) {
+                          double eps, double l2,
+                          c10::optional<at::Tensor> Vh) {
   check_cuda_i32(uniq, "uniq");
   const int D = (int)(V.numel() / V.size(0));
+  if (Vh) CHK(Vh->scalar_type() == at::kBFloat16, "Vh mirror must be bf16");
   lightctr::sparse_adagrad_apply_launch(
       uniq.data_ptr<int>(), count.data_ptr<int>(), W.data_ptr<float>(),
       V.data_ptr<float>(), nW.data_ptr<float>(), nV.data_ptr<float>(),
       gradW.data_ptr<float>(), gradV.data_ptr<float>(), (float)lr, (float)eps,
-      (float)l2, (int)uniq.numel(), D, cur_stream());
+      (float)l2, (int)uniq.numel(), D,
+      Vh ? Vh->data_ptr() : nullptr, cur_stream());
 }
 
 void sparse_ftrl_apply(at::Tensor uniq, at::Tensor count, at::Tensor W,
@@ -377,16 +385,19 @@ void sparse_ftrl_apply(at::Tensor uniq, at::Tensor count, at::Tensor W,
                        at::Tensor zV, at::Tensor nV, at::Tensor gradW,
                        at::Tensor gradV, double alpha, double beta, double l1,
                        double l2, int64_t v_adagrad, double v_lr,
-                       double v_eps, double v_l2) {
+                       double v_eps, double v_l2,
+                       c10::optional<at::Tensor> Vh) {
   check_cuda_i32(uniq, "uniq");
   const int D = (int)(V.numel() / V.size(0));
+  if (Vh) CHK(Vh->scalar_type() == at::kBFloat16, "Vh mirror must be bf16");
   lightctr::sparse_ftrl_apply_launch(
       uniq.data_ptr<int>(), count.data_ptr<int>(), W.data_ptr<float>(),
       V.data_ptr<float>(), zW.data_ptr<float>(), nW.data_ptr<float>(),
       zV.data_ptr<float>(), nV.data_ptr<float>(), gradW.data_ptr<float>(),
       gradV.data_ptr<float>(), (float)alpha, (float)beta, (float)l1,
       (float)l2, (int)uniq.numel(), D, (int)v_adagrad, (float)v_lr,
-      (float)v_eps, (float)v_l2, cur_stream());
+      (float)v_eps, (float)v_l2, Vh ? Vh->data_ptr() : nullptr,
+      cur_stream());
 }
 
 at::Tensor bitmap_compact(at::Tensor bitmap, at::Tensor out_fids,
@@ -806,9 +817,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ffm_forward", &ffm_forward, "FFM pairwise forward (LDS-staged)");
   m.def("ffm_forward_pp", &ffm_forward_pp, "lane-per-pair FFM forward");
   m.def("ffm_row_emit", &ffm_row_emit,
-        "FFM per-row staged fp16 block emit -> (gw, gblocks)");
+        "FFM per-row staged fp16 block emit -> (gw, gblocks)",
+        py::arg("row_ptr"), py::arg("fields"), py::arg("fids"),
+        py::arg("vals"), py::arg("V"), py::arg("dpred"),
+        py::arg("scale") = 1.0);
   m.def("ffm_blocks_apply_f16", &ffm_blocks_apply_f16,
-        "segment-reduce fp16 blocks into slabs (interior stores)");
+        "segment-reduce fp16 blocks into slabs (interior stores)",
+        py::arg("sorted_fids"), py::arg("perm"), py::arg("gblocks"),
+        py::arg("gw"), py::arg("gradW"), py::arg("gradV"),
+        py::arg("touched"), py::arg("inv_scale") = 1.0);
   m.def("ffm_backward", &ffm_backward, "FFM fused pairwise backward scatter");
   m.def("ffm_sorted_backward", &ffm_sorted_backward,
         "FFM sorted segment-reduce backward (LDS block accumulate)");
@@ -820,7 +837,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("parse_libffm", &parse_libffm, "native libffm text parser",
         py::arg("path"), py::arg("max_rows") = -1);
   m.def("sparse_adagrad_apply", &sparse_adagrad_apply,
-        "generic sparse fused Adagrad (runtime D)");
+        "generic sparse fused Adagrad (runtime D)",
+        py::arg("uniq"), py::arg("count"), py::arg("W"), py::arg("V"),
+        py::arg("nW"), py::arg("nV"), py::arg("gradW"), py::arg("gradV"),
+        py::arg("lr"), py::arg("eps"), py::arg("l2"),
+        py::arg("Vh") = py::none());
   m.def("ps_apply", &ps_apply,
         "PS-side fused updater (sgd/adagrad/dcasgd/dcasgda)");
   m.def("sparse_ftrl_apply", &sparse_ftrl_apply,
@@ -830,7 +851,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("gradW"), py::arg("gradV"), py::arg("alpha"),
         py::arg("beta"), py::arg("l1"), py::arg("l2"),
         py::arg("v_adagrad") = 0, py::arg("v_lr") = 0.05,
-        py::arg("v_eps") = 1e-8, py::arg("v_l2") = 1e-5);
+        py::arg("v_eps") = 1e-8, py::arg("v_l2") = 1e-5,
+        py::arg("Vh") = py::none());
   m.def("gemm_bf16", &gemm_bf16, "MFMA bf16 GEMM (C fp32), fused bias+act");
   m.def("gemm_bf16_full", &gemm_bf16_full,
         "MFMA bf16 GEMM returning (C fp32, C bf16)");

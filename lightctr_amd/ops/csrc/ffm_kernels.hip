@@ -18,6 +18,74 @@
 
 namespace lightctr {
 
+typedef __attribute__((ext_vector_type(4))) __bf16 ffm_bf16x4;
+
+// V-element abstraction: the FFM kernels run on fp32 V (master weights)
+// or on a bf16 V mirror (BASELINE config #3 "FFM bf16": fp32 master +
+// optimizer, bf16 compute reads — halves every coalesced V byte).
+template <typename VT>
+struct VQuad;
+template <>
+struct VQuad<float> {
+  // 4 consecutive elements as float4 (16 B load)
+  static __device__ __forceinline__ float4 load(const float* p) {
+    return *(const float4*)p;
+  }
+};
+template <>
+struct VQuad<__bf16> {
+  // 4 consecutive elements in 8 B, widened to fp32
+  static __device__ __forceinline__ float4 load(const __bf16* p) {
+    const ffm_bf16x4 h = *(const ffm_bf16x4*)p;
+    return make_float4((float)h.x, (float)h.y, (float)h.z, (float)h.w);
+  }
+};
+
+// Per-V-type staging codec for the row_emit kernel: fp32 V stages as fp16
+// (halved LDS), bf16 V stages raw (no conversion at all).
+template <typename VT>
+struct FfmStage;
+template <>
+struct FfmStage<float> {
+  using ST = _Float16;
+  static __device__ __forceinline__ uint2 pack(const float* src) {
+    const float4 v = *(const float4*)src;
+    union {
+      __half2 h[2];
+      uint2 u;
+    } pk;
+    pk.h[0] = __floats2half2_rn(v.x, v.y);
+    pk.h[1] = __floats2half2_rn(v.z, v.w);
+    return pk.u;
+  }
+  static __device__ __forceinline__ float4 unpack(uint2 u) {
+    union {
+      __half2 h[2];
+      uint2 u2;
+    } in;
+    in.u2 = u;
+    const float2 a = __half22float2(in.h[0]);
+    const float2 b = __half22float2(in.h[1]);
+    return make_float4(a.x, a.y, b.x, b.y);
+  }
+};
+template <>
+struct FfmStage<__bf16> {
+  using ST = __bf16;
+  static __device__ __forceinline__ uint2 pack(const __bf16* src) {
+    return *(const uint2*)src;  // raw copy, 4 elements
+  }
+  static __device__ __forceinline__ float4 unpack(uint2 u) {
+    union {
+      ffm_bf16x4 h;
+      uint2 u2;
+    } in;
+    in.u2 = u;
+    return make_float4((float)in.h.x, (float)in.h.y, (float)in.h.z,
+                       (float)in.h.w);
+  }
+};
+
 // start offset of pair-row i in the flattened (i<j) enumeration of n items
 __device__ __forceinline__ int tri_start(int i, int n) {
   return i * (n - 1) - (i * (i - 1)) / 2;
@@ -78,13 +146,13 @@ __global__ void ffm_forward_kernel(const int* __restrict__ row_ptr,
 // registers — no cross-lane shuffles, 8x fewer wave-iterations than the
 // (pair-group, k) layout, and 4 KB of independent loads in flight per
 // wave-iteration instead of 512 B. Same math, same HBM byte count.
-template <int K>
+template <int K, typename VT>
 __global__ void ffm_forward_pp_kernel(const int* __restrict__ row_ptr,
                                       const int* __restrict__ fields,
                                       const int* __restrict__ fids,
                                       const float* __restrict__ vals,
                                       const float* __restrict__ W,
-                                      const float* __restrict__ V,
+                                      const VT* __restrict__ V,
                                       float* __restrict__ pred, int nfields,
                                       int B) {
   const int lane = threadIdx.x & (LCTR_WAVE - 1);
@@ -106,23 +174,21 @@ __global__ void ffm_forward_pp_kernel(const int* __restrict__ row_ptr,
     int i0, j0, i1, j1;
     tri_decode(p, n, &i0, &j0);
     tri_decode(p + LCTR_WAVE, n, &i1, &j1);
-    const float4* va0 =
-        (const float4*)&V[((size_t)fids[beg + i0] * nfields +
-                           fields[beg + j0]) * K];
-    const float4* vb0 =
-        (const float4*)&V[((size_t)fids[beg + j0] * nfields +
-                           fields[beg + i0]) * K];
-    const float4* va1 =
-        (const float4*)&V[((size_t)fids[beg + i1] * nfields +
-                           fields[beg + j1]) * K];
-    const float4* vb1 =
-        (const float4*)&V[((size_t)fids[beg + j1] * nfields +
-                           fields[beg + i1]) * K];
+    const VT* va0 =
+        &V[((size_t)fids[beg + i0] * nfields + fields[beg + j0]) * K];
+    const VT* vb0 =
+        &V[((size_t)fids[beg + j0] * nfields + fields[beg + i0]) * K];
+    const VT* va1 =
+        &V[((size_t)fids[beg + i1] * nfields + fields[beg + j1]) * K];
+    const VT* vb1 =
+        &V[((size_t)fids[beg + j1] * nfields + fields[beg + i1]) * K];
     float t0 = 0.f, t1 = 0.f;
 #pragma unroll
     for (int q = 0; q < K / 4; ++q) {
-      const float4 a0 = va0[q], b0 = vb0[q];
-      const float4 a1 = va1[q], b1 = vb1[q];
+      const float4 a0 = VQuad<VT>::load(va0 + 4 * q);
+      const float4 b0 = VQuad<VT>::load(vb0 + 4 * q);
+      const float4 a1 = VQuad<VT>::load(va1 + 4 * q);
+      const float4 b1 = VQuad<VT>::load(vb1 + 4 * q);
       t0 += a0.x * b0.x + a0.y * b0.y + a0.z * b0.z + a0.w * b0.w;
       t1 += a1.x * b1.x + a1.y * b1.y + a1.z * b1.z + a1.w * b1.w;
     }
@@ -132,16 +198,15 @@ __global__ void ffm_forward_pp_kernel(const int* __restrict__ row_ptr,
   for (; p < npairs; p += LCTR_WAVE) {
     int i, j;
     tri_decode(p, n, &i, &j);
-    const float4* va =
-        (const float4*)&V[((size_t)fids[beg + i] * nfields +
-                           fields[beg + j]) * K];
-    const float4* vb =
-        (const float4*)&V[((size_t)fids[beg + j] * nfields +
-                           fields[beg + i]) * K];
+    const VT* va =
+        &V[((size_t)fids[beg + i] * nfields + fields[beg + j]) * K];
+    const VT* vb =
+        &V[((size_t)fids[beg + j] * nfields + fields[beg + i]) * K];
     float t = 0.f;
 #pragma unroll
     for (int q = 0; q < K / 4; ++q) {
-      const float4 a = va[q], b = vb[q];
+      const float4 a = VQuad<VT>::load(va + 4 * q);
+      const float4 b = VQuad<VT>::load(vb + 4 * q);
       t += a.x * b.x + a.y * b.y + a.z * b.z + a.w * b.w;
     }
     acc += t * vals[beg + i] * vals[beg + j];
@@ -487,15 +552,19 @@ __global__ void ffm_fwd_staged_kernel(
 //     global_store_short (128 B/wave/inst)
 // Rows with duplicate fields (or n > maxn) fall back to the round-1
 // atomic-accumulate loop, which stays bit-correct for that case.
-template <int K>
+template <int K, typename VT>
 __global__ void ffm_row_emit_kernel(
     const int* __restrict__ row_ptr, const int* __restrict__ fields,
     const int* __restrict__ fids, const float* __restrict__ vals,
-    const float* __restrict__ V, const float* __restrict__ dpred,
+    const VT* __restrict__ V, const float* __restrict__ dpred,
     _Float16* __restrict__ gblocks, float* __restrict__ gw, int nfields,
-    int B, int maxn) {
-  // LDS layout: acc[nw][nf*(K+1)] fp32 (dup fallback only) | stage_h
-  // [maxn*D] fp16 | vals_s[maxn] fp32 | fmap[nfields] i32 | dup flag
+    int B, int maxn, float scale) {
+  // LDS layout: acc[nw][nf*(K+1)] fp32 (dup fallback only) | stage
+  // [maxn*D] (fp16 for fp32 V, raw bf16 for bf16 V) | vals_s[maxn] fp32 |
+  // fmap[nfields] i32 | dup flag. `scale` (a power of two) is folded into
+  // the emitted fp16 blocks so B=65536-scaled logloss gradients (~1e-8)
+  // stay inside fp16 normal range; the apply kernel divides it back out.
+  using ST = typename FfmStage<VT>::ST;
   extern __shared__ float lds[];
   constexpr int G = LCTR_WAVE / K;
   const int row = blockIdx.x;
@@ -511,7 +580,7 @@ __global__ void ffm_row_emit_kernel(
   const int D = nfields * K;
   const int fstride = K + 1;
   float* acc = lds + (size_t)wv * nfields * fstride;
-  _Float16* stage = (_Float16*)(lds + (size_t)nw * nfields * fstride);
+  ST* stage = (ST*)(lds + (size_t)nw * nfields * fstride);
   float* vals_s = (float*)(stage + (size_t)maxn * D);
   int* fmap = (int*)(vals_s + maxn);
   int* dupflag = fmap + nfields;
@@ -519,19 +588,13 @@ __global__ void ffm_row_emit_kernel(
 
   const bool staged = n <= maxn;
   if (staged) {
-    // vectorized stage: float4 V load -> 2x half2 -> one 8 B LDS store
+    // vectorized stage: 16 B (fp32) or 8 B (bf16) V quad -> one 8 B LDS
+    // store
     const int nq = D / 4;  // K % 4 == 0 always (K in {4,8,16,32,64})
     uint2* st2 = (uint2*)stage;
     for (int idx = tid; idx < n * nq; idx += blockDim.x) {
       const int e = idx / nq, q = idx - e * nq;
-      const float4 v4 = ((const float4*)(V + (size_t)fids[beg + e] * D))[q];
-      union {
-        __half2 h[2];
-        uint2 u;
-      } pk;
-      pk.h[0] = __floats2half2_rn(v4.x, v4.y);
-      pk.h[1] = __floats2half2_rn(v4.z, v4.w);
-      st2[idx] = pk.u;
+      st2[idx] = FfmStage<VT>::pack(V + (size_t)fids[beg + e] * D + 4 * q);
     }
   }
   for (int f = tid; f < nfields; f += blockDim.x) fmap[f] = -1;
@@ -560,14 +623,13 @@ __global__ void ffm_row_emit_kernel(
           union {
             __half2 h[2];
             uint2 u2;
-          } in, ov;
-          in.u2 = *(const uint2*)&stage[(size_t)(j * nfields + Fi) * K +
-                                        ((u * 4) & (K - 1))];
-          const float s = dxi * vals_s[j];
-          const float2 a = __half22float2(in.h[0]);
-          const float2 b = __half22float2(in.h[1]);
-          ov.h[0] = __floats2half2_rn(a.x * s, a.y * s);
-          ov.h[1] = __floats2half2_rn(b.x * s, b.y * s);
+          } ov;
+          const float4 v = FfmStage<VT>::unpack(
+              *(const uint2*)&stage[(size_t)(j * nfields + Fi) * K +
+                                    ((u * 4) & (K - 1))]);
+          const float s = dxi * vals_s[j] * scale;
+          ov.h[0] = __floats2half2_rn(v.x * s, v.y * s);
+          ov.h[1] = __floats2half2_rn(v.z * s, v.w * s);
           o = ov.u2;
         }
         out2[u] = o;
@@ -585,14 +647,15 @@ __global__ void ffm_row_emit_kernel(
     const float xi = vals[p];
     for (int j = g; j < n; j += G) {
       if (j == i) continue;
-      const float v = staged
-                          ? (float)stage[(j * nfields + Fi) * K + k]
-                          : V[((size_t)fids[beg + j] * nfields + Fi) * K + k];
+      const float v =
+          staged ? (float)stage[(j * nfields + Fi) * K + k]
+                 : (float)V[((size_t)fids[beg + j] * nfields + Fi) * K + k];
       atomicAdd(&acc[fields[beg + j] * fstride + k],
                 d * xi * vals[beg + j] * v);
     }
     for (int t = lane; t < D; t += LCTR_WAVE)
-      gblocks[(size_t)p * D + t] = (_Float16)acc[(t / K) * fstride + (t % K)];
+      gblocks[(size_t)p * D + t] =
+          (_Float16)(acc[(t / K) * fstride + (t % K)] * scale);
     if (lane == 0) gw[p] = d * xi;
   }
 }
@@ -606,7 +669,8 @@ __global__ void ffm_blocks_apply_f16_kernel(
     const int* __restrict__ sorted_fids, const int* __restrict__ perm,
     const _Float16* __restrict__ gblocks, const float* __restrict__ gw,
     float* __restrict__ gradW, float* __restrict__ gradV,
-    unsigned long long* __restrict__ touched, int D, int nnz, int chunk) {
+    unsigned long long* __restrict__ touched, int D, int nnz, int chunk,
+    float inv_scale) {
   extern __shared__ float lds_acc[];
   const int lane = threadIdx.x & 63;
   const int wave_in_blk = threadIdx.x >> 6;
@@ -625,13 +689,14 @@ __global__ void ffm_blocks_apply_f16_kernel(
     const bool tail_ok = tail_e >= nnz || sorted_fids[tail_e] != cur;
     if (head_ok && tail_ok) {
       for (int i = lane; i < D; i += LCTR_WAVE) {
-        gradV[(size_t)cur * D + i] = acc[i];
+        gradV[(size_t)cur * D + i] = acc[i] * inv_scale;
         acc[i] = 0.f;
       }
       if (lane == 0) gradW[cur] = accw;
     } else {
       for (int i = lane; i < D; i += LCTR_WAVE) {
-        if (acc[i] != 0.f) atomicAdd(&gradV[(size_t)cur * D + i], acc[i]);
+        if (acc[i] != 0.f)
+          atomicAdd(&gradV[(size_t)cur * D + i], acc[i] * inv_scale);
         acc[i] = 0.f;
       }
       if (lane == 0) atomicAdd(&gradW[cur], accw);
@@ -699,25 +764,34 @@ void ffm_fwd_staged_launch(const int* row_ptr, const int* fields,
 }
 
 void ffm_row_emit_launch(const int* row_ptr, const int* fields,
-                         const int* fids, const float* vals, const float* V,
-                         const float* dpred, void* gblocks, float* gw,
-                         int nfields, int B, int maxn, int K,
-                         hipStream_t stream) {
+                         const int* fids, const float* vals, const void* V,
+                         int v_bf16, const float* dpred, void* gblocks,
+                         float* gw, int nfields, int B, int maxn, int K,
+                         float scale, hipStream_t stream) {
   if (B <= 0) return;
   dim3 block(256);
   dim3 grid(B);
   const size_t lds = ffm_row_emit_lds_bytes(nfields, K, maxn);
-  DISPATCH_FFM_K(K, hipLaunchKernelGGL((ffm_row_emit_kernel<KC>), grid, block,
-                                       lds, stream, row_ptr, fields, fids,
-                                       vals, V, dpred, (_Float16*)gblocks, gw,
-                                       nfields, B, maxn));
+  if (v_bf16) {
+    DISPATCH_FFM_K(
+        K, hipLaunchKernelGGL((ffm_row_emit_kernel<KC, __bf16>), grid, block,
+                              lds, stream, row_ptr, fields, fids, vals,
+                              (const __bf16*)V, dpred, (_Float16*)gblocks, gw,
+                              nfields, B, maxn, scale));
+  } else {
+    DISPATCH_FFM_K(
+        K, hipLaunchKernelGGL((ffm_row_emit_kernel<KC, float>), grid, block,
+                              lds, stream, row_ptr, fields, fids, vals,
+                              (const float*)V, dpred, (_Float16*)gblocks, gw,
+                              nfields, B, maxn, scale));
+  }
 }
 
 void ffm_blocks_apply_f16_launch(const int* sorted_fids, const int* perm,
                                  const void* gblocks, const float* gw,
                                  float* gradW, float* gradV,
                                  unsigned long long* touched, int D, int nnz,
-                                 hipStream_t stream) {
+                                 float inv_scale, hipStream_t stream) {
   if (nnz <= 0) return;
   const int chunk = 96;
   const int wpb = 4;
@@ -727,20 +801,29 @@ void ffm_blocks_apply_f16_launch(const int* sorted_fids, const int* perm,
   const size_t lds = (size_t)wpb * D * sizeof(float);
   hipLaunchKernelGGL(ffm_blocks_apply_f16_kernel, grid, block, lds, stream,
                      sorted_fids, perm, (const _Float16*)gblocks, gw, gradW,
-                     gradV, touched, D, nnz, chunk);
+                     gradV, touched, D, nnz, chunk, inv_scale);
 }
 
 void ffm_forward_pp_launch(const int* row_ptr, const int* fields,
                            const int* fids, const float* vals,
-                           const float* W, const float* V, float* pred,
-                           int nfields, int B, int K, hipStream_t stream) {
+                           const float* W, const void* V, int v_bf16,
+                           float* pred, int nfields, int B, int K,
+                           hipStream_t stream) {
   if (B <= 0) return;
   const int wpb = 4;
   dim3 block(wpb * LCTR_WAVE);
   dim3 grid((B + wpb - 1) / wpb);
-  DISPATCH_FFM_K(K, hipLaunchKernelGGL((ffm_forward_pp_kernel<KC>), grid,
-                                       block, 0, stream, row_ptr, fields,
-                                       fids, vals, W, V, pred, nfields, B));
+  if (v_bf16) {
+    DISPATCH_FFM_K(K, hipLaunchKernelGGL((ffm_forward_pp_kernel<KC, __bf16>),
+                                         grid, block, 0, stream, row_ptr,
+                                         fields, fids, vals, W,
+                                         (const __bf16*)V, pred, nfields, B));
+  } else {
+    DISPATCH_FFM_K(K, hipLaunchKernelGGL((ffm_forward_pp_kernel<KC, float>),
+                                         grid, block, 0, stream, row_ptr,
+                                         fields, fids, vals, W,
+                                         (const float*)V, pred, nfields, B));
+  }
 }
 
 void ffm_block_emit_launch(const int* row_of_entry, const int* row_ptr,

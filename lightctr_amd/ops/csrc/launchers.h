@@ -63,23 +63,24 @@ void ffm_block_emit_launch(const int* row_of_entry, const int* row_ptr,
                            int nfields, int nnz, int K, ihipStream_t* stream);
 void ffm_forward_pp_launch(const int* row_ptr, const int* fields,
                            const int* fids, const float* vals,
-                           const float* W, const float* V, float* pred,
-                           int nfields, int B, int K, ihipStream_t* stream);
+                           const float* W, const void* V, int v_bf16,
+                           float* pred, int nfields, int B, int K,
+                           ihipStream_t* stream);
 bool ffm_staged_eligible(int nfields, int K, int maxn);
 void ffm_fwd_staged_launch(const int* row_ptr, const int* fields,
                            const int* fids, const float* vals, const float* W,
                            const float* V, float* pred, int nfields, int B,
                            int maxn, int K, ihipStream_t* stream);
 void ffm_row_emit_launch(const int* row_ptr, const int* fields,
-                         const int* fids, const float* vals, const float* V,
-                         const float* dpred, void* gblocks, float* gw,
-                         int nfields, int B, int maxn, int K,
-                         ihipStream_t* stream);
+                         const int* fids, const float* vals, const void* V,
+                         int v_bf16, const float* dpred, void* gblocks,
+                         float* gw, int nfields, int B, int maxn, int K,
+                         float scale, ihipStream_t* stream);
 void ffm_blocks_apply_f16_launch(const int* sorted_fids, const int* perm,
                                  const void* gblocks, const float* gw,
                                  float* gradW, float* gradV,
                                  unsigned long long* touched, int D, int nnz,
-                                 ihipStream_t* stream);
+                                 float inv_scale, ihipStream_t* stream);
 void ffm_blocks_apply_launch(const int* sorted_fids, const int* perm,
                              const float* gblocks, const float* gw,
                              float* gradW, float* gradV,
@@ -96,13 +97,15 @@ void ps_apply_launch(const long* lidx, int n, const float* gW,
 void sparse_adagrad_apply_launch(const int* uniq, const int* count, float* W,
                                  float* V, float* nW, float* nV, float* gradW,
                                  float* gradV, float lr, float eps, float l2,
-                                 int capacity, int D, ihipStream_t* stream);
+                                 int capacity, int D, void* Vh,
+                                 ihipStream_t* stream);
 void sparse_ftrl_apply_launch(const int* uniq, const int* count, float* W,
                               float* V, float* zW, float* nW, float* zV,
                               float* nV, float* gradW, float* gradV,
                               float alpha, float beta, float l1, float l2,
                               int capacity, int D, int v_adagrad, float v_lr,
-                              float v_eps, float v_l2, ihipStream_t* stream);
+                              float v_eps, float v_l2, void* Vh,
+                              ihipStream_t* stream);
 void fm_adagrad_apply_launch(const int* uniq, const int* count, float* W,
                              float* V, float* nW, float* nV, float* gradW,
                              float* gradV, float lr, float eps, float l2,

@@ -7,6 +7,15 @@
 
 namespace lightctr {
 
+typedef __attribute__((ext_vector_type(4))) __bf16 misc_bf16x4;
+
+// optional bf16 mirror write-back (FFM bf16 mode keeps V fp32 master +
+// a bf16 compute mirror refreshed for exactly the touched features)
+__device__ __forceinline__ void mirror4(__bf16* Vh, size_t off, float4 v) {
+  misc_bf16x4 h = {(__bf16)v.x, (__bf16)v.y, (__bf16)v.z, (__bf16)v.w};
+  *(misc_bf16x4*)&Vh[off] = h;
+}
+
 // declared in fm_kernels.hip
 __device__ __forceinline__ void ftrl_update_g(float* w, float* z, float* n,
                                               float g, float alpha, float beta,
@@ -32,7 +41,7 @@ __global__ void sparse_adagrad_apply_g_kernel(
     float* __restrict__ W, float* __restrict__ V, float* __restrict__ nW,
     float* __restrict__ nV, float* __restrict__ gradW,
     float* __restrict__ gradV, float lr, float eps, float l2, int capacity,
-    int D) {
+    int D, __bf16* __restrict__ Vh) {
   const int i = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   const int lane = threadIdx.x & (LCTR_WAVE - 1);
   if (i >= capacity || i >= *count) return;
@@ -60,6 +69,7 @@ __global__ void sparse_adagrad_apply_g_kernel(
       nV4[d4] = a;
       V4[d4] = v;
       gV4[d4] = make_float4(0.f, 0.f, 0.f, 0.f);
+      if (Vh) mirror4(Vh, base + 4 * (size_t)d4, v);
     }
   } else {
     for (int d = lane; d < D; d += LCTR_WAVE) {
@@ -68,6 +78,7 @@ __global__ void sparse_adagrad_apply_g_kernel(
       nV[base + d] = acc;
       V[base + d] -= lr * g * __frsqrt_rn(acc + eps);
       gradV[base + d] = 0.f;
+      if (Vh) Vh[base + d] = (__bf16)V[base + d];
     }
   }
   if (lane == 0 && W != nullptr) {
@@ -85,7 +96,7 @@ __global__ void sparse_ftrl_apply_g_kernel(
     float* __restrict__ nW, float* __restrict__ zV, float* __restrict__ nV,
     float* __restrict__ gradW, float* __restrict__ gradV, float alpha,
     float beta, float l1, float l2, int capacity, int D, int v_adagrad,
-    float v_lr, float v_eps, float v_l2) {
+    float v_lr, float v_eps, float v_l2, __bf16* __restrict__ Vh) {
   const int i = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   const int lane = threadIdx.x & (LCTR_WAVE - 1);
   if (i >= capacity || i >= *count) return;
@@ -112,6 +123,7 @@ __global__ void sparse_ftrl_apply_g_kernel(
       nV4[d4] = a;
       V4[d4] = v;
       gV4[d4] = make_float4(0.f, 0.f, 0.f, 0.f);
+      if (Vh) mirror4(Vh, base + 4 * (size_t)d4, v);
     }
   } else {
     for (int d = lane; d < D; d += LCTR_WAVE) {
@@ -125,6 +137,7 @@ __global__ void sparse_ftrl_apply_g_kernel(
                       gradV[base + d], alpha, beta, l1, l2);
       }
       gradV[base + d] = 0.f;
+      if (Vh) Vh[base + d] = (__bf16)V[base + d];
     }
   }
   if (lane == 0 && W != nullptr) {
@@ -201,12 +214,13 @@ void ps_apply_launch(const long* lidx, int n, const float* gW,
 void sparse_adagrad_apply_launch(const int* uniq, const int* count, float* W,
                                  float* V, float* nW, float* nV, float* gradW,
                                  float* gradV, float lr, float eps, float l2,
-                                 int capacity, int D, hipStream_t stream) {
+                                 int capacity, int D, void* Vh,
+                                 hipStream_t stream) {
   dim3 block(256);
   dim3 grid((capacity + 3) / 4);
   hipLaunchKernelGGL(sparse_adagrad_apply_g_kernel, grid, block, 0, stream,
                      uniq, count, W, V, nW, nV, gradW, gradV, lr, eps, l2,
-                     capacity, D);
+                     capacity, D, (__bf16*)Vh);
 }
 
 void sparse_ftrl_apply_launch(const int* uniq, const int* count, float* W,
@@ -214,12 +228,14 @@ void sparse_ftrl_apply_launch(const int* uniq, const int* count, float* W,
                               float* nV, float* gradW, float* gradV,
                               float alpha, float beta, float l1, float l2,
                               int capacity, int D, int v_adagrad, float v_lr,
-                              float v_eps, float v_l2, hipStream_t stream) {
+                              float v_eps, float v_l2, void* Vh,
+                              hipStream_t stream) {
   dim3 block(256);
   dim3 grid((capacity + 3) / 4);
   hipLaunchKernelGGL(sparse_ftrl_apply_g_kernel, grid, block, 0, stream, uniq,
                      count, W, V, zW, nW, zV, nV, gradW, gradV, alpha, beta,
-                     l1, l2, capacity, D, v_adagrad, v_lr, v_eps, v_l2);
+                     l1, l2, capacity, D, v_adagrad, v_lr, v_eps, v_l2,
+                     (__bf16*)Vh);
 }
 
 }  // namespace lightctr

@@ -304,6 +304,107 @@ def test_ffm_rowemit_fastpath_parity():
 
 
 @pytest.mark.gpu
+def test_ffm_bf16_forward_and_backward_parity():
+    """bf16 compute mirror (BASELINE config #3): forward and rowemit
+    backward on a bf16 V match the fp32 reference within bf16 rounding."""
+    from lightctr_amd.ops import hip_ops
+
+    g = torch.Generator().manual_seed(5)
+    B, nf, K, F = 96, 11, 8, 3000
+    fields = torch.arange(nf, dtype=torch.int32).repeat(B).cuda()
+    fids = torch.randint(0, F, (B * nf,), generator=g,
+                         dtype=torch.int32).cuda()
+    vals = (torch.rand(B * nf, generator=g) + 0.5).cuda()
+    row_ptr = (torch.arange(B + 1, dtype=torch.int32) * nf).cuda()
+    labels = (torch.rand(B, generator=g) > 0.5).float().cuda()
+    W = torch.randn(F, generator=g).cuda()
+    V = (torch.randn(F, nf, K, generator=g) * 0.1).cuda()
+    Vh = V.to(torch.bfloat16)
+    pred = hip_ops.ffm_forward(row_ptr, fields, fids, vals, W, Vh)
+    ref = ffm_ref.ffm_forward_ref(row_ptr.cpu(), fields.cpu(), fids.cpu(),
+                                  vals.cpu(), W.cpu(),
+                                  Vh.float().cpu())
+    assert torch.allclose(pred.cpu(), ref, atol=5e-3, rtol=1e-2), \
+        (pred.cpu() - ref).abs().max()
+
+    _, dpred = hip_ops.logloss_grad(pred, labels, 1.0 / B)
+    gw, gblocks = hip_ops.ffm_row_emit(row_ptr, fields, fids, vals, Vh,
+                                       dpred, scale=128.0)
+    sorted_fids, perm = torch.sort(fids)
+    gradW = torch.zeros(F).cuda()
+    gradV = torch.zeros(F, nf, K).cuda()
+    touched = torch.zeros((F + 63) // 64, dtype=torch.int64).cuda()
+    hip_ops.ffm_blocks_apply_f16(sorted_fids, perm, gblocks, gw, gradW,
+                                 gradV.view(F, -1), touched,
+                                 inv_scale=1.0 / 128.0)
+    gW_ref, gV_ref = ffm_ref.ffm_backward_ref(row_ptr, fields, fids, vals,
+                                              Vh.float(), dpred)
+    assert torch.allclose(gradW, gW_ref, atol=1e-5, rtol=1e-4)
+    ref_scale = gV_ref.abs().max()
+    assert torch.allclose(gradV, gV_ref, atol=float(ref_scale) * 3e-2,
+                          rtol=3e-2), (gradV - gV_ref).abs().max()
+
+
+@pytest.mark.gpu
+def test_ffm_rowemit_tiny_gradient_scale():
+    """At production batch sizes the 1/B-scaled logloss gradients (~1e-8)
+    sit below fp16's subnormal floor; the power-of-two block scale must
+    preserve them. Emits with dpred ~1e-8 and checks the reduced grads
+    against the fp32 reference — without the scale these flush to 0."""
+    from lightctr_amd.ops import hip_ops
+
+    g = torch.Generator().manual_seed(13)
+    B, nf, K, F = 64, 13, 8, 2000
+    fields = torch.arange(nf, dtype=torch.int32).repeat(B).cuda()
+    fids = torch.randint(0, F, (B * nf,), generator=g,
+                         dtype=torch.int32).cuda()
+    vals = (torch.rand(B * nf, generator=g) + 0.5).cuda()
+    row_ptr = (torch.arange(B + 1, dtype=torch.int32) * nf).cuda()
+    W = torch.randn(F, generator=g).cuda()
+    V = (torch.randn(F, nf, K, generator=g) * 0.1).cuda()
+    # dpred at the magnitude a B=65536 logloss step produces
+    dpred = ((torch.rand(B, generator=g).cuda() - 0.5) * 2e-8)
+    scale = float(1 << 16)
+    gw, gblocks = hip_ops.ffm_row_emit(row_ptr, fields, fids, vals, V,
+                                       dpred, scale=scale)
+    sorted_fids, perm = torch.sort(fids)
+    gradW = torch.zeros(F).cuda()
+    gradV = torch.zeros(F, nf, K).cuda()
+    touched = torch.zeros((F + 63) // 64, dtype=torch.int64).cuda()
+    hip_ops.ffm_blocks_apply_f16(sorted_fids, perm, gblocks, gw, gradW,
+                                 gradV.view(F, -1), touched,
+                                 inv_scale=1.0 / scale)
+    gW_ref, gV_ref = ffm_ref.ffm_backward_ref(row_ptr, fields, fids, vals,
+                                              V, dpred)
+    assert float(gV_ref.abs().max()) > 0
+    # the unscaled emit would produce gradV == 0 here
+    assert float(gradV.abs().max()) > 0
+    ref_scale = gV_ref.abs().max()
+    assert torch.allclose(gradV, gV_ref, atol=float(ref_scale) * 2e-2,
+                          rtol=2e-2), (gradV - gV_ref).abs().max()
+
+
+@pytest.mark.gpu
+def test_ffm_bf16_model_trains():
+    """FFMModel(dtype=bf16) end-to-end: loss decreases and the bf16 mirror
+    tracks the fp32 master after optimizer steps."""
+    from lightctr_amd.models.ffm import FFMHyper, FFMModel
+
+    gen = SyntheticCriteo(num_features=1 << 14, seed=4, device="cuda:0")
+    m = FFMModel(FFMHyper(num_features=1 << 14, num_fields=39, k=8,
+                          dtype="bf16"), device="cuda:0")
+    losses = []
+    for i in range(30):
+        row_ptr, fields, fids, vals, labels = gen.batch(512)
+        loss = m.train_step(row_ptr, fields, fids, vals, labels)
+        losses.append(float(loss.mean()))
+    assert losses[-1] < losses[0] - 0.02, losses[::10]
+    # mirror coherence on touched rows
+    diff = (m.Vh.float() - m.V).abs().max()
+    assert float(diff) < 0.01, float(diff)
+
+
+@pytest.mark.gpu
 def test_ffm_forward_staged_long_rows():
     """Forward parity on rows spanning the staged/fallback boundary."""
     from lightctr_amd.ops import hip_ops
