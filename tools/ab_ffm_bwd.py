@@ -33,24 +33,26 @@ def main():
     ap.add_argument("--reps", type=int, default=20)
     ap.add_argument("--batch", type=int, default=65536)
     ap.add_argument("--modes", default="sorted,rowemit,blocks")
+    ap.add_argument("--dtype", default="fp32", choices=["fp32", "bf16"])
     args = ap.parse_args()
 
     ops = require_hip_ops()
     F, nf, K = 1 << 24, 39, 8
-    m = FFMModel(FFMHyper(num_features=F, num_fields=nf, k=K), device="cuda")
+    m = FFMModel(FFMHyper(num_features=F, num_fields=nf, k=K,
+                          dtype=args.dtype), device="cuda")
     gen = SyntheticCriteo(num_features=F, seed=5, device="cuda")
     row_ptr, fields, fids, vals, labels = gen.batch(args.batch)
     nnz = fids.numel()
     B = args.batch
 
-    pred = ops.ffm_forward(row_ptr, fields, fids, vals, m.W, m.V)
+    pred = ops.ffm_forward(row_ptr, fields, fids, vals, m.W, m._Vc)
     _, dpred = ops.logloss_grad(pred, labels, 1.0 / B)
     sorted_fids, perm = sort_ids(fids, F)
     row_of_entry = ops.row_index(row_ptr, nnz)
 
     # isolated phase timings
     t = time_fn(lambda: ops.ffm_forward(row_ptr, fields, fids, vals, m.W,
-                                        m.V), args.reps)
+                                        m._Vc), args.reps)
     print(f"forward                : {t * 1e3:7.3f} ms")
     t = time_fn(lambda: sort_ids(fids, F), args.reps)
     print(f"sort_ids               : {t * 1e3:7.3f} ms")
@@ -65,15 +67,16 @@ def main():
             print(f"bwd[sorted]            : {t * 1e3:7.3f} ms")
         elif mode == "rowemit":
             def emit():
-                return ops.ffm_row_emit(row_ptr, fields, fids, vals, m.V,
-                                        dpred)
+                return ops.ffm_row_emit(row_ptr, fields, fids, vals, m._Vc,
+                                        dpred, scale=65536.0)
             te = time_fn(emit, args.reps)
             gw, gblocks = emit()
             gv = m.gradV.view(F, -1)
 
             def apply():
                 ops.ffm_blocks_apply_f16(sorted_fids, perm, gblocks, gw,
-                                         m.gradW, gv, m.touched)
+                                         m.gradW, gv, m.touched,
+                                         inv_scale=1.0 / 65536.0)
             ta = time_fn(apply, args.reps)
             print(f"bwd[rowemit] emit      : {te * 1e3:7.3f} ms")
             print(f"bwd[rowemit] apply     : {ta * 1e3:7.3f} ms")
@@ -96,8 +99,8 @@ def main():
 
     # full train_step per mode
     for mode in args.modes.split(","):
-        m2 = FFMModel(FFMHyper(num_features=F, num_fields=nf, k=K),
-                      device="cuda")
+        m2 = FFMModel(FFMHyper(num_features=F, num_fields=nf, k=K,
+                              dtype=args.dtype), device="cuda")
         m2.backward_mode = mode
         t = time_fn(lambda: m2.train_step(row_ptr, fields, fids, vals,
                                           labels), args.reps)
