@@ -219,8 +219,16 @@ def main():
                   "falling back to eager", flush=True)
             step = eager_step
 
+    # §5.3 failure detection: a stalled step loop (GPU hang / dead peer
+    # blocking a collective) is detected and converted into a clean
+    # non-zero exit instead of a silent hang
+    from lightctr_amd.utils.watchdog import TrainGuard
+
+    guard = TrainGuard(model, ckpt_path=None, soft_s=60.0, dead_s=300.0,
+                       log=lambda m: print(m, flush=True)).start()
     for i in range(args.warmup):
         step(i)
+        guard.step()
     if dist:
         dist.barrier()
     if have_gpu:
@@ -228,11 +236,13 @@ def main():
     t0 = time.perf_counter()
     for i in range(args.steps):
         step(i)
+        guard.step()
     if dist:
         dist.barrier()
     if have_gpu:
         torch.cuda.synchronize()
     elapsed = time.perf_counter() - t0
+    guard.stop()
     # MAX over ranks
     if dist:
         t = torch.tensor([elapsed], device=device if have_gpu else "cpu")

@@ -85,7 +85,30 @@ def cmd_train(args) -> int:
                        batch_size=args.batch, epochs=args.epochs)
     else:
         return _train_other(args, dev)
-    tr.train(log=lambda msg: print(msg, flush=True))
+    guard = None
+    if args.watchdog:
+        from .utils.watchdog import TrainGuard
+
+        guard = TrainGuard(tr.model, ckpt_path=args.ckpt or None,
+                           soft_s=args.watchdog_soft_s,
+                           dead_s=args.watchdog_dead_s,
+                           ckpt_every=args.ckpt_every,
+                           log=lambda m: print(m, flush=True)).start()
+        if args.resume:
+            guard.maybe_resume()
+        # one heartbeat per train_step, whichever trainer loop runs it
+        inner = tr.model.train_step
+
+        def guarded_step(*a, **kw):
+            return guard.run_step(inner, *a, **kw)
+
+        tr.model.train_step = guarded_step
+    try:
+        tr.train(log=lambda msg: print(msg, flush=True))
+    finally:
+        if guard is not None:
+            tr.model.train_step = inner
+            guard.stop()
     metrics = tr.evaluate()
     print(json.dumps({"final": metrics}))
     if args.save:
@@ -354,6 +377,20 @@ def build_parser():
         p.add_argument("--save", default=None)
         p.add_argument("--load", default=None)
         p.add_argument("--dump", default=None, help="score dump path")
+        # §5.3 failure detection: heartbeat watchdog + checkpoint-restart
+        p.add_argument("--watchdog", action="store_true", default=True,
+                       help="stall watchdog + checkpoint-abort (default on)")
+        p.add_argument("--no-watchdog", dest="watchdog",
+                       action="store_false")
+        p.add_argument("--watchdog-soft-s", type=float, default=30.0)
+        p.add_argument("--watchdog-dead-s", type=float, default=120.0)
+        p.add_argument("--ckpt", default=None,
+                       help="checkpoint path for the watchdog + periodic "
+                            "saves")
+        p.add_argument("--ckpt-every", type=int, default=0,
+                       help="checkpoint every N steps (0 = only on death)")
+        p.add_argument("--resume", action="store_true",
+                       help="resume from --ckpt if it exists")
     return ap
 
 

@@ -98,7 +98,13 @@ __device__ __forceinline__ float p8_act(float v, int act) {
   return v;
 }
 
-template <bool TAIL>
+// LITE: same staging pipeline and counted boundary vmcnt, but only TWO
+// barriers per K-tile (after phase 1 — publishing every wave's b-frag
+// consumption before B(t+2) overwrites the slot — and at the tile
+// boundary) instead of the template's eight; the compiler pipelines the
+// phases freely as in the round-1 kernel. A/B lever for the big square
+// shapes where the 8-barrier variant measured 9% behind round 1.
+template <bool TAIL, bool LITE>
 __global__ __launch_bounds__(512, 1) void gemm256p8_bf16_kernel(
     const __bf16* __restrict__ A, const __bf16* __restrict__ Bst,
     const float* __restrict__ bias, float* __restrict__ C,
@@ -181,10 +187,12 @@ __global__ __launch_bounds__(512, 1) void gemm256p8_bf16_kernel(
         a[fm][kc] = *(const p8bf16x8*)&curA[ra * P8_BK + p8swz(ra, ks)];
       }
     if (sA) P8_STAGE_A(t + 1, 0);
-    asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");
-    __builtin_amdgcn_s_barrier();
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    __builtin_amdgcn_sched_barrier(0);
+    if constexpr (!LITE) {
+      asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_sched_barrier(0);
+    }
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int fm = 0; fm < 2; ++fm)
@@ -195,7 +203,7 @@ __global__ __launch_bounds__(512, 1) void gemm256p8_bf16_kernel(
           acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a[fm][kc], b[fn][kc], acc[fm][fn], 0, 0, 0);
     __builtin_amdgcn_s_setprio(0);
-    __builtin_amdgcn_s_barrier();
+    if constexpr (!LITE) __builtin_amdgcn_s_barrier();
 
     // ---- phases 1..3: A quadrant p ----
 #pragma unroll
@@ -215,9 +223,11 @@ __global__ __launch_bounds__(512, 1) void gemm256p8_bf16_kernel(
       } else {
         if (sB) P8_STAGE_B(t + 2, 1);
       }
-      __builtin_amdgcn_s_barrier();
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      __builtin_amdgcn_sched_barrier(0);
+      if constexpr (!LITE) {
+        __builtin_amdgcn_s_barrier();
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_sched_barrier(0);
+      }
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int fm = 0; fm < 2; ++fm)
@@ -228,8 +238,16 @@ __global__ __launch_bounds__(512, 1) void gemm256p8_bf16_kernel(
             acc[p * 2 + fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 a[fm][kc], b[fn][kc], acc[p * 2 + fm][fn], 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
-      if (p < 3) {
+      if (p == 1 && LITE) {
+        // LITE mid-barrier: every wave is past its phase-0 MFMA issue
+        // (b-frag registers populated), so phase 2 may overwrite the
+        // B(t) parity slot with B(t+2)
+        asm volatile("" ::: "memory");
         __builtin_amdgcn_s_barrier();
+        asm volatile("" ::: "memory");
+      }
+      if (p < 3) {
+        if constexpr (!LITE) __builtin_amdgcn_s_barrier();
       } else {
         // tile boundary: counted drain (leave B(t+2) in flight), then the
         // barrier that publishes every wave's landed stages
@@ -279,14 +297,29 @@ void gemm256p8_bf16_launch(const void* A, const void* Bst, const float* bias,
                            hipStream_t stream) {
   dim3 block(512);
   dim3 grid(N / 256, M / 256);
-  if (K % P8_BK == 0)
-    hipLaunchKernelGGL((gemm256p8_bf16_kernel<false>), grid, block, 0, stream,
-                       (const __bf16*)A, (const __bf16*)Bst, bias, C,
-                       (__bf16*)Cbf, M, N, K, act);
-  else
-    hipLaunchKernelGGL((gemm256p8_bf16_kernel<true>), grid, block, 0, stream,
-                       (const __bf16*)A, (const __bf16*)Bst, bias, C,
-                       (__bf16*)Cbf, M, N, K, act);
+  static const bool lite = [] {
+    const char* e = getenv("LCTR_GEMM_P8_LITE");
+    return e && e[0] == '1';
+  }();
+  if (K % P8_BK == 0) {
+    if (lite)
+      hipLaunchKernelGGL((gemm256p8_bf16_kernel<false, true>), grid, block,
+                         0, stream, (const __bf16*)A, (const __bf16*)Bst,
+                         bias, C, (__bf16*)Cbf, M, N, K, act);
+    else
+      hipLaunchKernelGGL((gemm256p8_bf16_kernel<false, false>), grid, block,
+                         0, stream, (const __bf16*)A, (const __bf16*)Bst,
+                         bias, C, (__bf16*)Cbf, M, N, K, act);
+  } else {
+    if (lite)
+      hipLaunchKernelGGL((gemm256p8_bf16_kernel<true, true>), grid, block, 0,
+                         stream, (const __bf16*)A, (const __bf16*)Bst, bias,
+                         C, (__bf16*)Cbf, M, N, K, act);
+    else
+      hipLaunchKernelGGL((gemm256p8_bf16_kernel<true, false>), grid, block,
+                         0, stream, (const __bf16*)A, (const __bf16*)Bst,
+                         bias, C, (__bf16*)Cbf, M, N, K, act);
+  }
 }
 
 }  // namespace lightctr

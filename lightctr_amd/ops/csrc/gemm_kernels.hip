@@ -225,13 +225,20 @@ void gemm_bf16_launch(const void* A, const void* Bst, const float* bias,
   // barrier-spanning loads buy at this geometry. v1 stays preferred; v2
   // kept (correct, race-screened) as the documented experiment.
   if (gemm256p8_eligible(M, N, K, transA, transB)) {
-    // 8-phase counted-vmcnt schedule (round 2). LCTR_GEMM_P8=0 falls back
-    // to the round-1 per-tile-drain kernel for A/B.
-    static const bool p8 = [] {
+    // 8-phase counted-vmcnt schedule (round 2). Measured (gemm_p8/old.txt):
+    // 2x the round-1 kernel on K-tail shapes (65536x256x624: 555 vs 299
+    // TF — the masked-glds tail staging + counted waits), but ~8% behind
+    // it on big square K%64==0 shapes (904 vs 996 @4k^3) where the extra
+    // per-phase barriers outweigh the counted drain. Auto: p8 for tail
+    // shapes, round-1 kernel otherwise. LCTR_GEMM_P8=1/0 forces/disables.
+    static const int p8mode = [] {
       const char* e = getenv("LCTR_GEMM_P8");
-      return !(e && e[0] == '0');
+      if (!e) return 2;  // auto
+      return e[0] == '0' ? 0 : 1;
     }();
-    if (p8) {
+    const bool use_p8 =
+        p8mode == 1 || (p8mode == 2 && (K % 64 != 0));
+    if (use_p8) {
       gemm256p8_bf16_launch(A, Bst, bias, C, Cbf, M, N, K, act, stream);
       return;
     }

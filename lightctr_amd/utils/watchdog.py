@@ -64,6 +64,96 @@ class Watchdog:
             return list(self.state)
 
 
+class TrainGuard:
+    """Failure-detection wiring for a training loop (reference
+    master.h:202-262 heartbeat -> declare-dead semantics, modernized to
+    checkpoint-restart per SURVEY §5.3).
+
+    The training thread calls `step()` once per batch; a Watchdog monitor
+    thread declares the loop dead when no step lands for `dead_s`
+    seconds (GPU hang, dead peer rank blocking a collective, ...). On
+    death the guard saves a checkpoint of the model's last COMPLETED
+    step (train_step mutates state only after its collectives, so the
+    snapshot is step-consistent per rank) and exits the process with
+    status 3 so a supervisor can restart; restart resumes from the
+    checkpoint via `maybe_resume()`.
+    """
+
+    EXIT_CODE = 3
+
+    def __init__(self, model, ckpt_path: str | None, soft_s: float = 10.0,
+                 dead_s: float = 20.0, period_s: float = 1.0,
+                 ckpt_every: int = 0, log=print):
+        self.model = model
+        self.ckpt_path = ckpt_path
+        self.ckpt_every = ckpt_every
+        self.log = log
+        self.steps = 0
+        self._wd = Watchdog(1, soft_s=soft_s, dead_s=dead_s,
+                            period_s=period_s,
+                            on_soft=self._on_soft, on_dead=self._on_dead)
+
+    def _on_soft(self, _rank):
+        self.log(f"[watchdog] training loop stalled >"
+                 f"{self._wd.soft_s:.0f}s at step {self.steps}")
+
+    def _on_dead(self, _rank):
+        self._die(f"training loop DEAD (> {self._wd.dead_s:.0f}s without "
+                  f"a step)")
+
+    def _die(self, why: str):
+        import os
+
+        self.log(f"[watchdog] {why} at step {self.steps}; checkpointing "
+                 f"and aborting for restart-resume")
+        if self.ckpt_path is not None:
+            try:
+                self.model.save(self.ckpt_path)
+                self.log(f"[watchdog] checkpoint -> {self.ckpt_path}")
+            except Exception as e:  # noqa: BLE001
+                self.log(f"[watchdog] checkpoint failed: {e}")
+        os._exit(self.EXIT_CODE)
+
+    def run_step(self, fn, *a, **kw):
+        """Run one training step under the guard: a raised communication
+        / device error (gloo raises on a dead peer; RCCL may raise or
+        hang — the hang case is the watchdog's) checkpoints the last
+        completed step and exits for restart."""
+        try:
+            out = fn(*a, **kw)
+        except Exception as e:  # noqa: BLE001 — any step fault is fatal
+            self._die(f"step failed ({type(e).__name__}: {e})")
+        self.step()
+        return out
+
+    def maybe_resume(self) -> bool:
+        # sharded models save under "{prefix}.shardRofW.pt", single models
+        # under the literal path — let load() resolve and treat a missing
+        # file as "fresh start"
+        if not self.ckpt_path:
+            return False
+        try:
+            self.model.load(self.ckpt_path)
+        except FileNotFoundError:
+            return False
+        self.log(f"[watchdog] resumed from {self.ckpt_path}")
+        return True
+
+    def start(self):
+        self._wd.start()
+        return self
+
+    def step(self):
+        self.steps += 1
+        self._wd.heartbeat(0)
+        if (self.ckpt_path and self.ckpt_every
+                and self.steps % self.ckpt_every == 0):
+            self.model.save(self.ckpt_path)
+
+    def stop(self):
+        self._wd.stop()
+
+
 def gpu_health() -> dict:
     """Host-side device health via rocm-smi (temperature/power/usage);
     returns {} when no GPU or rocm-smi unavailable."""
