@@ -400,6 +400,31 @@ void sparse_ftrl_apply(at::Tensor uniq, at::Tensor count, at::Tensor W,
       cur_stream());
 }
 
+void auc_hist_add(at::Tensor pred, at::Tensor label, at::Tensor hist) {
+  check_cuda_f32(pred, "pred");
+  check_cuda_f32(label, "label");
+  CHK(hist.is_cuda() && hist.scalar_type() == at::kInt &&
+          hist.is_contiguous(),
+      "hist must be cuda int32 [2*buckets]");
+  const int buckets = (int)(hist.numel() / 2);
+  lightctr::auc_hist_add_launch(pred.data_ptr<float>(),
+                                label.data_ptr<float>(),
+                                (long)pred.numel(),
+                                (unsigned int*)hist.data_ptr(), buckets,
+                                cur_stream());
+}
+
+at::Tensor auc_scan(at::Tensor hist) {
+  CHK(hist.is_cuda() && hist.scalar_type() == at::kInt &&
+          hist.is_contiguous(),
+      "hist must be cuda int32 [2*buckets]");
+  const int buckets = (int)(hist.numel() / 2);
+  auto out = at::zeros({3}, hist.options().dtype(at::kDouble));
+  lightctr::auc_scan_launch((const unsigned int*)hist.data_ptr(), buckets,
+                            out.data_ptr<double>(), cur_stream());
+  return out;  // {correct-pair mass, P, N}
+}
+
 at::Tensor bitmap_compact(at::Tensor bitmap, at::Tensor out_fids,
                           at::Tensor out_count) {
   CHK(bitmap.is_cuda() && bitmap.is_contiguous(), "bitmap");
@@ -881,6 +906,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("nfm_backward_emit", &nfm_backward_emit,
         "NFM per-entry grads for sorted apply");
   m.def("bitmap_compact", &bitmap_compact, "touched bitmap -> fid list");
+  m.def("auc_hist_add", &auc_hist_add,
+        "atomic pos/neg histogram accumulate for AUC");
+  m.def("auc_scan", &auc_scan,
+        "device scan of the AUC histogram -> {correct, P, N}");
   m.def("fm_adagrad_apply", &fm_adagrad_apply, "sparse fused Adagrad");
   m.def("fm_ftrl_apply", &fm_ftrl_apply,
         "sparse fused FTRL-proximal (optionally Adagrad on V)",

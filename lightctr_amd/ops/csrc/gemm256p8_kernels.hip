@@ -297,9 +297,12 @@ void gemm256p8_bf16_launch(const void* A, const void* Bst, const float* bias,
                            hipStream_t stream) {
   dim3 block(512);
   dim3 grid(N / 256, M / 256);
+  // LITE measured faster than the 8-barrier variant at every shape
+  // (596 vs 555 TF on the W&D 65536x256x624 tail, 949 vs 904 @4k^3 —
+  // profiles/r2_04_gemm_p8lite.txt): default on, =0 reverts.
   static const bool lite = [] {
     const char* e = getenv("LCTR_GEMM_P8_LITE");
-    return e && e[0] == '1';
+    return !(e && e[0] == '0');
   }();
   if (K % P8_BK == 0) {
     if (lite)

@@ -94,6 +94,11 @@ void ps_apply_launch(const long* lidx, int n, const float* gW,
                      float* nV, float* shadowW, float* shadowV, int K,
                      int updater, float lr, float lam, float eps,
                      ihipStream_t* stream);
+void auc_hist_add_launch(const float* pred, const float* label, long n,
+                         unsigned int* hist, int buckets,
+                         ihipStream_t* stream);
+void auc_scan_launch(const unsigned int* hist, int buckets, double* out,
+                     ihipStream_t* stream);
 void sparse_adagrad_apply_launch(const int* uniq, const int* count, float* W,
                                  float* V, float* nW, float* nV, float* gradW,
                                  float* gradV, float lr, float eps, float l2,

@@ -198,6 +198,126 @@ __global__ void ps_apply_kernel(const long* __restrict__ lidx, int n,
   }
 }
 
+// ---------------------------------------------------------------------------
+// Histogram AUC (reference util/evaluator.h:61-94: 2^24-bucket pos/neg
+// histogram + trapezoid integration) as device kernels: an atomic
+// histogram accumulate and a single-workgroup device scan that computes
+// correct-pair mass + totals. Semantics match utils/metrics.HistAUC.
+// ---------------------------------------------------------------------------
+
+__global__ void auc_hist_add_kernel(const float* __restrict__ pred,
+                                    const float* __restrict__ label, long n,
+                                    unsigned int* __restrict__ hist,
+                                    int buckets) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long j = i; j < n; j += stride) {
+    float p = pred[j];
+    p = fminf(1.f, fmaxf(0.f, p));
+    // floor, matching the host path's `(pred * (buckets-1)).long()`
+    int b = (int)(p * (buckets - 1));
+    b = min(buckets - 1, max(0, b));
+    const int pos = label[j] > 0.5f ? 1 : 0;
+    atomicAdd(&hist[(size_t)pos * buckets + b], 1u);
+  }
+}
+
+// One workgroup (1024 threads = 16 waves): two passes over the buckets.
+// Pass 1 reduces P and N; pass 2 runs a block-wide inclusive scan of the
+// positive counts chunk by chunk and accumulates
+//   correct = sum_b neg_b * (P - cumpos_incl_b) + 0.5 * neg_b * pos_b
+// out = {correct, P, N} (fp64).
+__global__ void auc_scan_kernel(const unsigned int* __restrict__ hist,
+                                int buckets, double* __restrict__ out) {
+  __shared__ long wave_tot[16];
+  __shared__ double wave_red[16];
+  __shared__ long running_s;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wv = tid >> 6;
+
+  // pass 1: totals
+  long pos_t = 0, neg_t = 0;
+  for (int b = tid; b < buckets; b += blockDim.x) {
+    neg_t += hist[b];
+    pos_t += hist[(size_t)buckets + b];
+  }
+  double pt = (double)pos_t, nt = (double)neg_t;
+#pragma unroll
+  for (int s = 1; s < LCTR_WAVE; s <<= 1) {
+    pt += __shfl_xor(pt, s);
+    nt += __shfl_xor(nt, s);
+  }
+  if (lane == 0) wave_red[wv] = pt;
+  __syncthreads();
+  double P = 0;
+  for (int w = 0; w < 16; ++w) P += wave_red[w];
+  __syncthreads();
+  if (lane == 0) wave_red[wv] = nt;
+  __syncthreads();
+  double N = 0;
+  for (int w = 0; w < 16; ++w) N += wave_red[w];
+  __syncthreads();
+
+  // pass 2: chunked inclusive scan of pos + correct-mass accumulate
+  if (tid == 0) running_s = 0;
+  __syncthreads();
+  double corr = 0;
+  for (int base = 0; base < buckets; base += blockDim.x) {
+    const int b = base + tid;
+    const long pos = b < buckets ? hist[(size_t)buckets + b] : 0;
+    const long neg = b < buckets ? hist[b] : 0;
+    // block inclusive scan of pos
+    long scan = pos;
+#pragma unroll
+    for (int s = 1; s < LCTR_WAVE; s <<= 1) {
+      const long t = __shfl_up(scan, s);
+      if (lane >= s) scan += t;
+    }
+    if (lane == LCTR_WAVE - 1) wave_tot[wv] = scan;
+    __syncthreads();
+    long wbase = running_s;
+    for (int w = 0; w < wv; ++w) wbase += wave_tot[w];
+    const long cum_incl = wbase + scan;
+    corr += (double)neg * (P - (double)cum_incl) +
+            0.5 * (double)neg * (double)pos;
+    __syncthreads();
+    if (tid == 0) {
+      long tot = 0;
+      for (int w = 0; w < 16; ++w) tot += wave_tot[w];
+      running_s += tot;
+    }
+    __syncthreads();
+  }
+#pragma unroll
+  for (int s = 1; s < LCTR_WAVE; s <<= 1) corr += __shfl_xor(corr, s);
+  if (lane == 0) wave_red[wv] = corr;
+  __syncthreads();
+  if (tid == 0) {
+    double c = 0;
+    for (int w = 0; w < 16; ++w) c += wave_red[w];
+    out[0] = c;
+    out[1] = P;
+    out[2] = N;
+  }
+}
+
+void auc_hist_add_launch(const float* pred, const float* label, long n,
+                         unsigned int* hist, int buckets,
+                         hipStream_t stream) {
+  if (n <= 0) return;
+  dim3 block(256);
+  dim3 grid((int)min((long)4096, (n + 255) / 256));
+  hipLaunchKernelGGL(auc_hist_add_kernel, grid, block, 0, stream, pred,
+                     label, n, hist, buckets);
+}
+
+void auc_scan_launch(const unsigned int* hist, int buckets, double* out,
+                     hipStream_t stream) {
+  hipLaunchKernelGGL(auc_scan_kernel, dim3(1), dim3(1024), 0, stream, hist,
+                     buckets, out);
+}
+
 void ps_apply_launch(const long* lidx, int n, const float* gW,
                      const float* gV, float* W, float* V, float* nW,
                      float* nV, float* shadowW, float* shadowV, int K,

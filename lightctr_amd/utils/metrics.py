@@ -13,15 +13,37 @@ import torch
 
 
 class HistAUC:
-    """Streaming histogram AUC: add(pred, label) batches, then compute()."""
+    """Streaming histogram AUC: add(pred, label) batches, then compute().
+
+    On GPU the histogram accumulate and the scan run as in-tree HIP
+    kernels (ops/csrc/misc_kernels.hip auc_hist_add/auc_scan — the §2.4
+    "atomic-histogram kernel + device scan" row for evaluator.h:61-94);
+    the CPU path keeps the torch.bincount reference implementation.
+    """
 
     def __init__(self, buckets: int = 1 << 20, device: str = "cpu"):
         self.buckets = buckets
-        self.pos = torch.zeros(buckets, dtype=torch.float64, device=device)
-        self.neg = torch.zeros(buckets, dtype=torch.float64, device=device)
+        self.device = torch.device(device)
+        self._gpu = self.device.type == "cuda"
+        if self._gpu:
+            from ..ops._extension import require_hip_ops
+
+            self._ops = require_hip_ops()
+            # [2, buckets] uint32 as int32 storage: [0]=neg, [1]=pos
+            self.hist = torch.zeros(2 * buckets, dtype=torch.int32,
+                                    device=self.device)
+        else:
+            self.pos = torch.zeros(buckets, dtype=torch.float64,
+                                   device=device)
+            self.neg = torch.zeros(buckets, dtype=torch.float64,
+                                   device=device)
 
     def add(self, pred: torch.Tensor, label: torch.Tensor) -> None:
         """pred in [0,1] (probabilities), label in {0,1}."""
+        if self._gpu:
+            self._ops.auc_hist_add(pred.float().contiguous(),
+                                   label.float().contiguous(), self.hist)
+            return
         idx = (pred.clamp(0, 1) * (self.buckets - 1)).long()
         lb = label.bool()
         if lb.any():
@@ -34,6 +56,12 @@ class HistAUC:
             ).to(self.neg.dtype)
 
     def compute(self) -> float:
+        if self._gpu:
+            out = self._ops.auc_scan(self.hist)
+            correct, npos, nneg = (float(x) for x in out.cpu())
+            if npos == 0 or nneg == 0:
+                return 0.5
+            return correct / (npos * nneg)
         npos = self.pos.sum()
         nneg = self.neg.sum()
         if npos == 0 or nneg == 0:

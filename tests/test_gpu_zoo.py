@@ -153,3 +153,32 @@ def test_embedding_gpu_fused_kernel_trains():
     within = (sims[a][:, a].sum() - 5) / 20
     cross = sims[a][:, b].mean()
     assert float(within) > float(cross) + 0.1, (float(within), float(cross))
+
+
+@pytest.mark.gpu
+def test_auc_hist_kernel_matches_exact():
+    """GPU atomic-histogram + device-scan AUC (evaluator.h:61-94 row)
+    matches the exact rank AUC to 1e-4 on 10M predictions."""
+    from lightctr_amd.utils.metrics import HistAUC, auc_score
+
+    g = torch.Generator().manual_seed(3)
+    n = 10_000_000
+    # separable-ish scores: positives shifted up
+    lab = (torch.rand(n, generator=g) > 0.7).float()
+    raw = torch.randn(n, generator=g) * 0.8 + lab * 0.9
+    pred = torch.sigmoid(raw)
+    h = HistAUC(buckets=1 << 22, device="cuda:0")
+    # streamed adds (two chunks) exercise accumulation
+    h.add(pred[: n // 2].cuda(), lab[: n // 2].cuda())
+    h.add(pred[n // 2:].cuda(), lab[n // 2:].cuda())
+    got = h.compute()
+    exact = auc_score(pred[:1_000_000], lab[:1_000_000])
+    # histogram vs exact on the full set: compare against exact on the
+    # full set too (rank AUC on 10M is fine on CPU)
+    exact_full = auc_score(pred, lab)
+    assert abs(got - exact_full) < 1e-4, (got, exact_full)
+    assert abs(got - exact) < 5e-3  # sanity vs the 1M subsample
+    # CPU HistAUC agrees with the kernel
+    hc = HistAUC(buckets=1 << 18, device="cpu")
+    hc.add(pred, lab)
+    assert abs(hc.compute() - got) < 1e-4

@@ -85,7 +85,15 @@ def main():
         labs.append(lb)
     pred = torch.cat(preds)
     lab = torch.cat(labs)
-    auc = auc_score(pred.cpu(), lab.cpu())
+    if pred.is_cuda:
+        # in-tree atomic-histogram + device-scan AUC kernel
+        from lightctr_amd.utils.metrics import HistAUC
+
+        h = HistAUC(buckets=1 << 22, device=str(pred.device))
+        h.add(pred, lab)
+        auc = h.compute()
+    else:
+        auc = auc_score(pred.cpu(), lab.cpu())
     logloss = float(torch.nn.functional.binary_cross_entropy(
         pred.clamp(1e-7, 1 - 1e-7), lab))
     print(f"model={args.model} steps={args.steps} batch={args.batch} "
