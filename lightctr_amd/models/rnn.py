@@ -8,9 +8,12 @@ backward :77-118; train/train_rnn_algo.h: 28-step LSTM over image rows ->
 attention -> FC classifier).
 
 Rebuilt batched: gates for the whole batch are one [B, 4H] GEMM per step
-(plain library GEMM via torch on ROCm = hipBLASLt — the sanctioned path
-for non-flagship dense ops), BPTT is the exact manual reverse pass, and
-updates are per-tensor Adagrad (the reference uses 12 AdagradUpdater_Num
+(reference lstm_unit.h:324-334 Matrix::Multiply hot site). On GPU the
+gate/attention GEMMs run through the IN-TREE MFMA bf16 kernels
+(ops/csrc/gemm_kernels.hip, fp32 master weights, bf16 operands, fused
+bias); on CPU (and for the exact-vs-autograd tests) the fp32 torch
+matmul path is kept. BPTT is the exact manual reverse pass, and updates
+are per-tensor Adagrad (the reference uses 12 AdagradUpdater_Num
 instances; here the same optimizer over the packed tensors).
 """
 
@@ -33,11 +36,30 @@ class _Adagrad:
             p -= self.lr * g / (n + self.eps).sqrt()
 
 
+def _hip_gemm_ok(device) -> bool:
+    if torch.device(device).type != "cuda":
+        return False
+    from ..ops._extension import has_hip_ops
+
+    return has_hip_ops()
+
+
+def _gemm_bf16(A, Bst, bias, M, N, K, transA=0, transB=0):
+    """C = op(A) @ op(Bst)^T via the in-tree MFMA GEMM (bf16 operands,
+    fp32 accumulate/out)."""
+    from ..ops._extension import require_hip_ops
+
+    ops = require_hip_ops()
+    return ops.gemm_bf16(A.to(torch.bfloat16).contiguous(),
+                         Bst.to(torch.bfloat16).contiguous(), bias,
+                         M, N, K, transA, transB, 0, False)
+
+
 class LSTMUnit:
     """Batched 4-gate LSTM with stored sequence history + exact BPTT."""
 
     def __init__(self, in_dim: int, hidden: int, seed=0, device="cpu",
-                 clip: float = 15.0):
+                 clip: float = 15.0, backend: str | None = None):
         # elementwise delta clipping at +-clip during BPTT (the reference
         # LSTM's error_clip_threshold = 15, lstm_unit.h:56,177-179)
         self.clip = clip
@@ -49,6 +71,10 @@ class LSTMUnit:
         self.b = torch.zeros(4 * hidden, device=device)
         self.b[hidden:2 * hidden] = 1.0  # forget-gate bias
         self.device = device
+        # "hip_bf16" routes the gate GEMMs through the in-tree MFMA
+        # kernels; "torch" is the fp32 exact path (CPU + autograd tests)
+        self.backend = backend or (
+            "hip_bf16" if _hip_gemm_ok(device) else "torch")
 
     def forward(self, xs: torch.Tensor, train=True):
         """xs: [B, T, D]. Returns hs [B, T, H]."""
@@ -58,9 +84,14 @@ class LSTMUnit:
         c = torch.zeros(B, H, device=xs.device)
         cache = []
         hs = []
+        hip = self.backend == "hip_bf16"
+        Wbf = self.W.to(torch.bfloat16) if hip else None
         for t in range(T):
             z = torch.cat([xs[:, t, :], h], dim=1)  # [B, D+H]
-            a = z @ self.W.t() + self.b
+            if hip:
+                a = _gemm_bf16(z, Wbf, self.b, B, 4 * H, D + H)
+            else:
+                a = z @ self.W.t() + self.b
             i = torch.sigmoid(a[:, :H])
             f = torch.sigmoid(a[:, H:2 * H])
             g = torch.tanh(a[:, 2 * H:3 * H])
@@ -85,6 +116,8 @@ class LSTMUnit:
         dxs = torch.zeros(B, T, D, device=dhs.device)
         dh_next = torch.zeros(B, H, device=dhs.device)
         dc_next = torch.zeros(B, H, device=dhs.device)
+        hip = self.backend == "hip_bf16"
+        Wbf = self.W.to(torch.bfloat16) if hip else None
         for t in reversed(range(T)):
             z, i, f, g, o, c, tc = self._cache[t]
             c_prev = self._cache[t - 1][5] if t > 0 else torch.zeros_like(c)
@@ -99,9 +132,14 @@ class LSTMUnit:
             dc_next = dc * f
             da = torch.cat([di * i * (1 - i), df * f * (1 - f),
                             dg * (1 - g * g), do * o * (1 - o)], dim=1)
-            dW += da.t() @ z
+            if hip:
+                # dW = da^T @ z: [4H, D+H]; dz = da @ W: [B, D+H]
+                dW += _gemm_bf16(da, z, None, 4 * H, self.D + H, B, 1, 1)
+                dz = _gemm_bf16(da, Wbf, None, B, self.D + H, 4 * H, 0, 1)
+            else:
+                dW += da.t() @ z
+                dz = da @ self.W
             db += da.sum(dim=0)
-            dz = da @ self.W
             dxs[:, t, :] = dz[:, :D]
             dh_next = dz[:, D:]
         return dxs, dW, db
@@ -112,17 +150,26 @@ class AttentionUnit:
     steps; context = sum alpha_t h_t (reference attention_unit.h)."""
 
     def __init__(self, hidden: int, attn_dim: int = 32, seed=0,
-                 device="cpu"):
+                 device="cpu", backend: str | None = None):
         g = torch.Generator().manual_seed(seed + 5)
         self.Wa = (torch.randn(attn_dim, hidden, generator=g)
                    * (1.0 / hidden) ** 0.5).to(device)
         self.ba = torch.zeros(attn_dim, device=device)
         self.v = (torch.randn(attn_dim, generator=g)
                   * (1.0 / attn_dim) ** 0.5).to(device)
+        self.backend = backend or (
+            "hip_bf16" if _hip_gemm_ok(device) else "torch")
 
     def forward(self, hs: torch.Tensor, train=True):
         """hs: [B, T, H] -> context [B, H]."""
-        u = torch.tanh(hs @ self.Wa.t() + self.ba)  # [B, T, A]
+        B, T, H = hs.shape
+        A = self.Wa.shape[0]
+        if self.backend == "hip_bf16":
+            pre = _gemm_bf16(hs.reshape(B * T, H), self.Wa, self.ba,
+                             B * T, A, H).view(B, T, A)
+            u = torch.tanh(pre)
+        else:
+            u = torch.tanh(hs @ self.Wa.t() + self.ba)  # [B, T, A]
         scores = u @ self.v  # [B, T]
         alpha = torch.softmax(scores, dim=1)
         ctx = (alpha.unsqueeze(2) * hs).sum(dim=1)
@@ -141,9 +188,17 @@ class AttentionUnit:
         dv = torch.einsum("bt,bta->a", dscore, u)
         du = dscore.unsqueeze(2) * self.v.view(1, 1, -1)
         dpre = du * (1 - u * u)  # [B, T, A]
-        dWa = torch.einsum("bta,bth->ah", dpre, hs)
+        A = self.Wa.shape[0]
+        if self.backend == "hip_bf16":
+            d2 = dpre.reshape(B * T, A)
+            h2 = hs.reshape(B * T, H)
+            dWa = _gemm_bf16(d2, h2, None, A, H, B * T, 1, 1)
+            dhs = dhs + _gemm_bf16(d2, self.Wa, None, B * T, H, A,
+                                   0, 1).view(B, T, H)
+        else:
+            dWa = torch.einsum("bta,bth->ah", dpre, hs)
+            dhs = dhs + dpre @ self.Wa
         dba = dpre.sum(dim=(0, 1))
-        dhs = dhs + dpre @ self.Wa
         return dhs, dWa, dba, dv
 
 
