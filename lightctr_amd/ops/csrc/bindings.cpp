@@ -400,6 +400,36 @@ void sparse_ftrl_apply(at::Tensor uniq, at::Tensor count, at::Tensor W,
       cur_stream());
 }
 
+at::Tensor im2col_bf16(at::Tensor x, int64_t k, int64_t stride,
+                       int64_t pad) {
+  check_cuda_f32(x, "x");
+  CHK(x.dim() == 4, "x must be [B,C,H,W]");
+  const int B = (int)x.size(0), C = (int)x.size(1), H = (int)x.size(2),
+            W = (int)x.size(3);
+  const int OH = (int)((H + 2 * pad - k) / stride + 1);
+  const int OW = (int)((W + 2 * pad - k) / stride + 1);
+  auto col = at::empty({(long)B * OH * OW, (long)C * k * k},
+                       x.options().dtype(at::kBFloat16));
+  lightctr::im2col_bf16_launch(x.data_ptr<float>(), col.data_ptr(), B, C, H,
+                               W, (int)k, (int)stride, (int)pad, OH, OW,
+                               cur_stream());
+  return col;
+}
+
+at::Tensor col2im(at::Tensor dcol, int64_t B, int64_t C, int64_t H,
+                  int64_t W, int64_t k, int64_t stride, int64_t pad) {
+  check_cuda_f32(dcol, "dcol");
+  const int OH = (int)((H + 2 * pad - k) / stride + 1);
+  const int OW = (int)((W + 2 * pad - k) / stride + 1);
+  CHK(dcol.size(0) == (long)B * OH * OW && dcol.size(1) == C * k * k,
+      "dcol shape mismatch");
+  auto dx = at::empty({B, C, H, W}, dcol.options());
+  lightctr::col2im_launch(dcol.data_ptr<float>(), dx.data_ptr<float>(),
+                          (int)B, (int)C, (int)H, (int)W, (int)k,
+                          (int)stride, (int)pad, OH, OW, cur_stream());
+  return dx;
+}
+
 void auc_hist_add(at::Tensor pred, at::Tensor label, at::Tensor hist) {
   check_cuda_f32(pred, "pred");
   check_cuda_f32(label, "label");
@@ -906,6 +936,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("nfm_backward_emit", &nfm_backward_emit,
         "NFM per-entry grads for sorted apply");
   m.def("bitmap_compact", &bitmap_compact, "touched bitmap -> fid list");
+  m.def("im2col_bf16", &im2col_bf16,
+        "fused unfold -> bf16 GEMM operand [B*L, C*k*k]");
+  m.def("col2im", &col2im, "conv data-grad fold (gather form)");
   m.def("auc_hist_add", &auc_hist_add,
         "atomic pos/neg histogram accumulate for AUC");
   m.def("auc_scan", &auc_scan,

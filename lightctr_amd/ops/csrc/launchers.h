@@ -143,6 +143,12 @@ void gemm256_bf16_launch(const void* A, const void* Bst, const float* bias,
 // --- nn_kernels.hip ---
 void small_wgrad_launch(const void* Ast, const void* Bst, float* C, int M,
                         int N, int K, ihipStream_t* stream);
+void im2col_bf16_launch(const float* x, void* col, int B, int C, int H,
+                        int W, int k, int stride, int pad, int OH, int OW,
+                        ihipStream_t* stream);
+void col2im_launch(const float* dcol, float* dx, int B, int C, int H, int W,
+                   int k, int stride, int pad, int OH, int OW,
+                   ihipStream_t* stream);
 void gemv_n1_launch(const void* A, const void* b, const float* bias,
                     float* C, void* Cbf, int M, int K, int act,
                     ihipStream_t* stream);

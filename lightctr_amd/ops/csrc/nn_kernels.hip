@@ -338,4 +338,95 @@ void outer_k1_launch(const void* A, const void* Bst, const float* bias,
                      (__bf16*)Cbf, M, N, act);
 }
 
+// ---------------------------------------------------------------------------
+// In-tree convolution im2col / col2im (reference util/matrix.h:237-319
+// Matrix::convolution/deconvolution, SURVEY §2.4 "im2col+MFMA or direct
+// conv" row). im2col gathers patches STRAIGHT into the bf16 GEMM operand
+// layout [B*L, C*k*k] (one fused pass instead of F.unfold fp32 +
+// transpose + contiguous + to_bf16); col2im is the gather-form data-grad
+// (each input pixel sums the k*k output taps that touched it — no
+// atomics).
+// ---------------------------------------------------------------------------
+
+__global__ void im2col_bf16_kernel(const float* __restrict__ x,
+                                   __bf16* __restrict__ col, int B, int C,
+                                   int H, int W, int k, int stride, int pad,
+                                   int OH, int OW) {
+  const long L = (long)OH * OW;
+  const long total = (long)B * L * C * k * k;
+  const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long gstride = (long)gridDim.x * blockDim.x;
+  const int kk = k * k;
+  for (long i = i0; i < total; i += gstride) {
+    // i = ((b*L + l) * C*k*k) + (c*k + kh)*k + kw
+    const long row = i / (C * kk);
+    const int cc = (int)(i - row * (C * kk));
+    const int c = cc / kk;
+    const int kh = (cc % kk) / k;
+    const int kw = cc % k;
+    const long b = row / L;
+    const int l = (int)(row - b * L);
+    const int oh = l / OW, ow = l - (l / OW) * OW;
+    const int h = oh * stride - pad + kh;
+    const int w = ow * stride - pad + kw;
+    float v = 0.f;
+    if (h >= 0 && h < H && w >= 0 && w < W)
+      v = x[((b * C + c) * (long)H + h) * W + w];
+    col[i] = (__bf16)v;
+  }
+}
+
+__global__ void col2im_kernel(const float* __restrict__ dcol,
+                              float* __restrict__ dx, int B, int C, int H,
+                              int W, int k, int stride, int pad, int OH,
+                              int OW) {
+  const long total = (long)B * C * H * W;
+  const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long gstride = (long)gridDim.x * blockDim.x;
+  const long L = (long)OH * OW;
+  const int kk = k * k;
+  for (long i = i0; i < total; i += gstride) {
+    const int w = (int)(i % W);
+    const int h = (int)((i / W) % H);
+    const int c = (int)((i / ((long)W * H)) % C);
+    const long b = i / ((long)W * H * C);
+    float acc = 0.f;
+    for (int kh = 0; kh < k; ++kh) {
+      const int oh_num = h + pad - kh;
+      if (oh_num < 0 || oh_num % stride) continue;
+      const int oh = oh_num / stride;
+      if (oh >= OH) continue;
+      for (int kw = 0; kw < k; ++kw) {
+        const int ow_num = w + pad - kw;
+        if (ow_num < 0 || ow_num % stride) continue;
+        const int ow = ow_num / stride;
+        if (ow >= OW) continue;
+        const long row = b * L + (long)oh * OW + ow;
+        acc += dcol[row * (C * kk) + ((long)c * k + kh) * k + kw];
+      }
+    }
+    dx[i] = acc;
+  }
+}
+
+void im2col_bf16_launch(const float* x, void* col, int B, int C, int H,
+                        int W, int k, int stride, int pad, int OH, int OW,
+                        hipStream_t stream) {
+  const long total = (long)B * OH * OW * C * k * k;
+  dim3 block(256);
+  dim3 grid((unsigned)min((long)8192, (total + 255) / 256));
+  hipLaunchKernelGGL(im2col_bf16_kernel, grid, block, 0, stream, x,
+                     (__bf16*)col, B, C, H, W, k, stride, pad, OH, OW);
+}
+
+void col2im_launch(const float* dcol, float* dx, int B, int C, int H, int W,
+                   int k, int stride, int pad, int OH, int OW,
+                   hipStream_t stream) {
+  const long total = (long)B * C * H * W;
+  dim3 block(256);
+  dim3 grid((unsigned)min((long)8192, (total + 255) / 256));
+  hipLaunchKernelGGL(col2im_kernel, grid, block, 0, stream, dcol, dx, B, C,
+                     H, W, k, stride, pad, OH, OW);
+}
+
 }  // namespace lightctr

@@ -182,3 +182,47 @@ def test_auc_hist_kernel_matches_exact():
     hc = HistAUC(buckets=1 << 18, device="cpu")
     hc.add(pred, lab)
     assert abs(hc.compute() - got) < 1e-4
+
+
+def test_im2col_col2im_parity_and_cnn_trains():
+    """In-tree im2col/col2im kernels (matrix.h:237-319 row) vs the
+    F.unfold/F.fold reference, incl. stride+padding; then the LeNet-style
+    CNN trains end-to-end through them."""
+    import torch.nn.functional as F
+
+    from lightctr_amd.ops import hip_ops
+
+    g = torch.Generator().manual_seed(11)
+    for (B, C, H, W, k, st, pad) in [(4, 3, 14, 14, 3, 1, 1),
+                                     (2, 1, 28, 28, 5, 2, 2),
+                                     (3, 6, 13, 13, 3, 2, 0)]:
+        x = torch.randn(B, C, H, W, generator=g).cuda()
+        col = hip_ops.im2col_bf16(x, k, st, pad)
+        ref = F.unfold(x, k, stride=st, padding=pad)  # [B, C*k*k, L]
+        L = ref.shape[2]
+        ref2 = ref.transpose(1, 2).reshape(B * L, -1)
+        assert torch.allclose(col.float(), ref2.to(torch.bfloat16).float(),
+                              atol=1e-2), (B, C, H, W, k, st, pad)
+        dcol = torch.randn(B * L, C * k * k, generator=g).cuda()
+        dx = hip_ops.col2im(dcol, B, C, H, W, k, st, pad)
+        dref = F.fold(dcol.view(B, L, -1).transpose(1, 2), (H, W), k,
+                      stride=st, padding=pad)
+        assert torch.allclose(dx, dref, atol=1e-4, rtol=1e-4), \
+            (dx - dref).abs().max()
+
+    from lightctr_amd.models.cnn import CNNHyper, CNNModel
+
+    B = 64
+    X = torch.rand(B, 1, 16, 16, generator=g) * 0.2
+    y = torch.randint(0, 4, (B,), generator=g)
+    hs = 8
+    img = X.view(B, 16, 16)
+    for i in range(B):
+        q = int(y[i])
+        img[i, (q // 2) * hs:(q // 2 + 1) * hs,
+            (q % 2) * hs:(q % 2 + 1) * hs] += 0.8
+    m = CNNModel(CNNHyper(in_shape=(1, 16, 16), n_classes=4, lr=3e-3),
+                 device="cuda:0")
+    X, y = X.view(B, 1, 16, 16).cuda(), y.cuda()
+    losses = [m.train_step(X, y) for _ in range(40)]
+    assert losses[-1] < losses[0] * 0.7, (losses[0], losses[-1])
