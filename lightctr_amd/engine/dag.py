@@ -259,12 +259,97 @@ class LossOp(Node):
         return (grad * g, torch.zeros_like(self._y))
 
 
+class GraphExecutor:
+    """GPU-credible scheduler for a STATIC dag (round-2):
+    one topological order computed once, forward + backward issued
+    sequentially with no thread pool and no per-node locks (kernels queue
+    asynchronously on the HIP stream — Python never blocks on device
+    work), and after a warmup step the whole train step is captured into
+    a hipGraph (torch.cuda.CUDAGraph) and replayed, making the step
+    launch-bound instead of Python-bound. The threaded future/promise
+    path above stays as the dynamic-graph capability-parity engine
+    (reference node_abst.h:57-87 semantics)."""
+
+    def __init__(self, terminus: "TerminusNode", capture: bool = True):
+        self.terminus = terminus
+        self.order = self._topo(terminus)
+        self.capture = capture
+        self._graph = None
+        self._calls = 0
+
+    @staticmethod
+    def _topo(root: Node) -> list[Node]:
+        order, seen = [], set()
+
+        def visit(n: Node):
+            if id(n) in seen:
+                return
+            seen.add(id(n))
+            for i in n.inputs:
+                visit(i)
+            order.append(n)
+
+        visit(root)
+        return order
+
+    def _run_seq(self):
+        for n in self.order:
+            n.value = n.forward_compute(*[i.value for i in n.inputs])
+        # backward in reverse topo order: when a node is reached, all of
+        # its consumers (later in topo order) have contributed grads
+        grads: dict[int, torch.Tensor] = {
+            id(self.terminus): torch.ones_like(self.terminus.value)}
+        for n in reversed(self.order):
+            g = grads.pop(id(n), None)
+            if g is None:
+                continue  # node not on a grad path
+            n.grad = g
+            in_grads = n.backward_compute(g)
+            for inp, ig in zip(n.inputs, in_grads):
+                prev = grads.get(id(inp))
+                grads[id(inp)] = ig if prev is None else prev + ig
+        return self.terminus.value
+
+    def step(self):
+        dev = self.terminus_device()
+        if self._graph is not None:
+            self._graph.replay()
+            return self.terminus.value
+        loss = self._run_seq()
+        self._calls += 1
+        if (self.capture and dev is not None and dev.type == "cuda"
+                and self._calls == 2):
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            try:
+                with torch.cuda.graph(g):
+                    self._run_seq()
+                self._graph = g
+                # capture records without executing: replay once so this
+                # call still performs a real training step
+                g.replay()
+                loss = self.terminus.value
+            except RuntimeError:
+                self.capture = False  # capture-unsupported op: stay eager
+        return loss
+
+    def terminus_device(self):
+        v = self.terminus.value
+        if v is None:
+            for n in self.order:
+                if isinstance(n, SourceNode):
+                    return n.value_init.device
+            return None
+        return v.device
+
+
 class DAGPipeline:
     """Wiring helper (reference dag_pipeline.h:33-37)."""
 
     def __init__(self):
         self.sources: list[Node] = []
         self.terminus: TerminusNode | None = None
+        self._executor: GraphExecutor | None = None
 
     def add_flow(self, *nodes):
         for n in nodes:
@@ -284,7 +369,20 @@ class DAGPipeline:
             torch.ones_like(self.terminus.value))
         self.terminus.backward_run()
 
-    def step(self):
-        loss = self.run_forward()
-        self.run_backward()
-        return loss
+    def step(self, engine: str = "auto"):
+        """engine: "auto" (GraphExecutor with hipGraph capture on GPU,
+        threaded elsewhere), "graph", or "threaded"."""
+        if engine == "threaded":
+            loss = self.run_forward()
+            self.run_backward()
+            return loss
+        if self._executor is None:
+            self._executor = GraphExecutor(self.terminus)
+        use_graph = engine == "graph" or (
+            self._executor.terminus_device() is not None
+            and self._executor.terminus_device().type == "cuda")
+        if engine == "auto" and not use_graph:
+            loss = self.run_forward()
+            self.run_backward()
+            return loss
+        return self._executor.step()

@@ -106,6 +106,11 @@ def main():
     log(rank, "E ring allreduce + broadcast ok")
 
     # ---- D: PS mode (P2P pair groups, int8 wire, DCASGD, SSP) ----
+    if world < 2:
+        dist.barrier()
+        if rank == 0:
+            print("SHAKEOUT OK (world 1: PS P2P part skipped)", flush=True)
+        return
     from lightctr_amd.parallel.ps import (PSConfig, PSShard, ps_train_fm,
                                           setup_pair_groups)
 
