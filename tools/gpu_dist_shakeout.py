@@ -48,7 +48,10 @@ def main():
     F = 1 << 18
 
     # ---- A: sharded FM over NCCL all-to-all ----
-    h = FMHyper(num_features=F, k=16, optimizer="adagrad", seed=7)
+    # lr=0.02: the default 0.1 transiently diverges at this batch/feature
+    # density even for the single-process model (verified on CPU) — the
+    # shakeout pins transport, so keep the dynamics tame
+    h = FMHyper(num_features=F, k=16, optimizer="adagrad", lr=0.02, seed=7)
     m = ShardedFMModel(h, device=dev, wire="fp16")
     gen = SyntheticCriteo(num_features=F, seed=100 + rank, device=dev)
     losses = []
