@@ -474,6 +474,89 @@ __global__ void fm_segscan_apply_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// Sorted backward, phase 2, QUAD-PER-ENTRY segmented scan (round 2,
+// docs/STATUS.md lever 3): K/4 lanes per entry, each holding one float4
+// quad of the gv row (+ gw on the first lane) — 5 live registers instead
+// of the 17 that made the lane-per-entry segscan lose to the walk. The
+// wave covers 64/(K/4) consecutive sorted entries; run totals come from
+// a log2-step segmented inclusive scan over entry slots (shfl_up at
+// entry stride), tails flush with one dwordx4 store per lane (plain for
+// in-wave-head runs — slabs zeroed by the optimizer pass — atomics only
+// for wave-spanning runs).
+// ---------------------------------------------------------------------------
+template <int K>
+__global__ void fm_segscan4_apply_kernel(
+    const int* __restrict__ sorted_fids, const int* __restrict__ perm,
+    const float* __restrict__ gw, const float* __restrict__ gv,
+    float* __restrict__ gradW, float* __restrict__ gradV,
+    unsigned long long* __restrict__ touched, int nnz) {
+  static_assert(K % 4 == 0 && K >= 4 && K <= 64, "quad layout needs K%4==0");
+  constexpr int L = K / 4;            // lanes per entry
+  constexpr int EPW = LCTR_WAVE / L;  // entries per wave
+  const int lane = threadIdx.x & (LCTR_WAVE - 1);
+  const int wave = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  const long base = (long)wave * EPW;
+  if (base >= nnz) return;
+  const int ei = lane / L;
+  const int q = lane % L;
+  const long e = base + ei;
+  const bool valid = e < nnz;
+
+  const int fid = valid ? sorted_fids[e] : -1;
+  const bool head0 = valid && ((e == 0) || (sorted_fids[e - 1] != fid));
+
+  float4 v = make_float4(0.f, 0.f, 0.f, 0.f);
+  float w = 0.f;
+  if (valid) {
+    const long p = perm ? (long)perm[e] : e;
+    v = ((const float4*)&gv[(size_t)p * K])[q];
+    if (q == 0) w = gw[p];
+  }
+
+  unsigned f = head0 ? 1u : 0u;
+#pragma unroll
+  for (int s = L; s < LCTR_WAVE; s <<= 1) {
+    const unsigned tf = __shfl_up(f, s);
+    float4 tv;
+    tv.x = __shfl_up(v.x, s);
+    tv.y = __shfl_up(v.y, s);
+    tv.z = __shfl_up(v.z, s);
+    tv.w = __shfl_up(v.w, s);
+    const float tw = __shfl_up(w, s);
+    if ((lane >= s) && (f == 0u)) {
+      v.x += tv.x;
+      v.y += tv.y;
+      v.z += tv.z;
+      v.w += tv.w;
+      w += tw;
+    }
+    if (lane >= s) f |= tf;
+  }
+
+  const bool tail = valid && ((e + 1 >= nnz) || (sorted_fids[e + 1] != fid));
+  if (tail && f) {
+    // run's head is in-wave -> this entry's lanes hold the TOTAL and own
+    // the feature exclusively: plain coalesced row store
+    ((float4*)&gradV[(size_t)fid * K])[q] = v;
+    if (q == 0) {
+      gradW[fid] = w;
+      atomicOr(&touched[fid >> 6], 1ull << (fid & 63));
+    }
+  } else if (tail || (valid && ei == EPW - 1)) {
+    // spanning run (tail without in-wave head, or wave's last entry with
+    // the run continuing): atomic partial flush
+    atomicAdd(&gradV[(size_t)fid * K + q * 4 + 0], v.x);
+    atomicAdd(&gradV[(size_t)fid * K + q * 4 + 1], v.y);
+    atomicAdd(&gradV[(size_t)fid * K + q * 4 + 2], v.z);
+    atomicAdd(&gradV[(size_t)fid * K + q * 4 + 3], v.w);
+    if (q == 0) {
+      atomicAdd(&gradW[fid], w);
+      atomicOr(&touched[fid >> 6], 1ull << (fid & 63));
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Bitmap -> unique-fid list compaction. One thread per 64-feature word;
 // clears the word as it goes so the bitmap is reusable next step.
 // ---------------------------------------------------------------------------
@@ -658,9 +741,22 @@ void fm_sorted_apply_launch(const int* sorted_fids, const int* perm,
                             int chunk, hipStream_t stream) {
   if (nnz <= 0) return;
   // chunk 0 = auto (walk kernel, chunk 384 — measured best), -1 = the
-  // segmented-scan kernel (correct, measured 275 vs 218 us: the 6-step
-  // 17-register shfl_up chain costs more than the serial walk saves;
-  // kept selectable + parity-tested as a documented negative result)
+  // lane-per-entry segmented-scan kernel (correct, measured 275 vs 218
+  // us: the 6-step 17-register shfl_up chain costs more than the serial
+  // walk saves; kept selectable + parity-tested as a documented negative
+  // result), -2 = the round-2 quad-per-entry segscan (K/4 lanes per
+  // entry, 5 live registers).
+  if (chunk == -2 && (K % 4 == 0) && K <= 64 && opt_mode == 0) {
+    const int wpb = waves_per_block();
+    const int epw = LCTR_WAVE / (K / 4);
+    const int nwaves = (nnz + epw - 1) / epw;
+    dim3 block(wpb * LCTR_WAVE);
+    dim3 grid((nwaves + wpb - 1) / wpb);
+    DISPATCH_K(K, hipLaunchKernelGGL((fm_segscan4_apply_kernel<KC>), grid,
+                                     block, 0, stream, sorted_fids, perm,
+                                     gw, gv, gradW, gradV, touched, nnz));
+    return;
+  }
   const bool can_scan =
       (K == 4 || K == 8 || K == 16) && opt_mode == 0;
   if (chunk == -1 && can_scan) {

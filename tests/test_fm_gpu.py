@@ -347,7 +347,8 @@ def test_segscan_apply_matches_walk():
     sorted_fids, perm = sort_ids(fids, F)
 
     out = []
-    for chunk in (0, -1):
+    # 0 = walk, -1 = lane-per-entry segscan, -2 = quad-per-entry segscan
+    for chunk in (0, -1, -2):
         gradW = torch.zeros(F, device="cuda")
         gradV = torch.zeros(F, K, device="cuda")
         touched = torch.zeros((F + 63) // 64, dtype=torch.int64,
@@ -355,6 +356,7 @@ def test_segscan_apply_matches_walk():
         ops.fm_sorted_apply(sorted_fids, perm, gw, gv, gradW, gradV,
                             touched, chunk)
         out.append((gradW, gradV, touched))
-    assert torch.allclose(out[0][0], out[1][0], atol=1e-5)
-    assert torch.allclose(out[0][1], out[1][1], atol=1e-5)
-    assert torch.equal(out[0][2], out[1][2])
+    for i in (1, 2):
+        assert torch.allclose(out[0][0], out[i][0], atol=1e-5), i
+        assert torch.allclose(out[0][1], out[i][1], atol=1e-5), i
+        assert torch.equal(out[0][2], out[i][2]), i

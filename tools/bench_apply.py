@@ -54,14 +54,14 @@ def main():
     gw_s, gv_s = ops.fm_backward_emit(row_ptr, fids, vals, m.V, sumVX, dpred)
     gw_p, gv_p = ops.fm_backward_emit(row_ptr, fids, vals, m.V, sumVX, dpred,
                                       pos)
-    for chunk in (-1, 256, 384, 512):
+    for chunk in (-2, -1, 256, 384, 512):
         a = t(lambda: ops.fm_sorted_apply(sorted_fids, perm, gw_s, gv_s,
                                           m.gradW, m.gradV, m.touched,
                                           chunk))
         b = t(lambda: ops.fm_sorted_apply(sorted_fids, None, gw_p, gv_p,
                                           m.gradW, m.gradV, m.touched,
                                           chunk))
-        tag = "segscan" if chunk == -1 else f"walk{chunk}"
+        tag = {-1: "segscan", -2: "segscan4"}.get(chunk, f"walk{chunk}")
         print(f"apply {tag:8s}  gather {a:8.1f} us   seq {b:8.1f} us")
 
     # correctness cross-check: gather vs scatter paths agree
@@ -82,6 +82,13 @@ def main():
     dw = (gW1 - m.gradW).abs().max().item()
     tok = bool(torch.equal(t1, m.touched))
     print(f"segscan vs walk gradV maxdiff {dmax:.3e} gradW {dw:.3e} touched_eq {tok}")
+    m.gradW.zero_(); m.gradV.zero_(); m.touched.zero_()
+    ops.fm_sorted_apply(sorted_fids, perm, gw_s, gv_s, m.gradW, m.gradV,
+                        m.touched, -2)
+    dmax = (gV1 - m.gradV).abs().max().item()
+    dw = (gW1 - m.gradW).abs().max().item()
+    tok = bool(torch.equal(t1, m.touched))
+    print(f"segscan4 vs walk gradV maxdiff {dmax:.3e} gradW {dw:.3e} touched_eq {tok}")
 
 
 if __name__ == "__main__":
