@@ -106,3 +106,55 @@ def test_plsa_separates_topics():
     # topic-word distributions concentrate on their half
     t0 = m.p_w_z[int(zd[0])]
     assert float(t0[:10].sum()) > 0.9 or float(t0[10:].sum()) > 0.9
+
+
+def test_proc_text_topic_tool(tmp_path):
+    """Text prep parity with reference data/proc_text_topic.py: vocab by
+    frequency (stopwords/markup dropped), dense count rows, and the
+    sparse triples feed PLSAModel.fit directly."""
+    import subprocess
+    import sys
+
+    src = tmp_path / "corpus.txt"
+    src.write_text(
+        "<doc id=1>\n"
+        "The cat sat on the mat with a cat\n"
+        "dogs and cats play ball\n"
+        "<doc id=2>\n"
+        "ball games are fun fun fun\n"
+        "the mat was red\n")
+    vocab = tmp_path / "vocab.txt"
+    train = tmp_path / "train.csv"
+    rc = subprocess.run(
+        [sys.executable, "tools/proc_text_topic.py", str(src), "10",
+         "--vocab-out", str(vocab), "--train-out", str(train)],
+        capture_output=True, text=True)
+    assert rc.returncode == 0, rc.stderr
+    lines = vocab.read_text().splitlines()
+    assert 0 < len(lines) <= 10
+    terms = {ln.split()[1]: (int(ln.split()[0]), int(ln.split()[2]))
+             for ln in lines}
+    assert "the" not in terms and "cat" in terms
+    assert terms["fun"][1] == 3  # corpus count survives
+    rows = train.read_text().splitlines()
+    assert len(rows) == 4  # markup lines dropped
+    assert all(len(r.split()) == len(lines) for r in rows)
+
+    # sparse mode triples train PLSA end-to-end
+    sp = tmp_path / "sparse.csv"
+    rc = subprocess.run(
+        [sys.executable, "tools/proc_text_topic.py", str(src), "10",
+         "--vocab-out", str(vocab), "--train-out", str(sp), "--sparse"],
+        capture_output=True, text=True)
+    assert rc.returncode == 0, rc.stderr
+    import torch
+
+    from lightctr_amd.models.plsa import PLSAHyper, PLSAModel
+
+    trip = [tuple(map(int, ln.split())) for ln in
+            sp.read_text().splitlines()]
+    d = torch.tensor([t[0] for t in trip])
+    w = torch.tensor([t[1] for t in trip])
+    c = torch.tensor([t[2] for t in trip])
+    m = PLSAModel(PLSAHyper(n_topics=2, max_iters=5, seed=1))
+    m.fit(d, w, c, n_docs=int(d.max()) + 1, n_words=len(lines))
