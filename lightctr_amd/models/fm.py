@@ -117,10 +117,22 @@ class FMModel:
                 sorted_fids, perm = sort_ids(fids, self.h.num_features)
                 gw, gv = ops.fm_backward_emit(row_ptr, fids, vals, self.V,
                                               sumVX, dpred)
-                if self.fused_apply and not (
-                        self.h.optimizer == "ftrl"
-                        and self.h.ftrl_v == "adagrad"):
-                    if self.h.optimizer == "ftrl":
+                if self.fused_apply:
+                    if (self.h.optimizer == "ftrl"
+                            and self.h.ftrl_v == "adagrad"):
+                        # round 2: mode 3 = FTRL on W + Adagrad on V (the
+                        # default pairing) fused into the segment reduce;
+                        # only boundary-spanning fids take the slab +
+                        # bitmap + sparse-apply pass
+                        ops.fm_sorted_apply_fused(
+                            sorted_fids, perm, gw, gv, self.gradW,
+                            self.gradV, self.touched, self.W, self.V,
+                            self.nW, self.nV, self.zW, None, 3,
+                            self.h.ftrl_alpha, self.h.ftrl_beta,
+                            self.h.ftrl_l1, self.h.ftrl_l2,
+                            v_lr=self.h.lr, v_eps=self.h.eps,
+                            v_l2=self.h.l2)
+                    elif self.h.optimizer == "ftrl":
                         ops.fm_sorted_apply_fused(
                             sorted_fids, perm, gw, gv, self.gradW,
                             self.gradV, self.touched, self.W, self.V,

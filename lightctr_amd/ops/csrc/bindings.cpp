@@ -149,7 +149,8 @@ void fm_sorted_apply(at::Tensor sorted_fids, c10::optional<at::Tensor> perm,
       gw.data_ptr<float>(), gv.data_ptr<float>(), gradW.data_ptr<float>(),
       gradV.data_ptr<float>(), (unsigned long long*)touched.data_ptr(),
       (int)sorted_fids.numel(), K, 0, nullptr, nullptr, nullptr, nullptr,
-      nullptr, nullptr, 0.f, 0.f, 0.f, 0.f, (int)chunk, cur_stream());
+      nullptr, nullptr, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, (int)chunk,
+      cur_stream());
 }
 
 // Fused variant: interior feature segments get their optimizer update
@@ -162,7 +163,8 @@ void fm_sorted_apply_fused(at::Tensor sorted_fids,
                            at::Tensor W, at::Tensor V, at::Tensor nW,
                            at::Tensor nV, c10::optional<at::Tensor> zW,
                            c10::optional<at::Tensor> zV, int64_t opt_mode,
-                           double p0, double p1, double p2, double p3) {
+                           double p0, double p1, double p2, double p3,
+                           double v_lr, double v_eps, double v_l2) {
   check_cuda_i32(sorted_fids, "sorted_fids");
   const int* perm_ptr = nullptr;
   at::Tensor perm_i;
@@ -170,7 +172,9 @@ void fm_sorted_apply_fused(at::Tensor sorted_fids,
     perm_i = perm32(*perm).contiguous();
     perm_ptr = perm_i.data_ptr<int>();
   }
-  CHK(opt_mode == 1 || opt_mode == 2, "opt_mode 1=adagrad 2=ftrl");
+  CHK(opt_mode == 1 || opt_mode == 2 || opt_mode == 3,
+      "opt_mode 1=adagrad 2=ftrl 3=ftrlW+adagradV");
+  if (opt_mode == 3) CHK(zW.has_value(), "opt_mode 3 needs zW");
   const int K = (int)gradV.size(1);
   lightctr::fm_sorted_apply_launch(
       sorted_fids.data_ptr<int>(), perm_ptr,
@@ -181,7 +185,8 @@ void fm_sorted_apply_fused(at::Tensor sorted_fids,
       zW.has_value() ? zW->data_ptr<float>() : nullptr,
       nV.data_ptr<float>(),
       zV.has_value() ? zV->data_ptr<float>() : nullptr, (float)p0,
-      (float)p1, (float)p2, (float)p3, 384, cur_stream());
+      (float)p1, (float)p2, (float)p3, (float)v_lr, (float)v_eps,
+      (float)v_l2, 384, cur_stream());
 }
 
 // ---- FFM ----
@@ -868,7 +873,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("gv"), py::arg("gradW"), py::arg("gradV"),
         py::arg("touched"), py::arg("chunk") = 0);
   m.def("fm_sorted_apply_fused", &fm_sorted_apply_fused,
-        "segment-reduce + fused optimizer for interior segments");
+        "segment-reduce + fused optimizer for interior segments",
+        py::arg("sorted_fids"), py::arg("perm"), py::arg("gw"),
+        py::arg("gv"), py::arg("gradW"), py::arg("gradV"),
+        py::arg("touched"), py::arg("W"), py::arg("V"), py::arg("nW"),
+        py::arg("nV"), py::arg("zW"), py::arg("zV"), py::arg("opt_mode"),
+        py::arg("p0"), py::arg("p1"), py::arg("p2"), py::arg("p3"),
+        py::arg("v_lr") = 0.05, py::arg("v_eps") = 1e-8,
+        py::arg("v_l2") = 1e-5);
   m.def("ffm_forward", &ffm_forward, "FFM pairwise forward (LDS-staged)");
   m.def("ffm_forward_pp", &ffm_forward_pp, "lane-per-pair FFM forward");
   m.def("ffm_row_emit", &ffm_row_emit,

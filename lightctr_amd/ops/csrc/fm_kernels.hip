@@ -254,6 +254,9 @@ struct FmOptArgs {
   float* nV;
   float* zV;
   float p0, p1, p2, p3;  // adagrad: lr, eps, l2 | ftrl: alpha, beta, l1, l2
+  // opt_mode 3 (FTRL on W + Adagrad on V — the default FM optimizer
+  // pairing, models/fm.py ftrl_v="adagrad"): q* are the V-side knobs
+  float q0, q1, q2;  // v_lr, v_eps, v_l2
 };
 
 template <int K>
@@ -301,6 +304,15 @@ __global__ void fm_sorted_apply_kernel(
           const float aw = oa.nW[cur_fid] + gwv * gwv;
           oa.nW[cur_fid] = aw;
           oa.W[cur_fid] -= oa.p0 * gwv * __frsqrt_rn(aw + oa.p1);
+        }
+      } else if (opt_mode == 3) {  // FTRL on W, Adagrad on V
+        const float gg = acc + oa.q2 * V[off];
+        const float a = oa.nV[off] + gg * gg;
+        oa.nV[off] = a;
+        V[off] -= oa.q0 * gg * __frsqrt_rn(a + oa.q1);
+        if (k == 0) {
+          ftrl_update(&oa.W[cur_fid], &oa.zW[cur_fid], &oa.nW[cur_fid],
+                      accw, oa.p0, oa.p1, oa.p2, oa.p3);
         }
       } else {  // ftrl
         ftrl_update(&V[off], &oa.zV[off], &oa.nV[off], acc, oa.p0, oa.p1,
@@ -738,6 +750,7 @@ void fm_sorted_apply_launch(const int* sorted_fids, const int* perm,
                             int K, int opt_mode, float* V, float* W,
                             float* nW, float* zW, float* nV, float* zV,
                             float p0, float p1, float p2, float p3,
+                            float q0, float q1, float q2,
                             int chunk, hipStream_t stream) {
   if (nnz <= 0) return;
   // chunk 0 = auto (walk kernel, chunk 384 — measured best), -1 = the
@@ -787,7 +800,7 @@ void fm_sorted_apply_launch(const int* sorted_fids, const int* perm,
   const int nwaves = (nnz + chunk - 1) / chunk;
   dim3 block(wpb * LCTR_WAVE);
   dim3 grid((nwaves + wpb - 1) / wpb);
-  FmOptArgs oa{W, nW, zW, nV, zV, p0, p1, p2, p3};
+  FmOptArgs oa{W, nW, zW, nV, zV, p0, p1, p2, p3, q0, q1, q2};
   DISPATCH_K(K, hipLaunchKernelGGL((fm_sorted_apply_kernel<KC>), grid, block,
                                    0, stream, sorted_fids, perm, gw, gv, gradW,
                                    gradV, touched, nnz, chunk, opt_mode, V,

@@ -30,6 +30,7 @@ void fm_sorted_apply_launch(const int* sorted_fids, const int* perm,
                             int K, int opt_mode, float* V, float* W,
                             float* nW, float* zW, float* nV, float* zV,
                             float p0, float p1, float p2, float p3,
+                            float q0, float q1, float q2,
                             int chunk, ihipStream_t* stream);
 void bitmap_compact_launch(unsigned long long* bitmap, int nwords,
                            int* out_fids, int* out_count, int cap,

@@ -176,9 +176,12 @@ def test_fused_apply_matches_two_phase():
     optimizers, on skewed data (long runs exercise interior fast path)."""
     from lightctr_amd.data.synthetic import SyntheticCriteo
 
-    for opt in ("adagrad", "ftrl"):
+    # ftrl_v covers pure FTRL (mode 2) and the default FTRL-W/Adagrad-V
+    # pairing (round-2 fused mode 3)
+    for opt, ftrl_v in (("adagrad", "ftrl"), ("ftrl", "ftrl"),
+                        ("ftrl", "adagrad")):
         h = FMHyper(num_features=1 << 15, k=16, optimizer=opt, seed=31,
-                    ftrl_v="ftrl")
+                    ftrl_v=ftrl_v)
         a = FMModel(h, device="cuda:0")
         a.fused_apply = True
         b = FMModel(h, device="cuda:0")
