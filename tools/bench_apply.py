@@ -90,6 +90,14 @@ def main():
     tok = bool(torch.equal(t1, m.touched))
     print(f"segscan4 vs walk gradV maxdiff {dmax:.3e} gradW {dw:.3e} touched_eq {tok}")
 
+    # full-step A/B: fused mode-3 apply vs two-phase (default ftrl pairing)
+    for fused in (False, True):
+        mm = FMModel(h, device="cuda")
+        mm.fused_apply = fused
+        tt = t(lambda: mm.train_step(row_ptr, fids, vals, labels))
+        print(f"step fused={int(fused)}  {tt:8.1f} us  "
+              f"({65536 / tt:.1f}M ex/s)")
+
 
 if __name__ == "__main__":
     main()
