@@ -26,6 +26,8 @@ def main():
     ap.add_argument("--k", type=int, default=16)
     ap.add_argument("--optimizer", default="ftrl")
     ap.add_argument("--eval-rows", type=int, default=262144)
+    ap.add_argument("--dtype", default="fp32", choices=["fp32", "bf16"],
+                    help="FFM only: bf16 compute mirror (config #3)")
     args = ap.parse_args()
 
     from lightctr_amd.data.synthetic import SyntheticCriteo
@@ -43,7 +45,8 @@ def main():
 
         model = FFMModel(FFMHyper(num_features=args.features, num_fields=39,
                                   k=min(args.k, 8),
-                                  optimizer=args.optimizer), device=dev)
+                                  optimizer=args.optimizer,
+                                  dtype=args.dtype), device=dev)
     elif args.model == "nfm":
         from lightctr_amd.models.nfm import NFMHyper, NFMModel
 
