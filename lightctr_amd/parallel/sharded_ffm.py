@@ -133,10 +133,14 @@ class ShardedFFMModel:
         if self._use_hip:
             ops = require_hip_ops()
             sorted_own, perm_own = sort_ids(lidx32, self.F_local)
-            ops.fm_sorted_apply(sorted_own, perm_own, gW_recv.contiguous(),
-                                gV_recv.contiguous(),
-                                self.gradW, self.gradV.view(self.F_local, -1),
-                                self.touched)
+            # runtime-D (nf*K) segment reduce: the FFM block-apply kernel
+            # (fm_sorted_apply only dispatches K in {4..64} and aborted
+            # on D=312 — caught by the round-2 on-GPU shakeout)
+            ops.ffm_blocks_apply(sorted_own, perm_own,
+                                 gV_recv.contiguous(),
+                                 gW_recv.contiguous(), self.gradW,
+                                 self.gradV.view(self.F_local, -1),
+                                 self.touched)
             self.count.zero_()
             ops.bitmap_compact(self.touched, self.uniq, self.count)
             live = self.uniq[: min(self.uniq.numel(), int(lidx32.numel()))]
