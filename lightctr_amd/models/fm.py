@@ -76,16 +76,14 @@ class FMModel:
         self._use_hip = self.device.type == "cuda"
         self.backward_mode = "sorted"  # "sorted" (default) | "atomic"
         # fused-apply: interior feature segments get their optimizer update
-        # inside the segment-reduce kernel. Measured NET-NEUTRAL on MI355X
-        # (optimizer RMWs serialize on the segment walk's critical path:
-        # apply 185->385us while the separate pass drops 132->37us), so the
-        # two-phase path stays the default; kept as a switchable variant.
-        # fused optimizer-in-apply: measured NET-NEGATIVE in both rounds
-        # (round 2, mode 3 ftrlW+adagradV: 721 vs 693 us/step — the
-        # boundary-fid bitmap pass still runs and the heavier flush costs
-        # more than the separate 110 us sparse apply saves;
-        # profiles/r2_06_fm_step_fused_ab.txt). Kept selectable + parity
-        # tested.
+        # inside the segment-reduce kernel. Measured NET-NEGATIVE in both
+        # rounds (round 1: optimizer RMWs serialize the segment walk,
+        # apply 185->385us vs separate pass 132->37us; round 2, mode 3
+        # ftrlW+adagradV: 721 vs 693 us/step — the boundary-fid bitmap
+        # pass still runs and the heavier flush costs more than the
+        # separate ~110 us sparse apply saves;
+        # profiles/r2_06_fm_step_fused_ab.txt). Two-phase stays default;
+        # kept selectable + parity tested.
         self.fused_apply = False
         if self._use_hip:
             require_hip_ops()  # fail loudly if extension missing on GPU

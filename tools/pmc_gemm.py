@@ -1,0 +1,45 @@
+"""Minimal single-kernel GEMM loop for rocprofv3 --pmc counter runs
+(single counters only — large sets crash the tool on this pool).
+
+Usage: [LCTR_GEMM_P8=0|1 LCTR_GEMM_P8_LITE=0|1] \
+    rocprofv3 --pmc SQ_LDS_BANK_CONFLICT -d out -- \
+    python tools/pmc_gemm.py --m 4096 --n 4096 --k 4096 --reps 20
+"""
+import argparse
+import sys
+import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
+import torch
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--m", type=int, default=4096)
+    ap.add_argument("--n", type=int, default=4096)
+    ap.add_argument("--k", type=int, default=4096)
+    ap.add_argument("--reps", type=int, default=20)
+    args = ap.parse_args()
+
+    from lightctr_amd.ops import hip_ops
+
+    M, N, K = args.m, args.n, args.k
+    g = torch.Generator().manual_seed(0)
+    A = (torch.randn(M, K, generator=g)).to(torch.bfloat16).cuda()
+    Bst = (torch.randn(N, K, generator=g)).to(torch.bfloat16).cuda()
+    for _ in range(3):
+        hip_ops.gemm_bf16(A, Bst, None, M, N, K, 0, 0, 0, False)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.reps):
+        hip_ops.gemm_bf16(A, Bst, None, M, N, K, 0, 0, 0, False)
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / args.reps
+    print(f"{M}x{N}x{K}: {dt * 1e3:.3f} ms  "
+          f"{2.0 * M * N * K / dt / 1e12:.1f} TF")
+
+
+if __name__ == "__main__":
+    main()
