@@ -104,7 +104,7 @@ __device__ __forceinline__ float p8_act(float v, int act) {
 // boundary) instead of the template's eight; the compiler pipelines the
 // phases freely as in the round-1 kernel. A/B lever for the big square
 // shapes where the 8-barrier variant measured 9% behind round 1.
-template <bool TAIL, bool LITE>
+template <bool TAIL, bool LITE, bool XCD>
 __global__ __launch_bounds__(512, 1) void gemm256p8_bf16_kernel(
     const __bf16* __restrict__ A, const __bf16* __restrict__ Bst,
     const float* __restrict__ bias, float* __restrict__ C,
@@ -115,8 +115,23 @@ __global__ __launch_bounds__(512, 1) void gemm256p8_bf16_kernel(
   const int wave = threadIdx.x >> 6;
   const int wm = wave >> 2;  // 0..1
   const int wn = wave & 3;   // 0..3
-  const int M0 = blockIdx.y * 256;
-  const int N0 = blockIdx.x * 256;
+  int bx = blockIdx.x, by = blockIdx.y;
+  if constexpr (XCD) {
+    // T1 XCD-aware remap (bijective form, guide §5.5): default blockIdx
+    // round-robins the 8 XCDs, so neighbor tiles that share A rows / B
+    // cols land on different private L2s; give each XCD a contiguous
+    // grid chunk instead
+    const int nwg = gridDim.x * gridDim.y;
+    const int wg = by * gridDim.x + bx;
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = wg % 8, idx = wg / 8;
+    const int nw = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q)
+                   + idx;
+    by = nw / gridDim.x;
+    bx = nw - by * gridDim.x;
+  }
+  const int M0 = by * 256;
+  const int N0 = bx * 256;
 
   p8f32x4 acc[8][4] = {};
 
@@ -304,25 +319,21 @@ void gemm256p8_bf16_launch(const void* A, const void* Bst, const float* bias,
     const char* e = getenv("LCTR_GEMM_P8_LITE");
     return !(e && e[0] == '0');
   }();
-  if (K % P8_BK == 0) {
-    if (lite)
-      hipLaunchKernelGGL((gemm256p8_bf16_kernel<false, true>), grid, block,
-                         0, stream, (const __bf16*)A, (const __bf16*)Bst,
-                         bias, C, (__bf16*)Cbf, M, N, K, act);
-    else
-      hipLaunchKernelGGL((gemm256p8_bf16_kernel<false, false>), grid, block,
-                         0, stream, (const __bf16*)A, (const __bf16*)Bst,
-                         bias, C, (__bf16*)Cbf, M, N, K, act);
-  } else {
-    if (lite)
-      hipLaunchKernelGGL((gemm256p8_bf16_kernel<true, true>), grid, block, 0,
-                         stream, (const __bf16*)A, (const __bf16*)Bst, bias,
-                         C, (__bf16*)Cbf, M, N, K, act);
-    else
-      hipLaunchKernelGGL((gemm256p8_bf16_kernel<true, false>), grid, block,
-                         0, stream, (const __bf16*)A, (const __bf16*)Bst,
-                         bias, C, (__bf16*)Cbf, M, N, K, act);
-  }
+  static const bool xcd = [] {
+    const char* e = getenv("LCTR_GEMM_P8_XCD");
+    return e && e[0] == '1';
+  }();
+  const bool tail = (K % P8_BK) != 0;
+#define P8_LAUNCH(T, L, X)                                                    hipLaunchKernelGGL((gemm256p8_bf16_kernel<T, L, X>), grid, block, 0,                           stream, (const __bf16*)A, (const __bf16*)Bst, bias, C,                      (__bf16*)Cbf, M, N, K, act)
+  if (!tail && lite && !xcd) P8_LAUNCH(false, true, false);
+  else if (!tail && lite && xcd) P8_LAUNCH(false, true, true);
+  else if (!tail && !lite && !xcd) P8_LAUNCH(false, false, false);
+  else if (!tail && !lite && xcd) P8_LAUNCH(false, false, true);
+  else if (tail && lite && !xcd) P8_LAUNCH(true, true, false);
+  else if (tail && lite && xcd) P8_LAUNCH(true, true, true);
+  else if (tail && !lite && !xcd) P8_LAUNCH(true, false, false);
+  else P8_LAUNCH(true, false, true);
+#undef P8_LAUNCH
 }
 
 }  // namespace lightctr
