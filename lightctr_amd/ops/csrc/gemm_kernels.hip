@@ -279,7 +279,13 @@ void gemm_bf16_launch(const void* A, const void* Bst, const float* bias,
   const int xy = (int)(grid.x * grid.y);
   if (xy < 64 && K >= 4096 && Cbf == nullptr && bias == nullptr &&
       act == 0) {
-    int zw = min(64, max(2, 512 / xy));
+    // split-K fan-out target (blocks in flight); LCTR_SPLITK_TARGET
+    // overrides for sweeps
+    static const int sk_target = [] {
+      const char* e = getenv("LCTR_SPLITK_TARGET");
+      return e ? atoi(e) : 512;
+    }();
+    int zw = min(64, max(2, sk_target / xy));
     kchunk = ((K + zw - 1) / zw + GEMM_BK - 1) / GEMM_BK * GEMM_BK;
     grid.z = (K + kchunk - 1) / kchunk;
     LCTR_CHECK_HIP(hipMemsetAsync(C, 0, (size_t)M * N * sizeof(float),

@@ -21,23 +21,28 @@ def main():
     ap.add_argument("--n", type=int, default=4096)
     ap.add_argument("--k", type=int, default=4096)
     ap.add_argument("--reps", type=int, default=20)
+    ap.add_argument("--transA", type=int, default=0)
+    ap.add_argument("--transB", type=int, default=0)
     args = ap.parse_args()
 
     from lightctr_amd.ops import hip_ops
 
     M, N, K = args.m, args.n, args.k
+    tA, tB = args.transA, args.transB
     g = torch.Generator().manual_seed(0)
-    A = (torch.randn(M, K, generator=g)).to(torch.bfloat16).cuda()
-    Bst = (torch.randn(N, K, generator=g)).to(torch.bfloat16).cuda()
+    A = (torch.randn(K, M) if tA else torch.randn(M, K)) \
+        .to(torch.bfloat16).cuda()
+    Bst = (torch.randn(K, N) if tB else torch.randn(N, K)) \
+        .to(torch.bfloat16).cuda()
     for _ in range(3):
-        hip_ops.gemm_bf16(A, Bst, None, M, N, K, 0, 0, 0, False)
+        hip_ops.gemm_bf16(A, Bst, None, M, N, K, tA, tB, 0, False)
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(args.reps):
-        hip_ops.gemm_bf16(A, Bst, None, M, N, K, 0, 0, 0, False)
+        hip_ops.gemm_bf16(A, Bst, None, M, N, K, tA, tB, 0, False)
     torch.cuda.synchronize()
     dt = (time.perf_counter() - t0) / args.reps
-    print(f"{M}x{N}x{K}: {dt * 1e3:.3f} ms  "
+    print(f"{M}x{N}x{K} tA={tA} tB={tB}: {dt * 1e3:.3f} ms  "
           f"{2.0 * M * N * K / dt / 1e12:.1f} TF")
 
 
