@@ -13,6 +13,7 @@
 //   * backward scatters symmetric grads with fp32 atomics + touched bitmap
 //     (optimizer = generic sparse fused kernels in misc_kernels.hip).
 #include <hip/hip_fp16.h>
+#include <cstdlib>
 
 #include "common.h"
 
@@ -793,7 +794,10 @@ void ffm_blocks_apply_f16_launch(const int* sorted_fids, const int* perm,
                                  unsigned long long* touched, int D, int nnz,
                                  float inv_scale, hipStream_t stream) {
   if (nnz <= 0) return;
-  const int chunk = 96;
+  static const int chunk = [] {
+    const char* e = getenv("LCTR_FFM_APPLY_CHUNK");
+    return e ? atoi(e) : 96;
+  }();
   const int wpb = 4;
   const int nwaves = (nnz + chunk - 1) / chunk;
   dim3 block(wpb * LCTR_WAVE);
