@@ -267,6 +267,21 @@ void gemm_bf16_launch(const void* A, const void* Bst, const float* bias,
     small_wgrad_launch(A, Bst, C, M, N, K, stream);
     return;
   }
+  if (gemm_wgrad_eligible(M, N, K, transA, transB) && bias == nullptr &&
+      act == 0 && Cbf == nullptr) {
+    // K-major x K-major wgrad: blocked glds staging + ds_read_b64_tr_b16
+    // fragment reads (the generic path's scalar transposed staging
+    // measured 125 us on the 256x624x65536 W&D shape — 9x off floor).
+    // LCTR_WGRAD128=0 reverts to the generic kernel.
+    static const bool wg = [] {
+      const char* e = getenv("LCTR_WGRAD128");
+      return !(e && e[0] == '0');
+    }();
+    if (wg) {
+      gemm_wgrad_bf16_launch(A, Bst, C, M, N, K, stream);
+      return;
+    }
+  }
   if (K == 1 && transA == 0) {
     outer_k1_launch(A, Bst, bias, C, Cbf, M, N, act, stream);
     return;

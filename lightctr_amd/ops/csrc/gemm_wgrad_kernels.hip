@@ -1,0 +1,204 @@
+// Weight-gradient GEMM for MI355X/gfx950: C[M,N] = sum_k At[k,m]*Bt[k,n]
+// with BOTH operands stored K-major (At = [K,M] row-major, Bt = [K,N] —
+// the dZ / activation layouts of the MLP backward, reference
+// fullyconnLayer.h:166-179 weight-delta accumulation).
+//
+// The generic kernel's TRANSA/TRANSB path stages these with per-element
+// scalar LDS writes (measured 125 us for 256x624x65536 — 9x off the
+// ~14 us traffic floor: VALU/LDS-write bound, guide common-mistake #1).
+// This kernel instead:
+//   * stages [64k x 128row] panels with global_load_lds into a BLOCKED
+//     LDS image [k/8][row/16][8][16] (each 16 B glds write is one
+//     contiguous 8-element half-row of a 8x16 block; per-lane source
+//     addresses make the blocking free)
+//   * reads MFMA fragments with the gfx950 hardware transpose read
+//     ds_read_b64_tr_b16 (lane l, elem j <- lds[addr_l + j*16]): two
+//     tr reads yield the 8 K-major values of the lane's fragment column
+//     — the "attention-V recipe" from the CDNA4 guide
+//   * split-K over gridDim.z with fp32 atomic accumulation into C
+//     (pre-zeroed by the launcher), as in the generic kernel.
+// Eligibility (launcher): transA && transB, M%16==0, N%16==0, M%8==0
+// row strides (always true given %16), K >= 1024. Partial K-tails are
+// staged by a scalar blocked path.
+#include <cstdlib>
+
+#include "common.h"
+
+namespace lightctr {
+
+typedef __attribute__((ext_vector_type(8))) __bf16 wg_bf16x8;
+typedef __attribute__((ext_vector_type(4))) __bf16 wg_bf16x4;
+typedef __attribute__((ext_vector_type(4))) float wg_f32x4;
+
+#define WG_BM 128
+#define WG_BN 128
+#define WG_BK 64
+
+// Stage a [BK x 128] K-major panel (g = base of [K, ld] array, columns
+// col0..col0+127, k rows k0..k0+BK) into the blocked LDS image
+// [BK/8][128/16][8][16] with glds. 256 threads, 4 glds per thread.
+// Caller guarantees full panel in range and 16 B alignment.
+__device__ __forceinline__ void wg_stage_glds(const __bf16* __restrict__ g,
+                                              long ld, int k0, int col0,
+                                              __bf16* dst) {
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+#pragma unroll
+  for (int it = 0; it < 4; ++it) {
+    const int instr = wave * 4 + it;        // 16 glds instructions total
+    const int gb = instr * 4 + (lane >> 4);  // global block 0..63
+    const int kb = gb >> 3;                  // k-block 0..7
+    const int mb = gb & 7;                   // row-block 0..7
+    const int kk = (lane & 15) >> 1;
+    const int mmh = (lane & 1) * 8;
+    const __bf16* src =
+        g + (long)(k0 + kb * 8 + kk) * ld + col0 + mb * 16 + mmh;
+    // lane-linear dest covers exactly (block gb, kk, half mmh)
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)src,
+        (__attribute__((address_space(3))) void*)(dst + instr * 512), 16, 0,
+        0);
+  }
+}
+
+// Scalar blocked staging for K-tails / range edges (zero-padded).
+__device__ __forceinline__ void wg_stage_scalar(const __bf16* __restrict__ g,
+                                                long ld, int k0, int col0,
+                                                int K, int ncols,
+                                                __bf16* dst) {
+  const int tid = threadIdx.x;
+  // element e = (block gb, kk, mm): 8192 elements / 256 threads = 32 each
+  for (int e = tid; e < 8192; e += 256) {
+    const int gb = e >> 7;
+    const int within = e & 127;
+    const int kk = within >> 4;
+    const int mm = within & 15;
+    const int kb = gb >> 3;
+    const int mb = gb & 7;
+    const int k = k0 + kb * 8 + kk;
+    const int c = col0 + mb * 16 + mm;
+    __bf16 v = (__bf16)0.f;
+    if (k < K && c < ncols) v = g[(long)k * ld + c];
+    dst[e] = v;
+  }
+}
+
+__device__ __forceinline__ wg_bf16x8 wg_frag(const __bf16* base, int fb,
+                                             int kc, int lane) {
+  // fragment (row-block fb, k-chunk kc): lane l needs column (l&15) of
+  // k = kc*32 + (l>>4)*8 + 0..7  ->  block kb = kc*4 + (l>>4), rows kk
+  // 0..7 of row-block fb. Two tr reads (kk 0..3, 4..7).
+  const int kb = kc * 4 + (lane >> 4);
+  const __bf16* p = base + ((kb * 8 + fb) << 7) + (lane & 15);
+  const wg_bf16x4 lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (__attribute__((address_space(3))) wg_bf16x4*)p);
+  const wg_bf16x4 hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (__attribute__((address_space(3))) wg_bf16x4*)(p + 64));
+  wg_bf16x8 out;
+  out[0] = lo.x; out[1] = lo.y; out[2] = lo.z; out[3] = lo.w;
+  out[4] = hi.x; out[5] = hi.y; out[6] = hi.z; out[7] = hi.w;
+  return out;
+}
+
+__global__ __launch_bounds__(256) void gemm_wgrad_bf16_kernel(
+    const __bf16* __restrict__ At, const __bf16* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K, int kchunk) {
+  __shared__ __bf16 smem[2 * WG_BK * WG_BM];  // As | Bs (16 KB each)
+  __bf16* As = smem;
+  __bf16* Bs = smem + WG_BK * WG_BM;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wm = wave >> 1, wn = wave & 1;  // 2x2 waves, 64x64 each
+  const int M0 = blockIdx.y * WG_BM;
+  const int N0 = blockIdx.x * WG_BN;
+  const bool a_full = M0 + WG_BM <= M;
+  const bool b_full = N0 + WG_BN <= N;
+
+  wg_f32x4 acc[4][4] = {};
+
+  const int kbeg = (kchunk > 0) ? blockIdx.z * kchunk : 0;
+  const int kend = (kchunk > 0) ? min(K, kbeg + kchunk) : K;
+  for (int k0 = kbeg; k0 < kend; k0 += WG_BK) {
+    const bool k_full = k0 + WG_BK <= kend;
+    if (a_full && k_full)
+      wg_stage_glds(At, M, k0, M0, As);
+    else
+      wg_stage_scalar(At, M, k0, M0, min(kend, K), M, As);
+    if (b_full && k_full)
+      wg_stage_glds(Bt, N, k0, N0, Bs);
+    else
+      wg_stage_scalar(Bt, N, k0, N0, min(kend, K), N, Bs);
+    __syncthreads();
+
+#pragma unroll
+    for (int kc = 0; kc < 2; ++kc) {
+      wg_bf16x8 a[4], b[4];
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        a[f] = wg_frag(As, wm * 4 + f, kc, lane);
+        b[f] = wg_frag(Bs, wn * 4 + f, kc, lane);
+      }
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < 4; ++fn)
+          acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[fm], b[fn], acc[fm][fn], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // C write (atomic when split-K): D map col = lane&15, row = (l>>4)*4+r
+#pragma unroll
+  for (int fm = 0; fm < 4; ++fm) {
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn) {
+      const int col = N0 + wn * 64 + fn * 16 + (lane & 15);
+      if (col >= N) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = M0 + wm * 64 + fm * 16 + (lane >> 4) * 4 + r;
+        if (row >= M) continue;
+        if (kchunk > 0)
+          atomicAdd(&C[(size_t)row * N + col], acc[fm][fn][r]);
+        else
+          C[(size_t)row * N + col] = acc[fm][fn][r];
+      }
+    }
+  }
+}
+
+bool gemm_wgrad_eligible(int M, int N, int K, int transA, int transB) {
+  // both operands K-major; blocked glds staging wants 16-col row blocks
+  // and 16 B-aligned 8-element half-rows (M%8, N%8; %16 keeps the edge
+  // logic out of the fragment path)
+  return transA == 1 && transB == 1 && M % 16 == 0 && N % 16 == 0 &&
+         K >= 1024;
+}
+
+void gemm_wgrad_bf16_launch(const void* At, const void* Bt, float* C, int M,
+                            int N, int K, hipStream_t stream) {
+  dim3 block(256);
+  dim3 grid((N + WG_BN - 1) / WG_BN, (M + WG_BM - 1) / WG_BM);
+  int kchunk = 0;
+  const int xy = (int)(grid.x * grid.y);
+  static const int sk_target = [] {
+    const char* e = getenv("LCTR_SPLITK_TARGET");
+    return e ? atoi(e) : 512;
+  }();
+  if (xy < 256 && K >= 2048) {
+    int zw = min(64, max(1, sk_target / xy));
+    if (zw > 1) {
+      kchunk = ((K + zw - 1) / zw + WG_BK - 1) / WG_BK * WG_BK;
+      grid.z = (K + kchunk - 1) / kchunk;
+      LCTR_CHECK_HIP(
+          hipMemsetAsync(C, 0, (size_t)M * N * sizeof(float), stream));
+    }
+  }
+  hipLaunchKernelGGL(gemm_wgrad_bf16_kernel, grid, block, 0, stream,
+                     (const __bf16*)At, (const __bf16*)Bt, C, M, N, K,
+                     kchunk);
+}
+
+}  // namespace lightctr

@@ -123,6 +123,11 @@ void fm_ftrl_apply_launch(const int* uniq, const int* count, float* W,
                           int v_adagrad, float v_lr, float v_eps, float v_l2,
                           ihipStream_t* stream);
 
+// --- gemm_wgrad_kernels.hip ---
+bool gemm_wgrad_eligible(int M, int N, int K, int transA, int transB);
+void gemm_wgrad_bf16_launch(const void* At, const void* Bt, float* C, int M,
+                            int N, int K, ihipStream_t* stream);
+
 // --- gemm_kernels.hip ---
 void gemm_bf16_launch(const void* A, const void* Bst, const float* bias,
                       float* C, void* Cbf, int M, int N, int K, int transA,

@@ -254,6 +254,32 @@ def test_gemm_v2_race_screen():
 
 
 @pytest.mark.gpu
+def test_gemm_wgrad128_parity_and_race():
+    """K-major x K-major wgrad kernel (blocked glds + ds_read_b64_tr_b16
+    fragment reads): parity vs fp32 reference across full/partial tiles
+    and K-tails, plus a 10-run screen of the split-K atomics."""
+    from lightctr_amd.ops import hip_ops
+
+    for (M, N, K) in [(256, 624, 4096), (128, 256, 2048), (256, 256, 1056),
+                      (144, 112, 1024)]:
+        g = torch.Generator().manual_seed(M + K)
+        At = (torch.randn(K, M, generator=g) * 0.5).to(torch.bfloat16).cuda()
+        Bt = (torch.randn(K, N, generator=g) * 0.5).to(torch.bfloat16).cuda()
+        C = hip_ops.gemm_bf16(At, Bt, None, M, N, K, 1, 1, 0, False)
+        ref = At.float().t() @ Bt.float()
+        assert torch.allclose(C, ref, atol=2e-2 * K ** 0.5, rtol=1e-2), \
+            (M, N, K, (C - ref).abs().max())
+    M, N, K = 256, 624, 8192
+    g = torch.Generator().manual_seed(3)
+    At = (torch.randn(K, M, generator=g) * 0.5).to(torch.bfloat16).cuda()
+    Bt = (torch.randn(K, N, generator=g) * 0.5).to(torch.bfloat16).cuda()
+    ref = At.float().t() @ Bt.float()
+    for it in range(10):
+        C = hip_ops.gemm_bf16(At, Bt, None, M, N, K, 1, 1, 0, False)
+        assert torch.allclose(C, ref, atol=2e-2 * K ** 0.5, rtol=1e-2), it
+
+
+@pytest.mark.gpu
 def test_gemm_p8_race_screen():
     """The 8-phase counted-vmcnt schedule is a new sync structure (guide
     two-lane discipline): repeated runs at a multi-tile shape and a
