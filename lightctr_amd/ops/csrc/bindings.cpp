@@ -405,6 +405,21 @@ void sparse_ftrl_apply(at::Tensor uniq, at::Tensor count, at::Tensor W,
       cur_stream());
 }
 
+at::Tensor gemm_wgrad_bf16(at::Tensor At, at::Tensor Bt, int64_t M,
+                           int64_t N, int64_t K) {
+  CHK(At.is_cuda() && At.scalar_type() == at::kBFloat16 &&
+          At.is_contiguous(), "At must be cuda bf16 [K,M]");
+  CHK(Bt.is_cuda() && Bt.scalar_type() == at::kBFloat16 &&
+          Bt.is_contiguous(), "Bt must be cuda bf16 [K,N]");
+  CHK(lightctr::gemm_wgrad_eligible((int)M, (int)N, (int)K, 1, 1),
+      "shape not eligible for the wgrad kernel");
+  auto C = at::empty({M, N}, At.options().dtype(at::kFloat));
+  lightctr::gemm_wgrad_bf16_launch(At.data_ptr(), Bt.data_ptr(),
+                                   C.data_ptr<float>(), (int)M, (int)N,
+                                   (int)K, cur_stream());
+  return C;
+}
+
 at::Tensor im2col_bf16(at::Tensor x, int64_t k, int64_t stride,
                        int64_t pad) {
   check_cuda_f32(x, "x");
@@ -948,6 +963,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("nfm_backward_emit", &nfm_backward_emit,
         "NFM per-entry grads for sorted apply");
   m.def("bitmap_compact", &bitmap_compact, "touched bitmap -> fid list");
+  m.def("gemm_wgrad_bf16", &gemm_wgrad_bf16,
+        "K-major x K-major wgrad GEMM (tr16 fragment reads)");
   m.def("im2col_bf16", &im2col_bf16,
         "fused unfold -> bf16 GEMM operand [B*L, C*k*k]");
   m.def("col2im", &col2im, "conv data-grad fold (gather form)");

@@ -272,10 +272,11 @@ void gemm_bf16_launch(const void* A, const void* Bst, const float* bias,
     // K-major x K-major wgrad: blocked glds staging + ds_read_b64_tr_b16
     // fragment reads (the generic path's scalar transposed staging
     // measured 125 us on the 256x624x65536 W&D shape — 9x off floor).
-    // LCTR_WGRAD128=0 reverts to the generic kernel.
+    // OPT-IN while under debug (LCTR_WGRAD128=1): first cut measured a
+    // parity failure + a partial-tile slowdown.
     static const bool wg = [] {
       const char* e = getenv("LCTR_WGRAD128");
-      return !(e && e[0] == '0');
+      return e && e[0] == '1';
     }();
     if (wg) {
       gemm_wgrad_bf16_launch(A, Bst, C, M, N, K, stream);

@@ -265,7 +265,7 @@ def test_gemm_wgrad128_parity_and_race():
         g = torch.Generator().manual_seed(M + K)
         At = (torch.randn(K, M, generator=g) * 0.5).to(torch.bfloat16).cuda()
         Bt = (torch.randn(K, N, generator=g) * 0.5).to(torch.bfloat16).cuda()
-        C = hip_ops.gemm_bf16(At, Bt, None, M, N, K, 1, 1, 0, False)
+        C = hip_ops.gemm_wgrad_bf16(At, Bt, M, N, K)
         ref = At.float().t() @ Bt.float()
         assert torch.allclose(C, ref, atol=2e-2 * K ** 0.5, rtol=1e-2), \
             (M, N, K, (C - ref).abs().max())
@@ -275,7 +275,7 @@ def test_gemm_wgrad128_parity_and_race():
     Bt = (torch.randn(K, N, generator=g) * 0.5).to(torch.bfloat16).cuda()
     ref = At.float().t() @ Bt.float()
     for it in range(10):
-        C = hip_ops.gemm_bf16(At, Bt, None, M, N, K, 1, 1, 0, False)
+        C = hip_ops.gemm_wgrad_bf16(At, Bt, M, N, K)
         assert torch.allclose(C, ref, atol=2e-2 * K ** 0.5, rtol=1e-2), it
 
 

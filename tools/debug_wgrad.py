@@ -13,7 +13,7 @@ def check(M, N, K):
     g = torch.Generator().manual_seed(M + K)
     At = (torch.randn(K, M, generator=g) * 0.5).to(torch.bfloat16).cuda()
     Bt = (torch.randn(K, N, generator=g) * 0.5).to(torch.bfloat16).cuda()
-    C = hip_ops.gemm_bf16(At, Bt, None, M, N, K, 1, 1, 0, False)
+    C = hip_ops.gemm_wgrad_bf16(At, Bt, M, N, K)
     ref = At.float().t() @ Bt.float()
     d = (C - ref).abs()
     tol = 2e-2 * K ** 0.5
