@@ -420,6 +420,16 @@ at::Tensor gemm_wgrad_bf16(at::Tensor At, at::Tensor Bt, int64_t M,
   return C;
 }
 
+std::vector<at::Tensor> wg_probe(at::Tensor g, int64_t ld, int64_t mode) {
+  CHK(g.is_cuda() && g.scalar_type() == at::kBFloat16, "g bf16");
+  auto out_lds = at::zeros({64 * 128}, g.options());
+  auto out_frag = at::zeros({2 * 2 * 64 * 8}, g.options().dtype(at::kFloat));
+  lightctr::wg_probe_launch(g.data_ptr(), (int)ld, out_lds.data_ptr(),
+                            out_frag.data_ptr<float>(), (int)mode,
+                            cur_stream());
+  return {out_lds, out_frag};
+}
+
 at::Tensor im2col_bf16(at::Tensor x, int64_t k, int64_t stride,
                        int64_t pad) {
   check_cuda_f32(x, "x");
@@ -963,6 +973,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("nfm_backward_emit", &nfm_backward_emit,
         "NFM per-entry grads for sorted apply");
   m.def("bitmap_compact", &bitmap_compact, "touched bitmap -> fid list");
+  m.def("wg_probe", &wg_probe, "wgrad staging/tr-read debug probe");
   m.def("gemm_wgrad_bf16", &gemm_wgrad_bf16,
         "K-major x K-major wgrad GEMM (tr16 fragment reads)");
   m.def("im2col_bf16", &im2col_bf16,

@@ -169,6 +169,40 @@ __global__ __launch_bounds__(256) void gemm_wgrad_bf16_kernel(
   }
 }
 
+// Debug probe: mode 0 = stage a [64][128] panel from g via wg_stage_glds
+// and dump the LDS image linearly; mode 1 = copy g directly into LDS
+// (identity image) — both then dump wave-0 fragments read via wg_frag.
+__global__ void wg_probe_kernel(const __bf16* __restrict__ g, int ld,
+                                __bf16* __restrict__ out_lds,
+                                float* __restrict__ out_frag, int mode) {
+  __shared__ __bf16 s[WG_BK * WG_BM];
+  if (mode == 0) {
+    wg_stage_glds(g, ld, 0, 0, s);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+  } else {
+    for (int e = threadIdx.x; e < WG_BK * WG_BM; e += 256) s[e] = g[e];
+    __syncthreads();
+  }
+  for (int e = threadIdx.x; e < WG_BK * WG_BM; e += 256) out_lds[e] = s[e];
+  const int lane = threadIdx.x & 63;
+  if (threadIdx.x < 64) {
+    for (int fb = 0; fb < 2; ++fb)
+      for (int kc = 0; kc < 2; ++kc) {
+        const wg_bf16x8 f = wg_frag(s, fb, kc, lane);
+        for (int e = 0; e < 8; ++e)
+          out_frag[(((size_t)fb * 2 + kc) * 64 + lane) * 8 + e] =
+              (float)f[e];
+      }
+  }
+}
+
+void wg_probe_launch(const void* g, int ld, void* out_lds, float* out_frag,
+                     int mode, hipStream_t stream) {
+  hipLaunchKernelGGL(wg_probe_kernel, dim3(1), dim3(256), 0, stream,
+                     (const __bf16*)g, ld, (__bf16*)out_lds, out_frag, mode);
+}
+
 bool gemm_wgrad_eligible(int M, int N, int K, int transA, int transB) {
   // both operands K-major; blocked glds staging wants 16-col row blocks
   // and 16 B-aligned 8-element half-rows (M%8, N%8; %16 keeps the edge

@@ -127,6 +127,8 @@ void fm_ftrl_apply_launch(const int* uniq, const int* count, float* W,
 bool gemm_wgrad_eligible(int M, int N, int K, int transA, int transB);
 void gemm_wgrad_bf16_launch(const void* At, const void* Bt, float* C, int M,
                             int N, int K, ihipStream_t* stream);
+void wg_probe_launch(const void* g, int ld, void* out_lds, float* out_frag,
+                     int mode, ihipStream_t* stream);
 
 // --- gemm_kernels.hip ---
 void gemm_bf16_launch(const void* A, const void* Bst, const float* bias,
