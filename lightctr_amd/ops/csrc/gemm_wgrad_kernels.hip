@@ -7,13 +7,13 @@
 // scalar LDS writes (measured 125 us for 256x624x65536 — 9x off the
 // ~14 us traffic floor: VALU/LDS-write bound, guide common-mistake #1).
 // This kernel instead:
-//   * stages [64k x 128row] panels with global_load_lds into a BLOCKED
-//     LDS image [k/8][row/16][8][16] (each 16 B glds write is one
-//     contiguous 8-element half-row of a 8x16 block; per-lane source
-//     addresses make the blocking free)
-//   * reads MFMA fragments with the gfx950 hardware transpose read
-//     ds_read_b64_tr_b16 (lane l, elem j <- lds[addr_l + j*16]): two
-//     tr reads yield the 8 K-major values of the lane's fragment column
+//   * stages [64k x 128row] panels with 16 B vector loads packed into a
+//     [k/8][row/4][8][4] blocked LDS image via two ds_write_b64 per
+//     load (probe-verified: ds_read_b64_tr_b16 reads elements
+//     {p, p+4, p+8, p+12} — stride FOUR 16-bit elements, which no
+//     glds-fillable image can serve, tools/debug_wgrad_probe.py)
+//   * reads MFMA fragments with the hardware transpose read: two tr
+//     reads (block offsets 0 and +16) yield the lane's 8 K-major values
 //     — the "attention-V recipe" from the CDNA4 guide
 //   * split-K over gridDim.z with fp32 atomic accumulation into C
 //     (pre-zeroed by the launcher), as in the generic kernel.
@@ -34,31 +34,35 @@ typedef __attribute__((ext_vector_type(4))) float wg_f32x4;
 #define WG_BN 128
 #define WG_BK 64
 
+// Blocked image: element (k, m) of the [BK x 128] panel lives at
+//   ((k/8)*32 + m/4)*32 + (k%8)*4 + (m%4)
+// i.e. [k/8][m/4] blocks of [8 k][4 m] — the k stride inside a block is
+// exactly the tr read's element stride of 4.
+__device__ __forceinline__ int wg_blk(int k, int m) {
+  return (((k >> 3) << 5) + (m >> 2)) * 32 + ((k & 7) << 2) + (m & 3);
+}
+
 // Stage a [BK x 128] K-major panel (g = base of [K, ld] array, columns
-// col0..col0+127, k rows k0..k0+BK) into the blocked LDS image
-// [BK/8][128/16][8][16] with glds. 256 threads, 4 glds per thread.
-// Caller guarantees full panel in range and 16 B alignment.
-__device__ __forceinline__ void wg_stage_glds(const __bf16* __restrict__ g,
-                                              long ld, int k0, int col0,
-                                              __bf16* dst) {
+// col0..col0+127, k rows k0..k0+BK) into the blocked image: 16 B vector
+// loads (8 contiguous row elements), each split into two 8 B
+// ds_write_b64 (the two 4-wide m-blocks it spans). 256 threads, 4
+// load+write pairs each. Caller guarantees range + 16 B alignment.
+__device__ __forceinline__ void wg_stage_vec(const __bf16* __restrict__ g,
+                                             long ld, int k0, int col0,
+                                             __bf16* dst) {
   const int tid = threadIdx.x;
-  const int lane = tid & 63;
-  const int wave = tid >> 6;
 #pragma unroll
   for (int it = 0; it < 4; ++it) {
-    const int instr = wave * 4 + it;        // 16 glds instructions total
-    const int gb = instr * 4 + (lane >> 4);  // global block 0..63
-    const int kb = gb >> 3;                  // k-block 0..7
-    const int mb = gb & 7;                   // row-block 0..7
-    const int kk = (lane & 15) >> 1;
-    const int mmh = (lane & 1) * 8;
-    const __bf16* src =
-        g + (long)(k0 + kb * 8 + kk) * ld + col0 + mb * 16 + mmh;
-    // lane-linear dest covers exactly (block gb, kk, half mmh)
-    __builtin_amdgcn_global_load_lds(
-        (const __attribute__((address_space(1))) void*)src,
-        (__attribute__((address_space(3))) void*)(dst + instr * 512), 16, 0,
-        0);
+    const int t = tid + it * 256;   // 1024 load slots
+    const int k = t >> 4;           // 0..63
+    const int m0 = (t & 15) * 8;    // 0..120
+    const wg_bf16x8 v =
+        *(const wg_bf16x8*)(g + (long)(k0 + k) * ld + col0 + m0);
+    wg_bf16x4 lo, hi;
+    lo.x = v[0]; lo.y = v[1]; lo.z = v[2]; lo.w = v[3];
+    hi.x = v[4]; hi.y = v[5]; hi.z = v[6]; hi.w = v[7];
+    *(wg_bf16x4*)(dst + wg_blk(k, m0)) = lo;
+    *(wg_bf16x4*)(dst + wg_blk(k, m0 + 4)) = hi;
   }
 }
 
@@ -68,33 +72,31 @@ __device__ __forceinline__ void wg_stage_scalar(const __bf16* __restrict__ g,
                                                 int K, int ncols,
                                                 __bf16* dst) {
   const int tid = threadIdx.x;
-  // element e = (block gb, kk, mm): 8192 elements / 256 threads = 32 each
   for (int e = tid; e < 8192; e += 256) {
-    const int gb = e >> 7;
-    const int within = e & 127;
-    const int kk = within >> 4;
-    const int mm = within & 15;
-    const int kb = gb >> 3;
-    const int mb = gb & 7;
-    const int k = k0 + kb * 8 + kk;
-    const int c = col0 + mb * 16 + mm;
+    // e enumerated as (k, m) in panel order
+    const int k = e >> 7;
+    const int m = e & 127;
+    const int gk = k0 + k;
+    const int c = col0 + m;
     __bf16 v = (__bf16)0.f;
-    if (k < K && c < ncols) v = g[(long)k * ld + c];
-    dst[e] = v;
+    if (gk < K && c < ncols) v = g[(long)gk * ld + c];
+    dst[wg_blk(k, m)] = v;
   }
 }
 
 __device__ __forceinline__ wg_bf16x8 wg_frag(const __bf16* base, int fb,
                                              int kc, int lane) {
-  // fragment (row-block fb, k-chunk kc): lane l needs column (l&15) of
-  // k = kc*32 + (l>>4)*8 + 0..7  ->  block kb = kc*4 + (l>>4), rows kk
-  // 0..7 of row-block fb. Two tr reads (kk 0..3, 4..7).
+  // fragment (16-row block fb, k-chunk kc): lane l needs column
+  // m = fb*16 + (l&15) of k = kc*32 + (l>>4)*8 + 0..7. tr reads
+  // {p, p+4, p+8, p+12}: p at (kb, m) block base + m%4 gives kk 0..3,
+  // +16 gives kk 4..7.
   const int kb = kc * 4 + (lane >> 4);
-  const __bf16* p = base + ((kb * 8 + fb) << 7) + (lane & 15);
+  const int m = fb * 16 + (lane & 15);
+  const __bf16* p = base + ((kb << 5) + (m >> 2)) * 32 + (m & 3);
   const wg_bf16x4 lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
       (__attribute__((address_space(3))) wg_bf16x4*)p);
   const wg_bf16x4 hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-      (__attribute__((address_space(3))) wg_bf16x4*)(p + 64));
+      (__attribute__((address_space(3))) wg_bf16x4*)(p + 16));
   wg_bf16x8 out;
   out[0] = lo.x; out[1] = lo.y; out[2] = lo.z; out[3] = lo.w;
   out[4] = hi.x; out[5] = hi.y; out[6] = hi.z; out[7] = hi.w;
@@ -122,11 +124,11 @@ __global__ __launch_bounds__(256) void gemm_wgrad_bf16_kernel(
   for (int k0 = kbeg; k0 < kend; k0 += WG_BK) {
     const bool k_full = k0 + WG_BK <= kend;
     if (a_full && k_full)
-      wg_stage_glds(At, M, k0, M0, As);
+      wg_stage_vec(At, M, k0, M0, As);
     else
       wg_stage_scalar(At, M, k0, M0, min(kend, K), M, As);
     if (b_full && k_full)
-      wg_stage_glds(Bt, N, k0, N0, Bs);
+      wg_stage_vec(Bt, N, k0, N0, Bs);
     else
       wg_stage_scalar(Bt, N, k0, N0, min(kend, K), N, Bs);
     __syncthreads();
@@ -177,8 +179,7 @@ __global__ void wg_probe_kernel(const __bf16* __restrict__ g, int ld,
                                 float* __restrict__ out_frag, int mode) {
   __shared__ __bf16 s[WG_BK * WG_BM];
   if (mode == 0) {
-    wg_stage_glds(g, ld, 0, 0, s);
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    wg_stage_vec(g, ld, 0, 0, s);
     __syncthreads();
   } else {
     for (int e = threadIdx.x; e < WG_BK * WG_BM; e += 256) s[e] = g[e];

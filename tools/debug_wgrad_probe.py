@@ -14,8 +14,8 @@ BK, BM = 64, 128
 
 def blocked_index(k, m):
     kb, kk = k // 8, k % 8
-    mb, mm = m // 16, m % 16
-    return ((kb * 8 + mb) * 128) + kk * 16 + mm
+    mb, mm = m // 4, m % 4
+    return (kb * 32 + mb) * 32 + kk * 4 + mm
 
 
 def main():
@@ -75,12 +75,12 @@ def main():
             for kc in range(2):
                 for l in range(64):
                     for e in range(8):
-                        # expected per MY assumed tr semantics:
-                        # base p = ((kb*8+fb)*128) + (l&15), kb=kc*4+(l>>4)
-                        # elems p + (e%4)*16 + (e//4)*64
+                        # tr semantics (probe-verified): elems
+                        # {p, p+4, p+8, p+12}; second read at p+16
                         kb = kc * 4 + (l >> 4)
-                        p = (kb * 8 + fb) * 128 + (l & 15)
-                        idx = p + (e % 4) * 16 + (e // 4) * 64
+                        m = fb * 16 + (l & 15)
+                        p = (kb * 32 + m // 4) * 32 + (m % 4)
+                        idx = p + (e % 4) * 4 + (e // 4) * 16
                         want = float(ident[idx] if idx < BK * BM else -1)
                         got = float(fr2[fb, kc, l, e])
                         if got != want:
