@@ -79,7 +79,7 @@ class GBMModel:
         D = X.shape[1]
         out = torch.empty(X.shape, dtype=torch.uint8, device=self.device)
         for f in range(D):
-            col = X[:, f].to(self.device)
+            col = X[:, f].to(self.device).contiguous()
             b = torch.bucketize(col, self.bin_edges[f]).clamp(
                 0, self.h.n_bins - 1)
             b = torch.where(torch.isnan(col),
