@@ -7,14 +7,17 @@
 // scalar LDS writes (measured 125 us for 256x624x65536 — 9x off the
 // ~14 us traffic floor: VALU/LDS-write bound, guide common-mistake #1).
 // This kernel instead:
-//   * stages [64k x 128row] panels with 16 B vector loads packed into a
-//     [k/8][row/4][8][4] blocked LDS image via two ds_write_b64 per
-//     load (probe-verified: ds_read_b64_tr_b16 reads elements
-//     {p, p+4, p+8, p+12} — stride FOUR 16-bit elements, which no
-//     glds-fillable image can serve, tools/debug_wgrad_probe.py)
-//   * reads MFMA fragments with the hardware transpose read: two tr
-//     reads (block offsets 0 and +16) yield the lane's 8 K-major values
-//     — the "attention-V recipe" from the CDNA4 guide
+//   * stages [64k x 128row] panels with global_load_lds into the PLAIN
+//     linear [k][128] image (fully coalesced 16 B pieces)
+//   * reads MFMA fragments with the hardware transpose read
+//     ds_read_b64_tr_b16. Probe-verified semantics
+//     (tools/debug_wgrad_probe.py): within each 16-lane group, lane i
+//     loads 4 contiguous elements at its address and the hardware
+//     redistributes them as out(lane 4*(i&3)+t, elem i>>2) =
+//     load[i][t]. Feeding addr_i = (k0g + (i>>2))*128 + col0 + 4*(i&3)
+//     therefore hands every lane its fragment column — k-major operands
+//     consumed straight from a row-major image, no scalar transposes
+//     anywhere (the "attention-V recipe" from the CDNA4 guide)
 //   * split-K over gridDim.z with fp32 atomic accumulation into C
 //     (pre-zeroed by the launcher), as in the generic kernel.
 // Eligibility (launcher): transA && transB, M%16==0, N%16==0, M%8==0
@@ -34,69 +37,62 @@ typedef __attribute__((ext_vector_type(4))) float wg_f32x4;
 #define WG_BN 128
 #define WG_BK 64
 
-// Blocked image: element (k, m) of the [BK x 128] panel lives at
-//   ((k/8)*32 + m/4)*32 + (k%8)*4 + (m%4)
-// i.e. [k/8][m/4] blocks of [8 k][4 m] — the k stride inside a block is
-// exactly the tr read's element stride of 4.
-__device__ __forceinline__ int wg_blk(int k, int m) {
-  return (((k >> 3) << 5) + (m >> 2)) * 32 + ((k & 7) << 2) + (m & 3);
-}
-
 // Stage a [BK x 128] K-major panel (g = base of [K, ld] array, columns
-// col0..col0+127, k rows k0..k0+BK) into the blocked image: 16 B vector
-// loads (8 contiguous row elements), each split into two 8 B
-// ds_write_b64 (the two 4-wide m-blocks it spans). 256 threads, 4
-// load+write pairs each. Caller guarantees range + 16 B alignment.
-__device__ __forceinline__ void wg_stage_vec(const __bf16* __restrict__ g,
-                                             long ld, int k0, int col0,
-                                             __bf16* dst) {
+// col0..col0+127, k rows k0..k0+BK) into the plain linear [k][128]
+// image with glds: 16 instructions x 64 lanes x 16 B, fully coalesced.
+// Caller guarantees range + 16 B alignment.
+__device__ __forceinline__ void wg_stage_glds(const __bf16* __restrict__ g,
+                                              long ld, int k0, int col0,
+                                              __bf16* dst) {
   const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
 #pragma unroll
   for (int it = 0; it < 4; ++it) {
-    const int t = tid + it * 256;   // 1024 load slots
-    const int k = t >> 4;           // 0..63
-    const int m0 = (t & 15) * 8;    // 0..120
-    const wg_bf16x8 v =
-        *(const wg_bf16x8*)(g + (long)(k0 + k) * ld + col0 + m0);
-    wg_bf16x4 lo, hi;
-    lo.x = v[0]; lo.y = v[1]; lo.z = v[2]; lo.w = v[3];
-    hi.x = v[4]; hi.y = v[5]; hi.z = v[6]; hi.w = v[7];
-    *(wg_bf16x4*)(dst + wg_blk(k, m0)) = lo;
-    *(wg_bf16x4*)(dst + wg_blk(k, m0 + 4)) = hi;
+    const int instr = wave * 4 + it;         // 16 instructions
+    const int k = instr * 4 + (lane >> 4);   // 4 k-rows per instruction
+    const int m0 = (lane & 15) * 8;
+    const __bf16* src = g + (long)(k0 + k) * ld + col0 + m0;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)src,
+        (__attribute__((address_space(3))) void*)(dst + instr * 512), 16, 0,
+        0);
   }
 }
 
-// Scalar blocked staging for K-tails / range edges (zero-padded).
+// Scalar linear staging for K-tails / range edges (zero-padded).
 __device__ __forceinline__ void wg_stage_scalar(const __bf16* __restrict__ g,
                                                 long ld, int k0, int col0,
                                                 int K, int ncols,
                                                 __bf16* dst) {
   const int tid = threadIdx.x;
   for (int e = tid; e < 8192; e += 256) {
-    // e enumerated as (k, m) in panel order
     const int k = e >> 7;
     const int m = e & 127;
     const int gk = k0 + k;
     const int c = col0 + m;
     __bf16 v = (__bf16)0.f;
     if (gk < K && c < ncols) v = g[(long)gk * ld + c];
-    dst[wg_blk(k, m)] = v;
+    dst[e] = v;
   }
 }
 
 __device__ __forceinline__ wg_bf16x8 wg_frag(const __bf16* base, int fb,
                                              int kc, int lane) {
-  // fragment (16-row block fb, k-chunk kc): lane l needs column
-  // m = fb*16 + (l&15) of k = kc*32 + (l>>4)*8 + 0..7. tr reads
-  // {p, p+4, p+8, p+12}: p at (kb, m) block base + m%4 gives kk 0..3,
-  // +16 gives kk 4..7.
-  const int kb = kc * 4 + (lane >> 4);
-  const int m = fb * 16 + (lane & 15);
-  const __bf16* p = base + ((kb << 5) + (m >> 2)) * 32 + (m & 3);
+  // fragment (16-row block fb, k-chunk kc) from the LINEAR [k][128]
+  // image via the cooperative transpose read. Within a 16-lane group
+  // (i = lane & 15), hardware redistributes lane i's 4 loaded elements
+  // as out(lane 4*(i&3)+t, elem i>>2) = load[i][t]; with
+  //   addr_i = (k0g + (i>>2)) * 128 + fb*16 + 4*(i&3)
+  // every lane ends up with its column m = fb*16 + (l&15) at
+  // k = k0g + (0..3); the second read at +4 rows gives k +4..7.
+  const int i = lane & 15;
+  const int k0g = kc * 32 + (lane >> 4) * 8;
+  const __bf16* p = base + (k0g + (i >> 2)) * 128 + fb * 16 + 4 * (i & 3);
   const wg_bf16x4 lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
       (__attribute__((address_space(3))) wg_bf16x4*)p);
   const wg_bf16x4 hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-      (__attribute__((address_space(3))) wg_bf16x4*)(p + 16));
+      (__attribute__((address_space(3))) wg_bf16x4*)(p + 4 * 128));
   wg_bf16x8 out;
   out[0] = lo.x; out[1] = lo.y; out[2] = lo.z; out[3] = lo.w;
   out[4] = hi.x; out[5] = hi.y; out[6] = hi.z; out[7] = hi.w;
@@ -124,11 +120,11 @@ __global__ __launch_bounds__(256) void gemm_wgrad_bf16_kernel(
   for (int k0 = kbeg; k0 < kend; k0 += WG_BK) {
     const bool k_full = k0 + WG_BK <= kend;
     if (a_full && k_full)
-      wg_stage_vec(At, M, k0, M0, As);
+      wg_stage_glds(At, M, k0, M0, As);
     else
       wg_stage_scalar(At, M, k0, M0, min(kend, K), M, As);
     if (b_full && k_full)
-      wg_stage_vec(Bt, N, k0, N0, Bs);
+      wg_stage_glds(Bt, N, k0, N0, Bs);
     else
       wg_stage_scalar(Bt, N, k0, N0, min(kend, K), N, Bs);
     __syncthreads();
@@ -179,7 +175,8 @@ __global__ void wg_probe_kernel(const __bf16* __restrict__ g, int ld,
                                 float* __restrict__ out_frag, int mode) {
   __shared__ __bf16 s[WG_BK * WG_BM];
   if (mode == 0) {
-    wg_stage_vec(g, ld, 0, 0, s);
+    wg_stage_glds(g, ld, 0, 0, s);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
   } else {
     for (int e = threadIdx.x; e < WG_BK * WG_BM; e += 256) s[e] = g[e];

@@ -13,9 +13,7 @@ BK, BM = 64, 128
 
 
 def blocked_index(k, m):
-    kb, kk = k // 8, k % 8
-    mb, mm = m // 4, m % 4
-    return (kb * 32 + mb) * 32 + kk * 4 + mm
+    return k * 128 + m  # linear image
 
 
 def main():
@@ -75,12 +73,13 @@ def main():
             for kc in range(2):
                 for l in range(64):
                     for e in range(8):
-                        # tr semantics (probe-verified): elems
-                        # {p, p+4, p+8, p+12}; second read at p+16
-                        kb = kc * 4 + (l >> 4)
-                        m = fb * 16 + (l & 15)
-                        p = (kb * 32 + m // 4) * 32 + (m % 4)
-                        idx = p + (e % 4) * 4 + (e // 4) * 16
+                        # cooperative tr model: out(l, j) =
+                        # load[4j + (i>>2)][i&3] over per-lane addrs
+                        # addr_i = (k0g + (i>>2))*128 + fb*16 + 4*(i&3);
+                        # net effect: element (k0g + e, fb*16 + (l&15))
+                        i = l & 15
+                        k0g = kc * 32 + (l >> 4) * 8
+                        idx = (k0g + e) * 128 + fb * 16 + i
                         want = float(ident[idx] if idx < BK * BM else -1)
                         got = float(fr2[fb, kc, l, e])
                         if got != want:
