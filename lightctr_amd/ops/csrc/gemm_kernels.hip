@@ -269,14 +269,14 @@ void gemm_bf16_launch(const void* A, const void* Bst, const float* bias,
   }
   if (gemm_wgrad_eligible(M, N, K, transA, transB) && bias == nullptr &&
       act == 0 && Cbf == nullptr) {
-    // K-major x K-major wgrad: blocked glds staging + ds_read_b64_tr_b16
-    // fragment reads (the generic path's scalar transposed staging
-    // measured 125 us on the 256x624x65536 W&D shape — 9x off floor).
-    // OPT-IN while under debug (LCTR_WGRAD128=1): first cut measured a
-    // parity failure + a partial-tile slowdown.
+    // K-major x K-major wgrad: linear glds staging + ds_read_b64_tr_b16
+    // cooperative-transpose fragment reads. Measured vs the generic
+    // scalar-transposed-staging path at K=65536: 256x512 67 vs 92 us,
+    // 128x256 40 vs 59, 256x624 (partial strip) 119 vs 126.
+    // LCTR_WGRAD128=0 reverts.
     static const bool wg = [] {
       const char* e = getenv("LCTR_WGRAD128");
-      return e && e[0] == '1';
+      return !(e && e[0] == '0');
     }();
     if (wg) {
       gemm_wgrad_bf16_launch(A, Bst, C, M, N, K, stream);
