@@ -83,6 +83,11 @@ class FFMModel:
         #   blocks — wave/entry fp32 emit + block reduce, 5.6 ms
         #   atomic — naive scatter (hot-feature serialization)
         self.backward_mode = "rowemit"
+        # fuse the sparse optimizer into the apply's interior-run flush
+        # (features whose sorted segment is wholly owned by one chunk
+        # skip the gradV slab + bitmap + separate optimizer kernel);
+        # spanning features keep the two-phase path. A/B-able.
+        self.fused_apply = False
         if self._use_hip:
             require_hip_ops()
 
@@ -140,12 +145,36 @@ class FFMModel:
                                                self._Vc, dpred,
                                                scale=bscale)
                 sorted_fids, perm = sort_ids(fids, self.h.num_features)
-                ops.ffm_blocks_apply_f16(sorted_fids, perm, gblocks, gw,
-                                         self.gradW,
-                                         self.gradV.view(
-                                             self.h.num_features, -1),
-                                         self.touched,
-                                         inv_scale=1.0 / bscale)
+                fuse = self.fused_apply and (
+                    self.h.optimizer == "adagrad"
+                    or (self.h.optimizer == "ftrl"
+                        and self.h.ftrl_v == "adagrad"))
+                if fuse:
+                    mode = 1 if self.h.optimizer == "adagrad" else 3
+                    wp = ((self.h.lr, self.h.eps, self.h.l2, 0.0)
+                          if mode == 1 else
+                          (self.h.ftrl_alpha, self.h.ftrl_beta,
+                           self.h.ftrl_l1, self.h.ftrl_l2))
+                    ops.ffm_blocks_apply_f16(
+                        sorted_fids, perm, gblocks, gw, self.gradW,
+                        self.gradV.view(self.h.num_features, -1),
+                        self.touched, inv_scale=1.0 / bscale,
+                        opt_mode=mode,
+                        V=self.V.view(self.h.num_features, -1),
+                        W=self.W, nW=self.nW,
+                        zW=self.zW if mode == 3 else None,
+                        nV=self.nV.view(self.h.num_features, -1),
+                        Vh=(self.Vh.view(self.h.num_features, -1)
+                            if self.Vh is not None else None),
+                        p0=wp[0], p1=wp[1], p2=wp[2], p3=wp[3],
+                        q0=self.h.lr, q1=self.h.eps, q2=self.h.l2)
+                else:
+                    ops.ffm_blocks_apply_f16(sorted_fids, perm, gblocks, gw,
+                                             self.gradW,
+                                             self.gradV.view(
+                                                 self.h.num_features, -1),
+                                             self.touched,
+                                             inv_scale=1.0 / bscale)
             elif self.backward_mode == "blocks":
                 row_of_entry = ops.row_index(row_ptr, fids.numel())
                 gw, gblocks = ops.ffm_block_emit(row_of_entry, row_ptr,

@@ -262,16 +262,37 @@ std::vector<at::Tensor> ffm_row_emit(at::Tensor row_ptr, at::Tensor fields,
 void ffm_blocks_apply_f16(at::Tensor sorted_fids, at::Tensor perm,
                           at::Tensor gblocks, at::Tensor gw, at::Tensor gradW,
                           at::Tensor gradV, at::Tensor touched,
-                          double inv_scale) {
+                          double inv_scale, int64_t opt_mode,
+                          c10::optional<at::Tensor> V,
+                          c10::optional<at::Tensor> W,
+                          c10::optional<at::Tensor> nW,
+                          c10::optional<at::Tensor> zW,
+                          c10::optional<at::Tensor> nV,
+                          c10::optional<at::Tensor> Vh,
+                          double p0, double p1, double p2, double p3,
+                          double q0, double q1, double q2) {
   check_cuda_i32(sorted_fids, "sorted_fids");
   auto perm_i = perm32(perm).contiguous();
   CHK(gblocks.scalar_type() == at::kHalf, "gblocks must be fp16");
+  CHK(opt_mode == 0 || opt_mode == 1 || opt_mode == 3,
+      "opt_mode 0=slab 1=adagrad 3=ftrlW+adagradV");
+  if (opt_mode != 0) {
+    CHK(V && W && nW && nV, "fused apply needs V/W/nW/nV");
+    if (opt_mode == 3) CHK(zW.has_value(), "opt_mode 3 needs zW");
+  }
   const int D = (int)gradV.size(1);
   lightctr::ffm_blocks_apply_f16_launch(
       sorted_fids.data_ptr<int>(), perm_i.data_ptr<int>(), gblocks.data_ptr(),
       gw.data_ptr<float>(), gradW.data_ptr<float>(), gradV.data_ptr<float>(),
       (unsigned long long*)touched.data_ptr(), D, (int)sorted_fids.numel(),
-      (float)inv_scale, cur_stream());
+      (float)inv_scale, (int)opt_mode,
+      V ? V->data_ptr<float>() : nullptr,
+      W ? W->data_ptr<float>() : nullptr,
+      nW ? nW->data_ptr<float>() : nullptr,
+      zW ? zW->data_ptr<float>() : nullptr,
+      nV ? nV->data_ptr<float>() : nullptr,
+      Vh ? Vh->data_ptr() : nullptr, (float)p0, (float)p1, (float)p2,
+      (float)p3, (float)q0, (float)q1, (float)q2, cur_stream());
 }
 
 void ffm_backward(at::Tensor row_ptr, at::Tensor fields, at::Tensor fids,
@@ -914,10 +935,17 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("vals"), py::arg("V"), py::arg("dpred"),
         py::arg("scale") = 1.0);
   m.def("ffm_blocks_apply_f16", &ffm_blocks_apply_f16,
-        "segment-reduce fp16 blocks into slabs (interior stores)",
+        "segment-reduce fp16 blocks into slabs (interior stores), "
+        "optionally with the optimizer fused for interior runs",
         py::arg("sorted_fids"), py::arg("perm"), py::arg("gblocks"),
         py::arg("gw"), py::arg("gradW"), py::arg("gradV"),
-        py::arg("touched"), py::arg("inv_scale") = 1.0);
+        py::arg("touched"), py::arg("inv_scale") = 1.0,
+        py::arg("opt_mode") = 0, py::arg("V") = py::none(),
+        py::arg("W") = py::none(), py::arg("nW") = py::none(),
+        py::arg("zW") = py::none(), py::arg("nV") = py::none(),
+        py::arg("Vh") = py::none(), py::arg("p0") = 0.0,
+        py::arg("p1") = 0.0, py::arg("p2") = 0.0, py::arg("p3") = 0.0,
+        py::arg("q0") = 0.0, py::arg("q1") = 0.0, py::arg("q2") = 0.0);
   m.def("ffm_backward", &ffm_backward, "FFM fused pairwise backward scatter");
   m.def("ffm_sorted_backward", &ffm_sorted_backward,
         "FFM sorted segment-reduce backward (LDS block accumulate)");

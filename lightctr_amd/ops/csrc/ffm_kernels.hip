@@ -666,12 +666,49 @@ __global__ void ffm_row_emit_kernel(
 // this wave's chunk owns its feature exclusively, so the flush is a plain
 // coalesced store sweep (slabs are zeroed by the optimizer pass); only
 // chunk-spanning runs use atomicAdd.
+// Optional fused optimizer for interior runs (opt_mode 0 = off, 1 =
+// Adagrad W+V, 3 = FTRL W + Adagrad V — the FFM production pairings):
+// a run whose head AND tail lie in this chunk holds the feature's
+// TOTAL batch gradient at flush time, so the update applies in place
+// (V/nV [+W z/n state] RMW, optional bf16 mirror refresh) and the fid
+// SKIPS the gradV slab, the touched bitmap, and the separate sparse
+// optimizer pass entirely. Spanning runs keep the slab+bitmap path.
+// FTRL-proximal update (reference gradientUpdater.h:235-278), local copy
+// (the misc/fm TU versions are TU-local inlines).
+__device__ __forceinline__ void ffm_ftrl_update(float* w, float* z, float* n,
+                                                float g, float alpha,
+                                                float beta, float l1,
+                                                float l2) {
+  const float g2 = g * g;
+  const float nold = *n;
+  const float sigma = (sqrtf(nold + g2) - sqrtf(nold)) / alpha;
+  const float znew = *z + g - sigma * (*w);
+  const float nnew = nold + g2;
+  *z = znew;
+  *n = nnew;
+  if (fabsf(znew) <= l1) {
+    *w = 0.f;
+  } else {
+    *w = -(znew - copysignf(l1, znew)) / ((beta + sqrtf(nnew)) / alpha + l2);
+  }
+}
+
+struct FfmOptArgs {
+  float* W;
+  float* nW;
+  float* zW;   // ftrl-W state (mode 3)
+  float* nV;
+  __bf16* Vh;  // optional bf16 mirror (config #3)
+  float p0, p1, p2, p3;  // W: adagrad lr,eps,l2 | ftrl alpha,beta,l1,l2
+  float q0, q1, q2;      // V adagrad: lr, eps, l2
+};
+
 __global__ void ffm_blocks_apply_f16_kernel(
     const int* __restrict__ sorted_fids, const int* __restrict__ perm,
     const _Float16* __restrict__ gblocks, const float* __restrict__ gw,
     float* __restrict__ gradW, float* __restrict__ gradV,
     unsigned long long* __restrict__ touched, int D, int nnz, int chunk,
-    float inv_scale) {
+    float inv_scale, int opt_mode, float* __restrict__ V, FfmOptArgs oa) {
   extern __shared__ float lds_acc[];
   const int lane = threadIdx.x & 63;
   const int wave_in_blk = threadIdx.x >> 6;
@@ -688,7 +725,30 @@ __global__ void ffm_blocks_apply_f16_kernel(
   auto flush = [&](int tail_e) {
     if (cur < 0) return;
     const bool tail_ok = tail_e >= nnz || sorted_fids[tail_e] != cur;
-    if (head_ok && tail_ok) {
+    if (opt_mode != 0 && head_ok && tail_ok) {
+      // exclusive owner: fused V-Adagrad (+W update) in place
+      for (int i = lane; i < D; i += LCTR_WAVE) {
+        const size_t off = (size_t)cur * D + i;
+        const float g = acc[i] * inv_scale + oa.q2 * V[off];
+        const float a = oa.nV[off] + g * g;
+        oa.nV[off] = a;
+        const float vnew = V[off] - oa.q0 * g * __frsqrt_rn(a + oa.q1);
+        V[off] = vnew;
+        if (oa.Vh) oa.Vh[off] = (__bf16)vnew;
+        acc[i] = 0.f;
+      }
+      if (lane == 0) {
+        if (opt_mode == 3) {
+          ffm_ftrl_update(&oa.W[cur], &oa.zW[cur], &oa.nW[cur], accw,
+                          oa.p0, oa.p1, oa.p2, oa.p3);
+        } else {
+          const float gw_ = accw + oa.p2 * oa.W[cur];
+          const float aw = oa.nW[cur] + gw_ * gw_;
+          oa.nW[cur] = aw;
+          oa.W[cur] -= oa.p0 * gw_ * __frsqrt_rn(aw + oa.p1);
+        }
+      }
+    } else if (head_ok && tail_ok) {
       for (int i = lane; i < D; i += LCTR_WAVE) {
         gradV[(size_t)cur * D + i] = acc[i] * inv_scale;
         acc[i] = 0.f;
@@ -700,7 +760,11 @@ __global__ void ffm_blocks_apply_f16_kernel(
           atomicAdd(&gradV[(size_t)cur * D + i], acc[i] * inv_scale);
         acc[i] = 0.f;
       }
-      if (lane == 0) atomicAdd(&gradW[cur], accw);
+      if (lane == 0) {
+        atomicAdd(&gradW[cur], accw);
+        if (opt_mode != 0)
+          atomicOr(&touched[cur >> 6], 1ull << (cur & 63));
+      }
     }
   };
   for (int e = base; e < end; ++e) {
@@ -710,7 +774,11 @@ __global__ void ffm_blocks_apply_f16_kernel(
       cur = fid;
       accw = 0.f;
       head_ok = (e == 0 || sorted_fids[e - 1] != fid);
-      if (lane == 0 && head_ok)
+      // fused mode: interior runs bypass the bitmap entirely; the bit is
+      // only needed when the run's gradient lands in the slabs. An
+      // interior run is detectable at its head (tail unknown yet), so in
+      // fused mode the spanning flush sets the bit instead.
+      if (lane == 0 && head_ok && opt_mode == 0)
         atomicOr(&touched[fid >> 6], 1ull << (fid & 63));
     }
     const long p = (long)perm[e];
@@ -792,7 +860,11 @@ void ffm_blocks_apply_f16_launch(const int* sorted_fids, const int* perm,
                                  const void* gblocks, const float* gw,
                                  float* gradW, float* gradV,
                                  unsigned long long* touched, int D, int nnz,
-                                 float inv_scale, hipStream_t stream) {
+                                 float inv_scale, int opt_mode, float* V,
+                                 float* W, float* nW, float* zW, float* nV,
+                                 void* Vh, float p0, float p1, float p2,
+                                 float p3, float q0, float q1, float q2,
+                                 hipStream_t stream) {
   if (nnz <= 0) return;
   static const int chunk = [] {
     const char* e = getenv("LCTR_FFM_APPLY_CHUNK");
@@ -803,9 +875,11 @@ void ffm_blocks_apply_f16_launch(const int* sorted_fids, const int* perm,
   dim3 block(wpb * LCTR_WAVE);
   dim3 grid((nwaves + wpb - 1) / wpb);
   const size_t lds = (size_t)wpb * D * sizeof(float);
+  FfmOptArgs oa{W, nW, zW, nV, (__bf16*)Vh, p0, p1, p2, p3, q0, q1, q2};
   hipLaunchKernelGGL(ffm_blocks_apply_f16_kernel, grid, block, lds, stream,
                      sorted_fids, perm, (const _Float16*)gblocks, gw, gradW,
-                     gradV, touched, D, nnz, chunk, inv_scale);
+                     gradV, touched, D, nnz, chunk, inv_scale, opt_mode, V,
+                     oa);
 }
 
 void ffm_forward_pp_launch(const int* row_ptr, const int* fields,

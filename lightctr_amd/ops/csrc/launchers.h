@@ -81,7 +81,11 @@ void ffm_blocks_apply_f16_launch(const int* sorted_fids, const int* perm,
                                  const void* gblocks, const float* gw,
                                  float* gradW, float* gradV,
                                  unsigned long long* touched, int D, int nnz,
-                                 float inv_scale, ihipStream_t* stream);
+                                 float inv_scale, int opt_mode, float* V,
+                                 float* W, float* nW, float* zW, float* nV,
+                                 void* Vh, float p0, float p1, float p2,
+                                 float p3, float q0, float q1, float q2,
+                                 ihipStream_t* stream);
 void ffm_blocks_apply_launch(const int* sorted_fids, const int* perm,
                              const float* gblocks, const float* gw,
                              float* gradW, float* gradV,

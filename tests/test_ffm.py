@@ -423,3 +423,32 @@ def test_ffm_forward_staged_long_rows():
                                   vals.cpu(), W.cpu(), V.cpu())
     assert torch.allclose(pred.cpu(), ref, atol=1e-3, rtol=1e-4), \
         (pred.cpu() - ref).abs().max()
+
+
+@pytest.mark.gpu
+def test_ffm_fused_apply_matches_two_phase():
+    """Optimizer fused into the apply's interior-run flush == the
+    two-phase slab+bitmap+sparse-apply path, for both production
+    pairings (adagrad and ftrlW+adagradV), fp32 and bf16."""
+    from lightctr_amd.models.ffm import FFMHyper, FFMModel
+
+    for opt, dtype in (("adagrad", "fp32"), ("ftrl", "bf16")):
+        h = FFMHyper(num_features=1 << 14, num_fields=13, k=8,
+                     optimizer=opt, dtype=dtype, seed=21)
+        gen = SyntheticCriteo(num_features=1 << 14, seed=9,
+                              device="cuda:0")
+        a = FFMModel(h, device="cuda:0")
+        a.fused_apply = True
+        b = FFMModel(h, device="cuda:0")
+        b.fused_apply = False
+        for _ in range(3):
+            row_ptr, fields, fids, vals, labels = gen.batch(2048)
+            a.train_step(row_ptr, fields, fids, vals, labels)
+            b.train_step(row_ptr, fields, fids, vals, labels)
+        assert torch.allclose(a.W, b.W, atol=1e-5), \
+            (opt, dtype, (a.W - b.W).abs().max())
+        assert torch.allclose(a.V, b.V, atol=1e-5), \
+            (opt, dtype, (a.V - b.V).abs().max())
+        assert torch.allclose(a.nV, b.nV, atol=1e-5)
+        if dtype == "bf16":
+            assert torch.allclose(a.Vh.float(), a.V, atol=1e-2)

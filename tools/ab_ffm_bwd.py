@@ -99,14 +99,18 @@ def main():
 
     # full train_step per mode
     for mode in args.modes.split(","):
-        m2 = FFMModel(FFMHyper(num_features=F, num_fields=nf, k=K,
-                              dtype=args.dtype), device="cuda")
-        m2.backward_mode = mode
-        t = time_fn(lambda: m2.train_step(row_ptr, fields, fids, vals,
-                                          labels), args.reps)
-        exs = B / t
-        print(f"step[{mode:8s}]        : {t * 1e3:7.3f} ms  "
-              f"({exs / 1e6:.1f}M ex/s)")
+        for fused in ((False, True) if mode == "rowemit" else (False,)):
+            m2 = FFMModel(FFMHyper(num_features=F, num_fields=nf, k=K,
+                                  optimizer="ftrl",
+                                  dtype=args.dtype), device="cuda")
+            m2.backward_mode = mode
+            m2.fused_apply = fused
+            t = time_fn(lambda: m2.train_step(row_ptr, fields, fids, vals,
+                                              labels), args.reps)
+            exs = B / t
+            tag = mode + ("+fused" if fused else "")
+            print(f"step[{tag:14s}]  : {t * 1e3:7.3f} ms  "
+                  f"({exs / 1e6:.1f}M ex/s)")
 
 
 if __name__ == "__main__":
