@@ -35,8 +35,9 @@ def main():
     elif args.model == "ffm":
         from lightctr_amd.models.ffm import FFMHyper, FFMModel
 
+        # bf16 compute mirror (config #3 serving precision)
         m = FFMModel(FFMHyper(num_features=args.features, num_fields=39,
-                              k=8), device=dev)
+                              k=8, dtype="bf16"), device=dev)
         fn = lambda: m.predict_proba(row_ptr, fields, fids, vals)
     elif args.model == "nfm":
         from lightctr_amd.models.nfm import NFMHyper, NFMModel
