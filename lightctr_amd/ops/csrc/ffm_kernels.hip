@@ -703,6 +703,7 @@ struct FfmOptArgs {
   float q0, q1, q2;      // V adagrad: lr, eps, l2
 };
 
+template <bool FUSED>
 __global__ void ffm_blocks_apply_f16_kernel(
     const int* __restrict__ sorted_fids, const int* __restrict__ perm,
     const _Float16* __restrict__ gblocks, const float* __restrict__ gw,
@@ -725,7 +726,7 @@ __global__ void ffm_blocks_apply_f16_kernel(
   auto flush = [&](int tail_e) {
     if (cur < 0) return;
     const bool tail_ok = tail_e >= nnz || sorted_fids[tail_e] != cur;
-    if (opt_mode != 0 && head_ok && tail_ok) {
+    if (FUSED && head_ok && tail_ok) {
       // exclusive owner: fused V-Adagrad (+W update) in place
       for (int i = lane; i < D; i += LCTR_WAVE) {
         const size_t off = (size_t)cur * D + i;
@@ -762,8 +763,7 @@ __global__ void ffm_blocks_apply_f16_kernel(
       }
       if (lane == 0) {
         atomicAdd(&gradW[cur], accw);
-        if (opt_mode != 0)
-          atomicOr(&touched[cur >> 6], 1ull << (cur & 63));
+        if (FUSED) atomicOr(&touched[cur >> 6], 1ull << (cur & 63));
       }
     }
   };
@@ -778,7 +778,7 @@ __global__ void ffm_blocks_apply_f16_kernel(
       // only needed when the run's gradient lands in the slabs. An
       // interior run is detectable at its head (tail unknown yet), so in
       // fused mode the spanning flush sets the bit instead.
-      if (lane == 0 && head_ok && opt_mode == 0)
+      if (lane == 0 && head_ok && !FUSED)
         atomicOr(&touched[fid >> 6], 1ull << (fid & 63));
     }
     const long p = (long)perm[e];
@@ -878,10 +878,16 @@ void ffm_blocks_apply_f16_launch(const int* sorted_fids, const int* perm,
   dim3 grid((nwaves + wpb - 1) / wpb);
   const size_t lds = (size_t)wpb * D * sizeof(float);
   FfmOptArgs oa{W, nW, zW, nV, (__bf16*)Vh, p0, p1, p2, p3, q0, q1, q2};
-  hipLaunchKernelGGL(ffm_blocks_apply_f16_kernel, grid, block, lds, stream,
-                     sorted_fids, perm, (const _Float16*)gblocks, gw, gradW,
-                     gradV, touched, D, nnz, chunk, inv_scale, opt_mode, V,
-                     oa);
+  if (opt_mode != 0)
+    hipLaunchKernelGGL((ffm_blocks_apply_f16_kernel<true>), grid, block, lds,
+                       stream, sorted_fids, perm, (const _Float16*)gblocks,
+                       gw, gradW, gradV, touched, D, nnz, chunk, inv_scale,
+                       opt_mode, V, oa);
+  else
+    hipLaunchKernelGGL((ffm_blocks_apply_f16_kernel<false>), grid, block,
+                       lds, stream, sorted_fids, perm,
+                       (const _Float16*)gblocks, gw, gradW, gradV, touched,
+                       D, nnz, chunk, inv_scale, opt_mode, V, oa);
 }
 
 void ffm_forward_pp_launch(const int* row_ptr, const int* fields,
