@@ -433,7 +433,8 @@ def test_ffm_fused_apply_matches_two_phase():
     from lightctr_amd.models.ffm import FFMHyper, FFMModel
 
     for opt, dtype in (("adagrad", "fp32"), ("ftrl", "bf16")):
-        h = FFMHyper(num_features=1 << 14, num_fields=13, k=8,
+        # num_fields must match the generator's 39 fields
+        h = FFMHyper(num_features=1 << 14, num_fields=39, k=8,
                      optimizer=opt, dtype=dtype, seed=21)
         gen = SyntheticCriteo(num_features=1 << 14, seed=9,
                               device="cuda:0")
