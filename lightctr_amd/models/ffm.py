@@ -88,6 +88,7 @@ class FFMModel:
         # skip the gradV slab + bitmap + separate optimizer kernel);
         # spanning features keep the two-phase path. A/B-able.
         self.fused_apply = False
+        self._fields_checked = False
         if self._use_hip:
             require_hip_ops()
 
@@ -131,6 +132,13 @@ class FFMModel:
     def train_step(self, row_ptr, fields, fids, vals, labels) -> torch.Tensor:
         B = row_ptr.numel() - 1
         scale = 1.0 / B
+        if not self._fields_checked:
+            # one-time guard: an out-of-range field silently corrupts the
+            # [F, nfields, K] indexing on device (costs one sync, once)
+            fmax = int(fields.max()) if fields.numel() else 0
+            assert fmax < self.h.num_fields, \
+                f"field id {fmax} >= num_fields {self.h.num_fields}"
+            self._fields_checked = True
         if self._use_hip:
             ops = require_hip_ops()
             pred = ops.ffm_forward(row_ptr, fields, fids, vals, self.W,
