@@ -22,9 +22,19 @@ def main():
     a.fused_apply = True
     b = FFMModel(h, device="cuda:0")
     b.fused_apply = False
-    row_ptr, fields, fids, vals, labels = gen.batch(2048)
-    a.train_step(row_ptr, fields, fids, vals, labels)
-    b.train_step(row_ptr, fields, fids, vals, labels)
+    for step in range(3):
+        row_ptr, fields, fids, vals, labels = gen.batch(2048)
+        a.train_step(row_ptr, fields, fids, vals, labels)
+        b.train_step(row_ptr, fields, fids, vals, labels)
+        print(f"step {step}: dV={float((a.V - b.V).abs().max()):.3e} "
+              f"dW={float((a.W - b.W).abs().max()):.3e} "
+              f"dnV={float((a.nV - b.nV).abs().max()):.3e} "
+              f"dnW={float((a.nW - b.nW).abs().max()):.3e} "
+              f"dgV={float((a.gradV - b.gradV).abs().max()):.3e} "
+              f"a.gV={float(a.gradV.abs().max()):.3e} "
+              f"b.gV={float(b.gradV.abs().max()):.3e} "
+              f"a.tc={int(a.touched.ne(0).sum())} "
+              f"b.tc={int(b.touched.ne(0).sum())}")
 
     dW = (a.W - b.W).abs()
     dV = (a.V - b.V).abs().amax(dim=(1, 2))
