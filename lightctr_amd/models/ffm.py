@@ -86,8 +86,11 @@ class FFMModel:
         # fuse the sparse optimizer into the apply's interior-run flush
         # (features whose sorted segment is wholly owned by one chunk
         # skip the gradV slab + bitmap + separate optimizer kernel);
-        # spanning features keep the two-phase path. A/B-able.
-        self.fused_apply = False
+        # spanning features keep the two-phase path. Measured 3.65 vs
+        # 4.15 ms/step (18.0 vs 15.8M ex/s, bf16 B=65536) — default ON
+        # for the supported pairings (adagrad, ftrlW+adagradV); other
+        # pairings fall back to two-phase automatically.
+        self.fused_apply = True
         self._fields_checked = False
         if self._use_hip:
             require_hip_ops()
