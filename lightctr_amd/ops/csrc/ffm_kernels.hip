@@ -806,16 +806,18 @@ __global__ void ffm_blocks_apply_f16_kernel(
   flush(end);
 }
 
-static size_t ffm_row_emit_lds_bytes(int nfields, int K, int maxn) {
-  // acc (4 waves, dup fallback) | fp16 stage | vals_s | fmap | dup flag
-  return (size_t)4 * nfields * (K + 1) * sizeof(float) +
+static size_t ffm_row_emit_lds_bytes(int nfields, int K, int maxn,
+                                     int nwaves) {
+  // acc (per wave, dup fallback) | fp16 stage | vals_s | fmap | dup flag
+  return (size_t)nwaves * nfields * (K + 1) * sizeof(float) +
          (size_t)maxn * nfields * K * sizeof(_Float16) +
          (size_t)maxn * sizeof(float) + (size_t)nfields * sizeof(int) +
          sizeof(int);
 }
 
 bool ffm_staged_eligible(int nfields, int K, int maxn) {
-  return ffm_row_emit_lds_bytes(nfields, K, maxn) <= (64 << 10);
+  // sized for the largest selectable block (8 waves)
+  return ffm_row_emit_lds_bytes(nfields, K, maxn, 8) <= (64 << 10);
 }
 
 void ffm_fwd_staged_launch(const int* row_ptr, const int* fields,
@@ -838,9 +840,14 @@ void ffm_row_emit_launch(const int* row_ptr, const int* fields,
                          float* gw, int nfields, int B, int maxn, int K,
                          float scale, hipStream_t stream) {
   if (B <= 0) return;
-  dim3 block(256);
+  static const int bs = [] {
+    const char* e = getenv("LCTR_FFM_EMIT_BLOCK");
+    const int v = e ? atoi(e) : 256;
+    return (v == 128 || v == 256 || v == 512) ? v : 256;
+  }();
+  dim3 block(bs);
   dim3 grid(B);
-  const size_t lds = ffm_row_emit_lds_bytes(nfields, K, maxn);
+  const size_t lds = ffm_row_emit_lds_bytes(nfields, K, maxn, bs / 64);
   if (v_bf16) {
     DISPATCH_FFM_K(
         K, hipLaunchKernelGGL((ffm_row_emit_kernel<KC, __bf16>), grid, block,
