@@ -115,10 +115,13 @@ class PSShard:
     def serve(self, groups):
         """One serving thread per worker (blocking recv on the worker's
         pair group), like the reference Delivery's recv event-loops +
-        handler pool (network.h:385-474). Table mutations are guarded by a
-        lock (the reference PS is deliberately Hogwild on reads — pulls
-        here read under the same lock for simplicity; relaxing it is a
-        perf knob, not a semantics change). Returns when all workers FIN."""
+        handler pool (network.h:385-474). PULLS ARE LOCK-FREE (the
+        reference PS is deliberately Hogwild on reads,
+        paramserver.h:138-139: a pull may observe a value mid-update —
+        async-SGD semantics absorb it); only pushes serialize against
+        each other under a lock so concurrent same-key updates do not
+        interleave their read-modify-writes. Returns when all workers
+        FIN."""
         cfg = self.cfg
         self._lock = threading.Lock()
         workers = list(range(cfg.ps_shards, self.world))
@@ -157,12 +160,14 @@ class PSShard:
             return
         dist.send(reply, dst=w, group=group)
         lidx = fids // cfg.ps_shards
-        with self._lock:
-            Wv = self.W[lidx].clone()
-            Vv = self.V[lidx].clone()
-            if cfg.updater in ("dcasgd", "dcasgda"):
-                self.shadowW[wi, lidx] = Wv
-                self.shadowV[wi, lidx] = Vv
+        # Hogwild read — no lock (see serve() docstring). The shadow
+        # copies are per-worker slices written only by this worker's
+        # serving thread, so they need no lock either.
+        Wv = self.W[lidx].clone()
+        Vv = self.V[lidx].clone()
+        if cfg.updater in ("dcasgd", "dcasgda"):
+            self.shadowW[wi, lidx] = Wv
+            self.shadowV[wi, lidx] = Vv
         if cfg.wire == "fp16":
             dist.send(Wv.to(torch.float16), dst=w, group=group)
             dist.send(Vv.to(torch.float16), dst=w, group=group)
