@@ -727,8 +727,44 @@ __global__ void ffm_blocks_apply_f16_kernel(
     if (cur < 0) return;
     const bool tail_ok = tail_e >= nnz || sorted_fids[tail_e] != cur;
     if (FUSED && head_ok && tail_ok) {
-      // exclusive owner: fused V-Adagrad (+W update) in place
-      for (int i = lane; i < D; i += LCTR_WAVE) {
+      // exclusive owner: fused V-Adagrad (+W update) in place, float4
+      // RMWs (D % 4 == 0 for every dispatched K; the sparse-optimizer
+      // kernel's measured win on this latency-bound state sweep)
+      const int nq4 = (D & 3) == 0 ? (D >> 2) : 0;
+      float4* V4 = (float4*)&V[(size_t)cur * D];
+      float4* nV4 = (float4*)&oa.nV[(size_t)cur * D];
+      float4* acc4 = (float4*)acc;
+      ffm_bf16x4* Vh4 =
+          oa.Vh ? (ffm_bf16x4*)&oa.Vh[(size_t)cur * D] : nullptr;
+      for (int i = lane; i < nq4; i += LCTR_WAVE) {
+        const float4 ac = acc4[i];
+        float4 v = V4[i];
+        float4 a = nV4[i];
+        const float gx = ac.x * inv_scale + oa.q2 * v.x;
+        const float gy = ac.y * inv_scale + oa.q2 * v.y;
+        const float gz = ac.z * inv_scale + oa.q2 * v.z;
+        const float gw2 = ac.w * inv_scale + oa.q2 * v.w;
+        a.x += gx * gx;
+        a.y += gy * gy;
+        a.z += gz * gz;
+        a.w += gw2 * gw2;
+        v.x -= oa.q0 * gx * __frsqrt_rn(a.x + oa.q1);
+        v.y -= oa.q0 * gy * __frsqrt_rn(a.y + oa.q1);
+        v.z -= oa.q0 * gz * __frsqrt_rn(a.z + oa.q1);
+        v.w -= oa.q0 * gw2 * __frsqrt_rn(a.w + oa.q1);
+        nV4[i] = a;
+        V4[i] = v;
+        if (Vh4) {
+          ffm_bf16x4 h;
+          h.x = (__bf16)v.x;
+          h.y = (__bf16)v.y;
+          h.z = (__bf16)v.z;
+          h.w = (__bf16)v.w;
+          Vh4[i] = h;
+        }
+        acc4[i] = make_float4(0.f, 0.f, 0.f, 0.f);
+      }
+      for (int i = 4 * nq4 + lane; i < D; i += LCTR_WAVE) {  // safety tail
         const size_t off = (size_t)cur * D + i;
         const float g = acc[i] * inv_scale + oa.q2 * V[off];
         const float a = oa.nV[off] + g * g;
@@ -750,7 +786,16 @@ __global__ void ffm_blocks_apply_f16_kernel(
         }
       }
     } else if (head_ok && tail_ok) {
-      for (int i = lane; i < D; i += LCTR_WAVE) {
+      const int nq4 = (D & 3) == 0 ? (D >> 2) : 0;
+      float4* gV4 = (float4*)&gradV[(size_t)cur * D];
+      float4* acc4 = (float4*)acc;
+      for (int i = lane; i < nq4; i += LCTR_WAVE) {
+        const float4 ac = acc4[i];
+        gV4[i] = make_float4(ac.x * inv_scale, ac.y * inv_scale,
+                             ac.z * inv_scale, ac.w * inv_scale);
+        acc4[i] = make_float4(0.f, 0.f, 0.f, 0.f);
+      }
+      for (int i = 4 * nq4 + lane; i < D; i += LCTR_WAVE) {
         gradV[(size_t)cur * D + i] = acc[i] * inv_scale;
         acc[i] = 0.f;
       }
