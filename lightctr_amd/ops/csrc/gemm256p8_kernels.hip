@@ -104,13 +104,16 @@ __device__ __forceinline__ float p8_act(float v, int act) {
 // boundary) instead of the template's eight; the compiler pipelines the
 // phases freely as in the round-1 kernel. A/B lever for the big square
 // shapes where the 8-barrier variant measured 9% behind round 1.
-template <bool TAIL, bool LITE, bool XCD>
+// DEEPA: A triple-buffered (3 parity slots, 160 KB LDS total) so A can
+// stage TWO tiles ahead like B; the boundary wait loosens from
+// vmcnt(4) to vmcnt(8) (one full tile of loads stays in flight).
+template <bool TAIL, bool LITE, bool XCD, bool DEEPA = false>
 __global__ __launch_bounds__(512, 1) void gemm256p8_bf16_kernel(
     const __bf16* __restrict__ A, const __bf16* __restrict__ Bst,
     const float* __restrict__ bias, float* __restrict__ C,
     __bf16* __restrict__ Cbf, int M, int N, int K, int act) {
-  // LDS: A parity 0 | A parity 1 | B parity 0 | B parity 1 (16384 els each)
-  __shared__ __bf16 smem[4 * 256 * P8_BK];
+  // LDS: A parity slots (2 or 3) then B parity 0/1 (16384 els each)
+  __shared__ __bf16 smem[(DEEPA ? 5 : 4) * 256 * P8_BK];
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int wm = wave >> 2;  // 0..1
@@ -140,12 +143,16 @@ __global__ __launch_bounds__(512, 1) void gemm256p8_bf16_kernel(
 
   __bf16* Abuf0 = smem;
   __bf16* Abuf1 = smem + 16384;
-  __bf16* Bbuf0 = smem + 32768;
-  __bf16* Bbuf1 = smem + 49152;
+  __bf16* Abuf2 = DEEPA ? smem + 2 * 16384 : nullptr;
+  __bf16* Bbuf0 = smem + (DEEPA ? 3 : 2) * 16384;
+  __bf16* Bbuf1 = Bbuf0 + 16384;
 
+#define P8_ABUF(tt)                                                        \
+  (DEEPA ? ((tt) % 3 == 0 ? Abuf0 : ((tt) % 3 == 1 ? Abuf1 : Abuf2))      \
+         : (((tt) & 1) ? Abuf1 : Abuf0))
 #define P8_STAGE_A(tt, h)                                                   \
   do {                                                                      \
-    __bf16* dst_ = (((tt) & 1) ? Abuf1 : Abuf0) + (h) * 8192;               \
+    __bf16* dst_ = P8_ABUF(tt) + (h) * 8192;                                \
     const __bf16* g_ = A + (long)(M0 + (h) * 128) * K + (tt) * P8_BK;       \
     if (!TAIL || (tt) < NT - 1)                                             \
       p8_stage_half(g_, K, dst_);                                           \
@@ -162,24 +169,31 @@ __global__ __launch_bounds__(512, 1) void gemm256p8_bf16_kernel(
       p8_stage_half_tail(g_, K, krem, dst_);                                \
   } while (0)
 
-  // prologue: tile 0 (4 half-tiles) + B(tile 1) two tiles ahead
+  // prologue: tile 0 (+ tile 1's B; DEEPA also tile 1's A) ahead
   P8_STAGE_A(0, 0);
   P8_STAGE_A(0, 1);
   P8_STAGE_B(0, 0);
   P8_STAGE_B(0, 1);
   if (NT > 1) {
+    if (DEEPA) {
+      P8_STAGE_A(1, 0);
+      P8_STAGE_A(1, 1);
+    }
     P8_STAGE_B(1, 0);
     P8_STAGE_B(1, 1);
-    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    if (DEEPA)
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
   } else {
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   }
   __builtin_amdgcn_s_barrier();
 
   for (int t = 0; t < NT; ++t) {
-    __bf16* curA = (t & 1) ? Abuf1 : Abuf0;
+    __bf16* curA = P8_ABUF(t);
     __bf16* curB = (t & 1) ? Bbuf1 : Bbuf0;
-    const bool sA = t + 1 < NT;
+    const bool sA = DEEPA ? (t + 2 < NT) : (t + 1 < NT);
     const bool sB = t + 2 < NT;
 
     // ---- phase 0: b-frags (whole tile) + A quadrant 0 ----
@@ -201,7 +215,7 @@ __global__ __launch_bounds__(512, 1) void gemm256p8_bf16_kernel(
         const int ks = kc * 32 + (lane >> 4) * 8;
         a[fm][kc] = *(const p8bf16x8*)&curA[ra * P8_BK + p8swz(ra, ks)];
       }
-    if (sA) P8_STAGE_A(t + 1, 0);
+    if (sA) P8_STAGE_A(DEEPA ? t + 2 : t + 1, 0);
     if constexpr (!LITE) {
       asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");
       __builtin_amdgcn_s_barrier();
@@ -232,7 +246,7 @@ __global__ __launch_bounds__(512, 1) void gemm256p8_bf16_kernel(
           a[fm][kc] = *(const p8bf16x8*)&curA[ra * P8_BK + p8swz(ra, ks)];
         }
       if (p == 1) {
-        if (sA) P8_STAGE_A(t + 1, 1);
+        if (sA) P8_STAGE_A(DEEPA ? t + 2 : t + 1, 1);
       } else if (p == 2) {
         if (sB) P8_STAGE_B(t + 2, 0);
       } else {
@@ -264,9 +278,14 @@ __global__ __launch_bounds__(512, 1) void gemm256p8_bf16_kernel(
       if (p < 3) {
         if constexpr (!LITE) __builtin_amdgcn_s_barrier();
       } else {
-        // tile boundary: counted drain (leave B(t+2) in flight), then the
-        // barrier that publishes every wave's landed stages
-        if (sB)
+        // tile boundary: counted drain, then the barrier that publishes
+        // every wave's landed stages. DEEPA: A(t+2)+B(t+2) may stay in
+        // flight (vmcnt(8)); else only B(t+2) (vmcnt(4)).
+        if (DEEPA && sA && sB)
+          asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+        else if (!DEEPA && sB)
+          asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        else if (DEEPA && sB)
           asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
         else
           asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
