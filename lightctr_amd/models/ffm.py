@@ -15,6 +15,7 @@ import torch
 
 from ..ops import ffm_ref, fm_ref
 from ..ops._extension import require_hip_ops, sort_ids
+from ..utils.checks import validate_csr_batch
 from ..utils.metrics import auc_score
 
 
@@ -135,6 +136,8 @@ class FFMModel:
     def train_step(self, row_ptr, fields, fids, vals, labels) -> torch.Tensor:
         B = row_ptr.numel() - 1
         scale = 1.0 / B
+        validate_csr_batch(row_ptr, fids, vals, self.h.num_features,
+                           fields, self.h.num_fields)
         if not self._fields_checked:
             # one-time guard: an out-of-range field silently corrupts the
             # [F, nfields, K] indexing on device (costs one sync, once)

@@ -23,6 +23,7 @@ import torch
 
 from ..ops import fm_ref
 from ..ops._extension import require_hip_ops, sort_ids
+from ..utils.checks import validate_csr_batch
 from ..utils.metrics import auc_score
 
 
@@ -107,6 +108,7 @@ class FMModel:
         caller reduces/syncs only when it wants the number."""
         B = row_ptr.numel() - 1
         scale = 1.0 / B
+        validate_csr_batch(row_ptr, fids, vals, self.h.num_features)
         if self._use_hip:
             ops = require_hip_ops()
             pred, sumVX = ops.fm_forward(row_ptr, fids, vals, self.W, self.V)

@@ -412,3 +412,44 @@ def test_failover_checkpoint_restart(tmp_path):
         p.join(timeout=120)
     assert procs[0].exitcode == 0 and procs[1].exitcode == 0, \
         (procs[0].exitcode, procs[1].exitcode)
+
+
+def test_batch_validators(monkeypatch):
+    """LIGHTCTR_CHECK=1 batch validation (§5.2 invariant layer): catches
+    malformed CSR, out-of-range ids/fields, non-finite values."""
+    import importlib
+
+    import lightctr_amd.utils.checks as checks
+
+    monkeypatch.setenv("LIGHTCTR_CHECK", "1")
+    importlib.reload(checks)
+    rp = torch.tensor([0, 2, 4], dtype=torch.int32)
+    fids = torch.tensor([1, 2, 3, 4], dtype=torch.int32)
+    vals = torch.ones(4)
+    checks.validate_csr_batch(rp, fids, vals, 10)  # ok
+    try:
+        checks.validate_csr_batch(rp, fids, vals, 3)  # fid 4 >= 3
+        raise SystemExit("should have raised")
+    except AssertionError:
+        pass
+    bad_vals = vals.clone()
+    bad_vals[1] = float("nan")
+    try:
+        checks.validate_csr_batch(rp, fids, bad_vals, 10)
+        raise SystemExit("should have raised")
+    except AssertionError:
+        pass
+    fields = torch.tensor([0, 1, 5, 1], dtype=torch.int32)
+    try:
+        checks.validate_csr_batch(rp, fids, vals, 10, fields, 4)
+        raise SystemExit("should have raised")
+    except AssertionError:
+        pass
+    checks.validate_finite("state", torch.ones(3))
+    try:
+        checks.validate_finite("state", torch.tensor([1.0, float("inf")]))
+        raise SystemExit("should have raised")
+    except AssertionError:
+        pass
+    monkeypatch.setenv("LIGHTCTR_CHECK", "0")
+    importlib.reload(checks)
