@@ -259,7 +259,7 @@ struct FmOptArgs {
   float q0, q1, q2;  // v_lr, v_eps, v_l2
 };
 
-template <int K>
+template <int K, bool PP = true>
 __global__ void fm_sorted_apply_kernel(
     const int* __restrict__ sorted_fids, const int* __restrict__ perm,
     const float* __restrict__ gw, const float* __restrict__ gv,
@@ -378,13 +378,20 @@ __global__ void fm_sorted_apply_kernel(
       accw += VW[u];                                                       \
     }                                                                      \
   } while (0)
-  if (sb < se) FM_WALK_LOAD(sb, vA, vwA, fA);
-  for (int e = sb; e < se; e += 16) {
-    if (e + 8 < se) FM_WALK_LOAD(e + 8, vB, vwB, fB);
-    FM_WALK_PROC(e, vA, vwA, fA);
-    if (e + 8 >= se) break;
-    if (e + 16 < se) FM_WALK_LOAD(e + 16, vA, vwA, fA);
-    FM_WALK_PROC(e + 8, vB, vwB, fB);
+  if (PP) {
+    if (sb < se) FM_WALK_LOAD(sb, vA, vwA, fA);
+    for (int e = sb; e < se; e += 16) {
+      if (e + 8 < se) FM_WALK_LOAD(e + 8, vB, vwB, fB);
+      FM_WALK_PROC(e, vA, vwA, fA);
+      if (e + 8 >= se) break;
+      if (e + 16 < se) FM_WALK_LOAD(e + 16, vA, vwA, fA);
+      FM_WALK_PROC(e + 8, vB, vwB, fB);
+    }
+  } else {  // round-1 single-buffer walk (within-probe A/B baseline)
+    for (int e = sb; e < se; e += 8) {
+      FM_WALK_LOAD(e, vA, vwA, fA);
+      FM_WALK_PROC(e, vA, vwA, fA);
+    }
   }
 #undef FM_WALK_LOAD
 #undef FM_WALK_PROC
@@ -813,16 +820,26 @@ void fm_sorted_apply_launch(const int* sorted_fids, const int* perm,
         return;
     }
   }
+  // chunk == -3 selects the round-1 single-buffer walk (A/B baseline)
+  const bool pp = chunk != -3;
+  if (chunk == -3) chunk = 0;
   if (chunk <= 0) chunk = 384;  // measured optimum, tools/bench_apply.py
   const int wpb = waves_per_block();
   const int nwaves = (nnz + chunk - 1) / chunk;
   dim3 block(wpb * LCTR_WAVE);
   dim3 grid((nwaves + wpb - 1) / wpb);
   FmOptArgs oa{W, nW, zW, nV, zV, p0, p1, p2, p3, q0, q1, q2};
-  DISPATCH_K(K, hipLaunchKernelGGL((fm_sorted_apply_kernel<KC>), grid, block,
-                                   0, stream, sorted_fids, perm, gw, gv, gradW,
-                                   gradV, touched, nnz, chunk, opt_mode, V,
-                                   oa));
+  if (pp) {
+    DISPATCH_K(K, hipLaunchKernelGGL((fm_sorted_apply_kernel<KC, true>),
+                                     grid, block, 0, stream, sorted_fids,
+                                     perm, gw, gv, gradW, gradV, touched,
+                                     nnz, chunk, opt_mode, V, oa));
+  } else {
+    DISPATCH_K(K, hipLaunchKernelGGL((fm_sorted_apply_kernel<KC, false>),
+                                     grid, block, 0, stream, sorted_fids,
+                                     perm, gw, gv, gradW, gradV, touched,
+                                     nnz, chunk, opt_mode, V, oa));
+  }
 }
 
 void bitmap_compact_launch(unsigned long long* bitmap, int nwords,
