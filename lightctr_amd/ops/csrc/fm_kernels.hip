@@ -341,13 +341,14 @@ __global__ void fm_sorted_apply_kernel(
     }
   };
 
-  // 2-deep batch pipeline (round 2; PMC: VALUBusy 9.6% / MemUnitStalled
-  // 0.6% at ~6.5 resident waves per SIMD — the walk idled on each
-  // batch's gather latency with nothing else in flight; issuing batch
-  // i+1's 8 gathers before consuming batch i keeps one batch of 64 B
-  // lines in flight per subgroup at all times). Named ping-pong buffers
-  // (NOT runtime-indexed arrays — those allocate in scratch, guide
-  // common-mistake 20).
+  // 2-deep batch pipeline — MEASURED NEGATIVE within-probe (210 vs
+  // 220 us at chunk 384): despite the PMC profile (VALUBusy 9.6%,
+  // MemUnitStalled 0.6% — latency idling), prefetching the next 8-entry
+  // batch costs more in register pressure / issue placement than the
+  // extra lines in flight buy at ~6.5 waves/SIMD. Kept selectable
+  // (chunk=-4) as the documented experiment; the single-buffer walk
+  // stays default. Named ping-pong buffers (NOT runtime-indexed arrays
+  // — those allocate in scratch, guide common-mistake 20).
   float vA[8], vwA[8], vB[8], vwB[8];
   int fA[8], fB[8];
 #define FM_WALK_LOAD(e0, VV, VW, FF)                                       \
@@ -820,9 +821,10 @@ void fm_sorted_apply_launch(const int* sorted_fids, const int* perm,
         return;
     }
   }
-  // chunk == -3 selects the round-1 single-buffer walk (A/B baseline)
-  const bool pp = chunk != -3;
-  if (chunk == -3) chunk = 0;
+  // chunk == -4 selects the 2-deep pipelined walk (measured negative;
+  // kept for A/B); default = single-buffer walk
+  const bool pp = chunk == -4;
+  if (chunk == -3 || chunk == -4) chunk = 0;
   if (chunk <= 0) chunk = 384;  // measured optimum, tools/bench_apply.py
   const int wpb = waves_per_block();
   const int nwaves = (nnz + chunk - 1) / chunk;

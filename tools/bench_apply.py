@@ -54,15 +54,15 @@ def main():
     gw_s, gv_s = ops.fm_backward_emit(row_ptr, fids, vals, m.V, sumVX, dpred)
     gw_p, gv_p = ops.fm_backward_emit(row_ptr, fids, vals, m.V, sumVX, dpred,
                                       pos)
-    for chunk in (-3, -2, -1, 256, 384, 512):
+    for chunk in (-4, -3, -2, -1, 256, 384, 512):
         a = t(lambda: ops.fm_sorted_apply(sorted_fids, perm, gw_s, gv_s,
                                           m.gradW, m.gradV, m.touched,
                                           chunk))
         b = t(lambda: ops.fm_sorted_apply(sorted_fids, None, gw_p, gv_p,
                                           m.gradW, m.gradV, m.touched,
                                           chunk))
-        tag = {-1: "segscan", -2: "segscan4",
-               -3: "walk-1buf"}.get(chunk, f"walk{chunk}")
+        tag = {-1: "segscan", -2: "segscan4", -3: "walk-1buf",
+               -4: "walk-pp2"}.get(chunk, f"walk{chunk}")
         print(f"apply {tag:8s}  gather {a:8.1f} us   seq {b:8.1f} us")
 
     # correctness cross-check: gather vs scatter paths agree
