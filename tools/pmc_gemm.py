@@ -23,6 +23,8 @@ def main():
     ap.add_argument("--reps", type=int, default=20)
     ap.add_argument("--transA", type=int, default=0)
     ap.add_argument("--transB", type=int, default=0)
+    ap.add_argument("--wgrad-direct", action="store_true",
+                    help="call gemm_wgrad_bf16 directly (bypass dispatch)")
     args = ap.parse_args()
 
     from lightctr_amd.ops import hip_ops
@@ -34,12 +36,17 @@ def main():
         .to(torch.bfloat16).cuda()
     Bst = (torch.randn(K, N) if tB else torch.randn(N, K)) \
         .to(torch.bfloat16).cuda()
+    if args.wgrad_direct:
+        fn = lambda: hip_ops.gemm_wgrad_bf16(A, Bst, M, N, K)
+    else:
+        fn = lambda: hip_ops.gemm_bf16(A, Bst, None, M, N, K, tA, tB, 0,
+                                       False)
     for _ in range(3):
-        hip_ops.gemm_bf16(A, Bst, None, M, N, K, tA, tB, 0, False)
+        fn()
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(args.reps):
-        hip_ops.gemm_bf16(A, Bst, None, M, N, K, tA, tB, 0, False)
+        fn()
     torch.cuda.synchronize()
     dt = (time.perf_counter() - t0) / args.reps
     print(f"{M}x{N}x{K} tA={tA} tB={tB}: {dt * 1e3:.3f} ms  "
