@@ -262,11 +262,9 @@ void gemm_bf16_launch(const void* A, const void* Bst, const float* bias,
     wrowsum_m1_launch(A, Bst, C, K, N, stream);
     return;
   }
-  if (transA == 1 && transB == 1 && (long)M * N <= 4096 && K >= 8192 &&
-      bias == nullptr && act == 0 && Cbf == nullptr) {
-    small_wgrad_launch(A, Bst, C, M, N, K, stream);
-    return;
-  }
+  // wgrad128 first: it beats small_wgrad even on tiny outputs
+  // (64x16x65536: 65 vs 93 us); small_wgrad keeps the shapes wgrad128
+  // cannot take (M or N not 16-aligned)
   if (gemm_wgrad_eligible(M, N, K, transA, transB) && bias == nullptr &&
       act == 0 && Cbf == nullptr) {
     // K-major x K-major wgrad: linear glds staging + ds_read_b64_tr_b16
@@ -282,6 +280,11 @@ void gemm_bf16_launch(const void* A, const void* Bst, const float* bias,
       gemm_wgrad_bf16_launch(A, Bst, C, M, N, K, stream);
       return;
     }
+  }
+  if (transA == 1 && transB == 1 && (long)M * N <= 4096 && K >= 8192 &&
+      bias == nullptr && act == 0 && Cbf == nullptr) {
+    small_wgrad_launch(A, Bst, C, M, N, K, stream);
+    return;
   }
   if (K == 1 && transA == 0) {
     outer_k1_launch(A, Bst, bias, C, Cbf, M, N, act, stream);
