@@ -710,15 +710,23 @@ __global__ void ffm_blocks_apply_f16_kernel(
     float* __restrict__ gradW, float* __restrict__ gradV,
     unsigned long long* __restrict__ touched, int D, int nnz, int chunk,
     float inv_scale, int opt_mode, float* __restrict__ V, FfmOptArgs oa) {
-  extern __shared__ float lds_acc[];
+  // REGISTER accumulator (round 2): each lane owns quads {lane, lane+64,
+  // ...} of the [D] block — at most 4 float4 for D <= 1024 (the staged
+  // rowemit caps D at ~819). The round-1 LDS accumulator cost a
+  // ds_read_b128 + ds_write_b128 round trip per quad per entry, the
+  // dominant dependency chain of this latency-bound walk (PMC: VALUBusy
+  // 20%, MemUnitStalled 0.4%).
+  constexpr int MAXQ = 4;  // quads per lane -> D <= 4*64*4 = 1024
   const int lane = threadIdx.x & 63;
-  const int wave_in_blk = threadIdx.x >> 6;
-  const int wave = blockIdx.x * (blockDim.x >> 6) + wave_in_blk;
+  const int wave = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   const int base = wave * chunk;
   if (base >= nnz) return;
   const int end = min(base + chunk, nnz);
-  float* acc = &lds_acc[(size_t)wave_in_blk * D];
-  for (int i = lane; i < D; i += LCTR_WAVE) acc[i] = 0.f;
+  const int nq = D >> 2;  // launcher guarantees D % 4 == 0
+
+  float4 acc[MAXQ];
+#pragma unroll
+  for (int t = 0; t < MAXQ; ++t) acc[t] = make_float4(0.f, 0.f, 0.f, 0.f);
 
   int cur = -1;
   bool head_ok = false;
@@ -727,19 +735,18 @@ __global__ void ffm_blocks_apply_f16_kernel(
     if (cur < 0) return;
     const bool tail_ok = tail_e >= nnz || sorted_fids[tail_e] != cur;
     if (FUSED && head_ok && tail_ok) {
-      // exclusive owner: fused V-Adagrad (+W update) in place, float4
-      // RMWs (D % 4 == 0 for every dispatched K; the sparse-optimizer
-      // kernel's measured win on this latency-bound state sweep)
-      const int nq4 = (D & 3) == 0 ? (D >> 2) : 0;
+      // exclusive owner: fused V-Adagrad (+W update) in place
       float4* V4 = (float4*)&V[(size_t)cur * D];
       float4* nV4 = (float4*)&oa.nV[(size_t)cur * D];
-      float4* acc4 = (float4*)acc;
       ffm_bf16x4* Vh4 =
           oa.Vh ? (ffm_bf16x4*)&oa.Vh[(size_t)cur * D] : nullptr;
-      for (int i = lane; i < nq4; i += LCTR_WAVE) {
-        const float4 ac = acc4[i];
-        float4 v = V4[i];
-        float4 a = nV4[i];
+#pragma unroll
+      for (int t = 0; t < MAXQ; ++t) {
+        const int q = t * LCTR_WAVE + lane;
+        if (q >= nq) break;
+        const float4 ac = acc[t];
+        float4 v = V4[q];
+        float4 a = nV4[q];
         const float gx = ac.x * inv_scale + oa.q2 * v.x;
         const float gy = ac.y * inv_scale + oa.q2 * v.y;
         const float gz = ac.z * inv_scale + oa.q2 * v.z;
@@ -752,27 +759,17 @@ __global__ void ffm_blocks_apply_f16_kernel(
         v.y -= oa.q0 * gy * __frsqrt_rn(a.y + oa.q1);
         v.z -= oa.q0 * gz * __frsqrt_rn(a.z + oa.q1);
         v.w -= oa.q0 * gw2 * __frsqrt_rn(a.w + oa.q1);
-        nV4[i] = a;
-        V4[i] = v;
+        nV4[q] = a;
+        V4[q] = v;
         if (Vh4) {
           ffm_bf16x4 h;
           h.x = (__bf16)v.x;
           h.y = (__bf16)v.y;
           h.z = (__bf16)v.z;
           h.w = (__bf16)v.w;
-          Vh4[i] = h;
+          Vh4[q] = h;
         }
-        acc4[i] = make_float4(0.f, 0.f, 0.f, 0.f);
-      }
-      for (int i = 4 * nq4 + lane; i < D; i += LCTR_WAVE) {  // safety tail
-        const size_t off = (size_t)cur * D + i;
-        const float g = acc[i] * inv_scale + oa.q2 * V[off];
-        const float a = oa.nV[off] + g * g;
-        oa.nV[off] = a;
-        const float vnew = V[off] - oa.q0 * g * __frsqrt_rn(a + oa.q1);
-        V[off] = vnew;
-        if (oa.Vh) oa.Vh[off] = (__bf16)vnew;
-        acc[i] = 0.f;
+        acc[t] = make_float4(0.f, 0.f, 0.f, 0.f);
       }
       if (lane == 0) {
         if (opt_mode == 3) {
@@ -786,25 +783,30 @@ __global__ void ffm_blocks_apply_f16_kernel(
         }
       }
     } else if (head_ok && tail_ok) {
-      const int nq4 = (D & 3) == 0 ? (D >> 2) : 0;
       float4* gV4 = (float4*)&gradV[(size_t)cur * D];
-      float4* acc4 = (float4*)acc;
-      for (int i = lane; i < nq4; i += LCTR_WAVE) {
-        const float4 ac = acc4[i];
-        gV4[i] = make_float4(ac.x * inv_scale, ac.y * inv_scale,
+#pragma unroll
+      for (int t = 0; t < MAXQ; ++t) {
+        const int q = t * LCTR_WAVE + lane;
+        if (q >= nq) break;
+        const float4 ac = acc[t];
+        gV4[q] = make_float4(ac.x * inv_scale, ac.y * inv_scale,
                              ac.z * inv_scale, ac.w * inv_scale);
-        acc4[i] = make_float4(0.f, 0.f, 0.f, 0.f);
-      }
-      for (int i = 4 * nq4 + lane; i < D; i += LCTR_WAVE) {
-        gradV[(size_t)cur * D + i] = acc[i] * inv_scale;
-        acc[i] = 0.f;
+        acc[t] = make_float4(0.f, 0.f, 0.f, 0.f);
       }
       if (lane == 0) gradW[cur] = accw;
     } else {
-      for (int i = lane; i < D; i += LCTR_WAVE) {
-        if (acc[i] != 0.f)
-          atomicAdd(&gradV[(size_t)cur * D + i], acc[i] * inv_scale);
-        acc[i] = 0.f;
+#pragma unroll
+      for (int t = 0; t < MAXQ; ++t) {
+        const int q = t * LCTR_WAVE + lane;
+        if (q >= nq) break;
+        const float4 ac = acc[t];
+        if (ac.x != 0.f || ac.y != 0.f || ac.z != 0.f || ac.w != 0.f) {
+          atomicAdd(&gradV[(size_t)cur * D + 4 * q + 0], ac.x * inv_scale);
+          atomicAdd(&gradV[(size_t)cur * D + 4 * q + 1], ac.y * inv_scale);
+          atomicAdd(&gradV[(size_t)cur * D + 4 * q + 2], ac.z * inv_scale);
+          atomicAdd(&gradV[(size_t)cur * D + 4 * q + 3], ac.w * inv_scale);
+        }
+        acc[t] = make_float4(0.f, 0.f, 0.f, 0.f);
       }
       if (lane == 0) {
         atomicAdd(&gradW[cur], accw);
@@ -819,33 +821,29 @@ __global__ void ffm_blocks_apply_f16_kernel(
       cur = fid;
       accw = 0.f;
       head_ok = (e == 0 || sorted_fids[e - 1] != fid);
-      // fused mode: interior runs bypass the bitmap entirely; the bit is
-      // only needed when the run's gradient lands in the slabs. An
-      // interior run is detectable at its head (tail unknown yet), so in
-      // fused mode the spanning flush sets the bit instead.
+      // fused mode: interior runs bypass the bitmap entirely (the
+      // spanning flush sets the bit instead — see flush above)
       if (lane == 0 && head_ok && !FUSED)
         atomicOr(&touched[fid >> 6], 1ull << (fid & 63));
     }
     const long p = (long)perm[e];
-    // quad loads (dwordx2): D % 4 == 0 whenever K % 4 == 0 (all dispatched
-    // K); generic tail kept for safety.
     const uint2* gb4 = (const uint2*)&gblocks[(size_t)p * D];
-    const int nq = D / 4;
-    for (int i = lane; i < nq; i += LCTR_WAVE) {
+#pragma unroll
+    for (int t = 0; t < MAXQ; ++t) {
+      const int q = t * LCTR_WAVE + lane;
+      if (q >= nq) break;
       union {
         uint2 u2;
         __half2 h[2];
       } in;
-      in.u2 = gb4[i];
+      in.u2 = gb4[q];
       const float2 a = __half22float2(in.h[0]);
       const float2 b = __half22float2(in.h[1]);
-      acc[4 * i] += a.x;
-      acc[4 * i + 1] += a.y;
-      acc[4 * i + 2] += b.x;
-      acc[4 * i + 3] += b.y;
+      acc[t].x += a.x;
+      acc[t].y += a.y;
+      acc[t].z += b.x;
+      acc[t].w += b.y;
     }
-    for (int i = 4 * nq + lane; i < D; i += LCTR_WAVE)
-      acc[i] += (float)gblocks[(size_t)p * D + i];
     if (lane == 0) accw += gw[p];
   }
   flush(end);
@@ -926,20 +924,23 @@ void ffm_blocks_apply_f16_launch(const int* sorted_fids, const int* perm,
     const char* e = getenv("LCTR_FFM_APPLY_CHUNK");
     return e ? atoi(e) : 64;
   }();
+  if (D % 4 != 0 || D > 1024) {
+    fprintf(stderr, "ffm_blocks_apply_f16: D=%d unsupported\n", D);
+    abort();
+  }
   const int wpb = 4;
   const int nwaves = (nnz + chunk - 1) / chunk;
   dim3 block(wpb * LCTR_WAVE);
   dim3 grid((nwaves + wpb - 1) / wpb);
-  const size_t lds = (size_t)wpb * D * sizeof(float);
   FfmOptArgs oa{W, nW, zW, nV, (__bf16*)Vh, p0, p1, p2, p3, q0, q1, q2};
   if (opt_mode != 0)
-    hipLaunchKernelGGL((ffm_blocks_apply_f16_kernel<true>), grid, block, lds,
+    hipLaunchKernelGGL((ffm_blocks_apply_f16_kernel<true>), grid, block, 0,
                        stream, sorted_fids, perm, (const _Float16*)gblocks,
                        gw, gradW, gradV, touched, D, nnz, chunk, inv_scale,
                        opt_mode, V, oa);
   else
     hipLaunchKernelGGL((ffm_blocks_apply_f16_kernel<false>), grid, block,
-                       lds, stream, sorted_fids, perm,
+                       0, stream, sorted_fids, perm,
                        (const _Float16*)gblocks, gw, gradW, gradV, touched,
                        D, nnz, chunk, inv_scale, opt_mode, V, oa);
 }
