@@ -83,7 +83,15 @@ class FFMModel:
         #            bandwidth bound per the round-1 PMC/ISA analysis)
         #   blocks — wave/entry fp32 emit + block reduce, 5.6 ms
         #   atomic — naive scatter (hot-feature serialization)
-        self.backward_mode = "rowemit"
+        # rowemit needs the staged-emit LDS footprint to fit (mirrors
+        # ffm_staged_eligible in ffm_kernels.hip, maxn=40 / 8 waves) and
+        # the apply's register-accumulator cap (D <= 1024); larger
+        # field*factor products fall back to the sorted walk
+        maxn, D = 40, nf * K
+        lds_bytes = (8 * nf * (K + 1) * 4 + maxn * D * 2 + maxn * 4
+                     + nf * 4 + 4)
+        self._rowemit_ok = lds_bytes <= (64 << 10) and D <= 1024
+        self.backward_mode = "rowemit" if self._rowemit_ok else "sorted"
         # fuse the sparse optimizer into the apply's interior-run flush
         # (features whose sorted segment is wholly owned by one chunk
         # skip the gradV slab + bitmap + separate optimizer kernel);

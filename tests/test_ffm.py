@@ -457,3 +457,25 @@ def test_ffm_fused_apply_matches_two_phase():
         assert torch.allclose(a.nV, b.nV, atol=1e-4)
         if dtype == "bf16":
             assert torch.allclose(a.Vh.float(), a.V, atol=1e-2)
+
+
+@pytest.mark.gpu
+def test_ffm_large_field_product_falls_back():
+    """nfields*K beyond the staged-emit/register caps must select the
+    sorted backward automatically (and still train)."""
+    from lightctr_amd.models.ffm import FFMHyper, FFMModel
+
+    m = FFMModel(FFMHyper(num_features=1 << 12, num_fields=80, k=16),
+                 device="cuda:0")
+    assert m.backward_mode == "sorted"
+    g = torch.Generator().manual_seed(3)
+    B, nf = 64, 80
+    fields = torch.arange(nf, dtype=torch.int32).repeat(B).cuda()
+    fids = torch.randint(0, 1 << 12, (B * nf,), generator=g,
+                         dtype=torch.int32).cuda()
+    vals = torch.ones(B * nf).cuda()
+    row_ptr = (torch.arange(B + 1, dtype=torch.int32) * nf).cuda()
+    labels = (torch.rand(B, generator=g) > 0.5).float().cuda()
+    for _ in range(3):
+        loss = m.train_step(row_ptr, fields, fids, vals, labels)
+        assert torch.isfinite(loss).all()
