@@ -49,6 +49,12 @@ class ShardedFFMModel:
         self.uniq = torch.zeros(self.F_local, dtype=torch.int32, device=dev)
         self.count = torch.zeros(1, dtype=torch.int32, device=dev)
         self._use_hip = dev.type == "cuda"
+        # rowemit caps (mirror models/ffm.py): fall back to the sorted
+        # walk when the staged emit / register accumulator cannot fit
+        maxn, D = 40, nf * K
+        lds_bytes = (8 * nf * (K + 1) * 4 + maxn * D * 2 + maxn * 4
+                     + nf * 4 + 4)
+        self._rowemit_ok = lds_bytes <= (64 << 10) and D <= 1024
         if self._use_hip:
             require_hip_ops()
 
@@ -109,16 +115,23 @@ class ShardedFFMModel:
             scratch = torch.zeros((U + 63) // 64, dtype=torch.int64,
                                   device=self.device)
             sorted_l, perm = sort_ids(fids_local, U)
-            # rowemit backward (round-2 default, measured 2.82 vs 3.91 ms
-            # single-GPU); power-of-two scale keeps the fp16 blocks away
-            # from denormal flush at large B*world
-            bscale = float(1 << min(24, max(0, (B * world).bit_length()
-                                           - 1)))
-            gw, gblocks = ops.ffm_row_emit(row_ptr, fields, fids_local,
-                                           vals, Vc, dpred, scale=bscale)
-            ops.ffm_blocks_apply_f16(sorted_l, perm, gblocks, gw, gWl,
-                                     gVl.view(U, -1), scratch,
-                                     inv_scale=1.0 / bscale)
+            if self._rowemit_ok:
+                # rowemit backward (round-2 default, 2.82 vs 3.91 ms
+                # single-GPU); power-of-two scale keeps the fp16 blocks
+                # away from denormal flush at large B*world
+                bscale = float(1 << min(24, max(0, (B * world).bit_length()
+                                               - 1)))
+                gw, gblocks = ops.ffm_row_emit(row_ptr, fields, fids_local,
+                                               vals, Vc, dpred,
+                                               scale=bscale)
+                ops.ffm_blocks_apply_f16(sorted_l, perm, gblocks, gw, gWl,
+                                         gVl.view(U, -1), scratch,
+                                         inv_scale=1.0 / bscale)
+            else:
+                row_of_entry = ops.row_index(row_ptr, fids.numel())
+                ops.ffm_sorted_backward(sorted_l, perm, row_of_entry,
+                                        row_ptr, fields, fids_local, vals,
+                                        Vl, dpred, gWl, gVl, scratch)
         else:
             pred = ffm_ref.ffm_forward_ref(row_ptr, fields, fids_local,
                                            vals, Wl, Vl)
