@@ -814,38 +814,62 @@ __global__ void ffm_blocks_apply_f16_kernel(
       }
     }
   };
-  for (int e = base; e < end; ++e) {
-    const int fid = sorted_fids[e];
-    if (fid != cur) {
-      flush(e);
-      cur = fid;
-      accw = 0.f;
-      head_ok = (e == 0 || sorted_fids[e - 1] != fid);
-      // fused mode: interior runs bypass the bitmap entirely (the
-      // spanning flush sets the bit instead — see flush above)
-      if (lane == 0 && head_ok && !FUSED)
-        atomicOr(&touched[fid >> 6], 1ull << (fid & 63));
-    }
-    const long p = (long)perm[e];
-    const uint2* gb4 = (const uint2*)&gblocks[(size_t)p * D];
-#pragma unroll
-    for (int t = 0; t < MAXQ; ++t) {
-      const int q = t * LCTR_WAVE + lane;
-      if (q >= nq) break;
-      union {
-        uint2 u2;
-        __half2 h[2];
-      } in;
-      in.u2 = gb4[q];
-      const float2 a = __half22float2(in.h[0]);
-      const float2 b = __half22float2(in.h[1]);
-      acc[t].x += a.x;
-      acc[t].y += a.y;
-      acc[t].z += b.x;
-      acc[t].w += b.y;
-    }
-    if (lane == 0) accw += gw[p];
+  // 2-deep entry pipeline: entry e+1's quad loads (scattered ~624 B)
+  // are issued before entry e is consumed, so each wave keeps one
+  // entry's lines in flight across the run-detection/flush serial
+  // chain. (The FM walk is already 8-entry batched; this kernel had no
+  // lookahead at all.) Named ping-pong registers.
+  uint2 ldA[MAXQ], ldB[MAXQ];
+  float gwA = 0.f, gwB = 0.f;
+#define FFM_APPLY_LOAD(e0, LD, GW)                                         \
+  do {                                                                     \
+    const long p_ = (long)perm[e0];                                        \
+    const uint2* gb4_ = (const uint2*)&gblocks[(size_t)p_ * D];            \
+    _Pragma("unroll") for (int t = 0; t < MAXQ; ++t) {                     \
+      const int q_ = t * LCTR_WAVE + lane;                                 \
+      if (q_ >= nq) break;                                                 \
+      LD[t] = gb4_[q_];                                                    \
+    }                                                                      \
+    GW = (lane == 0) ? gw[p_] : 0.f;                                       \
+  } while (0)
+#define FFM_APPLY_PROC(e0, LD, GW)                                         \
+  do {                                                                     \
+    const int fid_ = sorted_fids[e0];                                      \
+    if (fid_ != cur) {                                                     \
+      flush(e0);                                                           \
+      cur = fid_;                                                          \
+      accw = 0.f;                                                          \
+      head_ok = ((e0) == 0 || sorted_fids[(e0) - 1] != fid_);              \
+      if (lane == 0 && head_ok && !FUSED)                                  \
+        atomicOr(&touched[fid_ >> 6], 1ull << (fid_ & 63));                \
+    }                                                                      \
+    _Pragma("unroll") for (int t = 0; t < MAXQ; ++t) {                     \
+      const int q_ = t * LCTR_WAVE + lane;                                 \
+      if (q_ >= nq) break;                                                 \
+      union {                                                              \
+        uint2 u2;                                                          \
+        __half2 h[2];                                                      \
+      } in_;                                                               \
+      in_.u2 = LD[t];                                                      \
+      const float2 a_ = __half22float2(in_.h[0]);                          \
+      const float2 b_ = __half22float2(in_.h[1]);                          \
+      acc[t].x += a_.x;                                                    \
+      acc[t].y += a_.y;                                                    \
+      acc[t].z += b_.x;                                                    \
+      acc[t].w += b_.y;                                                    \
+    }                                                                      \
+    accw += GW;                                                            \
+  } while (0)
+  if (base < end) FFM_APPLY_LOAD(base, ldA, gwA);
+  for (int e = base; e < end; e += 2) {
+    if (e + 1 < end) FFM_APPLY_LOAD(e + 1, ldB, gwB);
+    FFM_APPLY_PROC(e, ldA, gwA);
+    if (e + 1 >= end) break;
+    if (e + 2 < end) FFM_APPLY_LOAD(e + 2, ldA, gwA);
+    FFM_APPLY_PROC(e + 1, ldB, gwB);
   }
+#undef FFM_APPLY_LOAD
+#undef FFM_APPLY_PROC
   flush(end);
 }
 
