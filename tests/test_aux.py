@@ -453,3 +453,16 @@ def test_batch_validators(monkeypatch):
         pass
     monkeypatch.setenv("LIGHTCTR_CHECK", "0")
     importlib.reload(checks)
+
+
+def test_tools_importable():
+    """Every tools/ script parses and imports its module-level deps (the
+    GPU-only main()s are not run)."""
+    import ast
+    import pathlib
+
+    for f in sorted(pathlib.Path("tools").glob("*.py")):
+        src = f.read_text()
+        tree = ast.parse(src)  # syntax
+        assert any(isinstance(n, ast.FunctionDef) and n.name == "main"
+                   for n in ast.walk(tree)) or "proc_" in f.name, f.name
