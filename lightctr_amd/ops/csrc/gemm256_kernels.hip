@@ -70,7 +70,42 @@ __device__ __forceinline__ float act_apply256(float v, int act) {
 // BURST: issue all four of tile t+1's half-tile stages at phase 0 so the
 // tile-end vmcnt(0) drain waits on loads ~3 phases old instead of on the
 // half-tile just issued in phase 3 (A/B via LCTR_GEMM_BURST=1).
-template <bool TAIL, bool BURST>
+//
+// PIPE (LCTR_GEMM_APIPE=1): software-pipeline the per-phase A-fragment
+// ds_reads one phase ahead. The baseline schedule (ISA-audited) emits
+// `ds_read_b128 x4; s_waitcnt lgkmcnt(0); v_mfma x16` per phase — the
+// LDS latency of the 4 reads is exposed on every phase because the
+// s_setprio intrinsics and the staging branches pin the reads to their
+// own phase. Issuing phase p+1's reads into named ping-pong buffers
+// BEFORE phase p's MFMA cluster lets the compiler's waitcnt pass emit a
+// counted `lgkmcnt(4)` instead (4 newer reads left in flight), hiding
+// the read latency under the previous 16-MFMA cluster.
+
+// phase-p A-fragment read into a named buffer (compile-time P keeps the
+// acc[] indices constant — runtime indexing would spill to scratch)
+#define G256_READ_A(P, AARR)                                             \
+  _Pragma("unroll") for (int fm = 0; fm < 2; ++fm)                       \
+  _Pragma("unroll") for (int kc = 0; kc < 2; ++kc) {                     \
+    const int ra = wm * 128 + ((P)*2 + fm) * 16 + (lane & 15);           \
+    const int ks = kc * 32 + (lane >> 4) * 8;                            \
+    AARR[fm][kc] = *(const bf16x8*)&curA[ra * G256_BK + swz256(ra, ks)]; \
+  }
+
+#define G256_MFMA(P, AARR)                                               \
+  __builtin_amdgcn_s_setprio(1);                                         \
+  _Pragma("unroll") for (int fm = 0; fm < 2; ++fm)                       \
+  _Pragma("unroll") for (int fn = 0; fn < 4; ++fn)                       \
+  _Pragma("unroll") for (int kc = 0; kc < 2; ++kc) acc[(P)*2 + fm][fn] = \
+      __builtin_amdgcn_mfma_f32_16x16x32_bf16(AARR[fm][kc], b[fn][kc],   \
+                                              acc[(P)*2 + fm][fn], 0, 0, \
+                                              0);                        \
+  __builtin_amdgcn_s_setprio(0);
+
+// PIPE=2 additionally makes the per-phase staging branch-free on the
+// full-tile path: the last tile re-stages itself (k0 clamped) instead
+// of branching on `more`, so the whole tile body is one basic block and
+// the waitcnt pass can count across all four phases.
+template <bool TAIL, bool BURST, int PIPE>
 __global__ __launch_bounds__(512, 1) void gemm256_bf16_kernel(
     const __bf16* __restrict__ A, const __bf16* __restrict__ Bst,
     const float* __restrict__ bias, float* __restrict__ C,
@@ -122,40 +157,98 @@ __global__ __launch_bounds__(512, 1) void gemm256_bf16_kernel(
         const int ks = kc * 32 + (lane >> 4) * 8;
         b[fn][kc] = *(const bf16x8*)&curB[rb * G256_BK + swz256(rb, ks)];
       }
-    // 4 phases: quadrant p = fm in [2p, 2p+2), all fn, full K-tile
-#pragma unroll
-    for (int p = 0; p < 4; ++p) {
-      if (more) {  // prefetch half-tile(s) of tile t+1
-        const bool next_full = !TAIL || (k0 + G256_BK <= K);
-        if (next_full) {
-          if (BURST) {
-            if (p == 0) {
-              stage_half_512(A + (long)M0 * K + k0, K, nxtA);
-              stage_half_512(A + (long)(M0 + 128) * K + k0, K,
-                             nxtA + 128 * G256_BK);
-              stage_half_512(Bst + (long)N0 * K + k0, K, nxtB);
-              stage_half_512(Bst + (long)(N0 + 128) * K + k0, K,
-                             nxtB + 128 * G256_BK);
-            }
-          } else if (p == 0) {
+    // prefetch half-tile(s) of tile t+1 for phase p (glds issue keeps
+    // the one-stage-per-phase cadence; BURST front-loads all four)
+    auto stage_phase = [&](int p) {
+      if (!more) return;
+      const bool next_full = !TAIL || (k0 + G256_BK <= K);
+      if (next_full) {
+        if (BURST) {
+          if (p == 0) {
             stage_half_512(A + (long)M0 * K + k0, K, nxtA);
-          } else if (p == 1) {
             stage_half_512(A + (long)(M0 + 128) * K + k0, K,
                            nxtA + 128 * G256_BK);
-          } else if (p == 2) {
             stage_half_512(Bst + (long)N0 * K + k0, K, nxtB);
-          } else {
             stage_half_512(Bst + (long)(N0 + 128) * K + k0, K,
                            nxtB + 128 * G256_BK);
           }
-        } else {  // K tail: zero-filled scalar staging covers all 256
-                  // rows per call — phases 0 (A) and 2 (B) only
-          if (p == 0)
-            stage_tail_512(A + (long)M0 * K, K, k0, K, nxtA);
-          else if (p == 2)
-            stage_tail_512(Bst + (long)N0 * K, K, k0, K, nxtB);
+        } else if (p == 0) {
+          stage_half_512(A + (long)M0 * K + k0, K, nxtA);
+        } else if (p == 1) {
+          stage_half_512(A + (long)(M0 + 128) * K + k0, K,
+                         nxtA + 128 * G256_BK);
+        } else if (p == 2) {
+          stage_half_512(Bst + (long)N0 * K + k0, K, nxtB);
+        } else {
+          stage_half_512(Bst + (long)(N0 + 128) * K + k0, K,
+                         nxtB + 128 * G256_BK);
         }
+      } else {  // K tail: zero-filled scalar staging covers all 256
+                // rows per call — phases 0 (A) and 2 (B) only
+        if (p == 0)
+          stage_tail_512(A + (long)M0 * K, K, k0, K, nxtA);
+        else if (p == 2)
+          stage_tail_512(Bst + (long)N0 * K, K, k0, K, nxtB);
       }
+    };
+    // branch-free staging (PIPE=2, full tiles only): clamp so the last
+    // tile harmlessly re-stages itself into the unused nxt buffer
+    const long k0c = (!TAIL && PIPE == 2)
+                         ? (more ? (long)k0 : (long)k0 - G256_BK)
+                         : (long)k0;
+    auto stage_flat = [&](int p) {
+      if (p == 0) stage_half_512(A + (long)M0 * K + k0c, K, nxtA);
+      else if (p == 1)
+        stage_half_512(A + (long)(M0 + 128) * K + k0c, K,
+                       nxtA + 128 * G256_BK);
+      else if (p == 2)
+        stage_half_512(Bst + (long)N0 * K + k0c, K, nxtB);
+      else
+        stage_half_512(Bst + (long)(N0 + 128) * K + k0c, K,
+                       nxtB + 128 * G256_BK);
+    };
+    if (PIPE == 2 && !TAIL) {
+      bf16x8 aA[2][2], aB[2][2];
+      G256_READ_A(0, aA);
+      stage_flat(0);
+      G256_READ_A(1, aB);
+      G256_MFMA(0, aA);
+      stage_flat(1);
+      G256_READ_A(2, aA);
+      G256_MFMA(1, aB);
+      stage_flat(2);
+      G256_READ_A(3, aB);
+      G256_MFMA(2, aA);
+      stage_flat(3);
+      G256_MFMA(3, aB);
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
+      cur ^= 1;
+      continue;
+    }
+    if (PIPE) {
+      bf16x8 aA[2][2], aB[2][2];
+      G256_READ_A(0, aA);
+      stage_phase(0);
+      G256_READ_A(1, aB);  // in flight under phase-0 MFMAs
+      G256_MFMA(0, aA);
+      stage_phase(1);
+      G256_READ_A(2, aA);
+      G256_MFMA(1, aB);
+      stage_phase(2);
+      G256_READ_A(3, aB);
+      G256_MFMA(2, aA);
+      stage_phase(3);
+      G256_MFMA(3, aB);
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
+      cur ^= 1;
+      continue;
+    }
+    // 4 phases: quadrant p = fm in [2p, 2p+2), all fn, full K-tile
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      stage_phase(p);
       bf16x8 a[2][2];
 #pragma unroll
       for (int fm = 0; fm < 2; ++fm)
@@ -217,25 +310,35 @@ void gemm256_bf16_launch(const void* A, const void* Bst, const float* bias,
     const char* e = getenv("LCTR_GEMM_BURST");
     return e && e[0] == '1';
   }();
+  // LCTR_GEMM_APIPE=1|2 selects the phase-pipelined A-read schedule
+  // (2 = also branch-free staging; re-read per launch so in-process
+  // A/B sweeps can toggle it)
+  const char* ep = getenv("LCTR_GEMM_APIPE");
+  const int pipe = ep ? (ep[0] == '2' ? 2 : (ep[0] == '1' ? 1 : 0)) : 0;
+#define G256_LAUNCH(T, B, P)                                               \
+  hipLaunchKernelGGL((gemm256_bf16_kernel<T, B, P>), grid, block, 0,       \
+                     stream, (const __bf16*)A, (const __bf16*)Bst, bias,   \
+                     C, (__bf16*)Cbf, M, N, K, act)
   if (K % G256_BK == 0) {
-    if (burst)
-      hipLaunchKernelGGL((gemm256_bf16_kernel<false, true>), grid, block, 0,
-                         stream, (const __bf16*)A, (const __bf16*)Bst, bias,
-                         C, (__bf16*)Cbf, M, N, K, act);
-    else
-      hipLaunchKernelGGL((gemm256_bf16_kernel<false, false>), grid, block, 0,
-                         stream, (const __bf16*)A, (const __bf16*)Bst, bias,
-                         C, (__bf16*)Cbf, M, N, K, act);
+    if (burst) {
+      if (pipe == 2) G256_LAUNCH(false, true, 2);
+      else if (pipe == 1) G256_LAUNCH(false, true, 1);
+      else G256_LAUNCH(false, true, 0);
+    } else {
+      if (pipe == 2) G256_LAUNCH(false, false, 2);
+      else if (pipe == 1) G256_LAUNCH(false, false, 1);
+      else G256_LAUNCH(false, false, 0);
+    }
   } else {
-    if (burst)
-      hipLaunchKernelGGL((gemm256_bf16_kernel<true, true>), grid, block, 0,
-                         stream, (const __bf16*)A, (const __bf16*)Bst, bias,
-                         C, (__bf16*)Cbf, M, N, K, act);
-    else
-      hipLaunchKernelGGL((gemm256_bf16_kernel<true, false>), grid, block, 0,
-                         stream, (const __bf16*)A, (const __bf16*)Bst, bias,
-                         C, (__bf16*)Cbf, M, N, K, act);
+    if (burst) {
+      if (pipe) G256_LAUNCH(true, true, 1);
+      else G256_LAUNCH(true, true, 0);
+    } else {
+      if (pipe) G256_LAUNCH(true, false, 1);
+      else G256_LAUNCH(true, false, 0);
+    }
   }
+#undef G256_LAUNCH
 }
 
 }  // namespace lightctr
