@@ -310,11 +310,15 @@ void gemm256_bf16_launch(const void* A, const void* Bst, const float* bias,
     const char* e = getenv("LCTR_GEMM_BURST");
     return e && e[0] == '1';
   }();
-  // LCTR_GEMM_APIPE=1|2 selects the phase-pipelined A-read schedule
-  // (2 = also branch-free staging; re-read per launch so in-process
-  // A/B sweeps can toggle it)
+  // LCTR_GEMM_APIPE=0|1|2 selects the per-phase A-read schedule
+  // (2 = pipelined + branch-free staging; re-read per launch so
+  // in-process A/B sweeps can toggle it). Default 2 for full tiles:
+  // within-process A/B (profiles/r2_09_gemm_apipe.txt) shows +2.3% at
+  // 8192^3 (1233 vs 1208 TF, 8/8 passes ahead) and no regression at
+  // 4096^3. TAIL shapes default to the round-1 schedule (rarely hit —
+  // K%64!=0 dispatches to the p8 kernel upstream).
   const char* ep = getenv("LCTR_GEMM_APIPE");
-  const int pipe = ep ? (ep[0] == '2' ? 2 : (ep[0] == '1' ? 1 : 0)) : 0;
+  const int pipe = ep ? (ep[0] == '2' ? 2 : (ep[0] == '1' ? 1 : 0)) : 2;
 #define G256_LAUNCH(T, B, P)                                               \
   hipLaunchKernelGGL((gemm256_bf16_kernel<T, B, P>), grid, block, 0,       \
                      stream, (const __bf16*)A, (const __bf16*)Bst, bias,   \
@@ -330,11 +334,12 @@ void gemm256_bf16_launch(const void* A, const void* Bst, const float* bias,
       else G256_LAUNCH(false, false, 0);
     }
   } else {
+    const bool tail_pipe = ep && pipe;  // only when explicitly requested
     if (burst) {
-      if (pipe) G256_LAUNCH(true, true, 1);
+      if (tail_pipe) G256_LAUNCH(true, true, 1);
       else G256_LAUNCH(true, true, 0);
     } else {
-      if (pipe) G256_LAUNCH(true, false, 1);
+      if (tail_pipe) G256_LAUNCH(true, false, 1);
       else G256_LAUNCH(true, false, 0);
     }
   }

@@ -107,7 +107,16 @@ __device__ __forceinline__ float p8_act(float v, int act) {
 // DEEPA: A triple-buffered (3 parity slots, 160 KB LDS total) so A can
 // stage TWO tiles ahead like B; the boundary wait loosens from
 // vmcnt(4) to vmcnt(8) (one full tile of loads stays in flight).
-template <bool TAIL, bool LITE, bool XCD, bool DEEPA = false>
+// APIPE (LCTR_GEMM_P8_APIPE=1, LITE only): phase-pipeline the per-phase
+// A-fragment ds_reads one MFMA cluster ahead through two named register
+// buffers, same transform as gemm256's LCTR_GEMM_APIPE (which measured
+// +2.3% @8192^3 — profiles/r2_09_gemm_apipe.txt). The a-reads target
+// curA, which no staging touches during the tile, so issuing them
+// before the previous phase's MFMA cluster is hazard-free; the LITE
+// mid-barrier constraint (B parity-slot overwrite) only orders b-frag
+// reads vs phase-2 staging and is preserved.
+template <bool TAIL, bool LITE, bool XCD, bool DEEPA = false,
+          bool APIPE = false>
 __global__ __launch_bounds__(512, 1) void gemm256p8_bf16_kernel(
     const __bf16* __restrict__ A, const __bf16* __restrict__ Bst,
     const float* __restrict__ bias, float* __restrict__ C,
@@ -216,6 +225,55 @@ __global__ __launch_bounds__(512, 1) void gemm256p8_bf16_kernel(
         a[fm][kc] = *(const p8bf16x8*)&curA[ra * P8_BK + p8swz(ra, ks)];
       }
     if (sA) P8_STAGE_A(DEEPA ? t + 2 : t + 1, 0);
+
+#define P8_READ_A_PH(P, ARR)                                             \
+  _Pragma("unroll") for (int fm = 0; fm < 2; ++fm)                       \
+  _Pragma("unroll") for (int kc = 0; kc < 2; ++kc) {                     \
+    const int ra = wm * 128 + ((P)*2 + fm) * 16 + (lane & 15);           \
+    const int ks = kc * 32 + (lane >> 4) * 8;                            \
+    ARR[fm][kc] = *(const p8bf16x8*)&curA[ra * P8_BK + p8swz(ra, ks)];   \
+  }
+#define P8_MFMA_PH(P, ARR)                                               \
+  __builtin_amdgcn_s_setprio(1);                                         \
+  _Pragma("unroll") for (int fm = 0; fm < 2; ++fm)                       \
+  _Pragma("unroll") for (int fn = 0; fn < 4; ++fn)                       \
+  _Pragma("unroll") for (int kc = 0; kc < 2; ++kc) acc[(P)*2 + fm][fn] = \
+      __builtin_amdgcn_mfma_f32_16x16x32_bf16(ARR[fm][kc], b[fn][kc],    \
+                                              acc[(P)*2 + fm][fn], 0, 0, \
+                                              0);                        \
+  __builtin_amdgcn_s_setprio(0);
+
+    if constexpr (APIPE && LITE) {
+      p8bf16x8 a2[2][2];
+      P8_READ_A_PH(1, a2);  // in flight under phase-0 MFMAs
+      P8_MFMA_PH(0, a);
+      if (sA) P8_STAGE_A(DEEPA ? t + 2 : t + 1, 1);
+      P8_READ_A_PH(2, a);
+      P8_MFMA_PH(1, a2);
+      // LITE mid-barrier: every wave is past its phase-0 MFMA issue
+      // (b-frag registers populated), so phase 2 may overwrite the
+      // B(t) parity slot with B(t+2)
+      asm volatile("" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      asm volatile("" ::: "memory");
+      if (sB) P8_STAGE_B(t + 2, 0);
+      P8_READ_A_PH(3, a2);
+      P8_MFMA_PH(2, a);
+      if (sB) P8_STAGE_B(t + 2, 1);
+      P8_MFMA_PH(3, a2);
+      // tile boundary: counted drain + publishing barrier (as below)
+      if (DEEPA && sA && sB)
+        asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+      else if (!DEEPA && sB)
+        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      else if (DEEPA && sB)
+        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      continue;
+    }
+
     if constexpr (!LITE) {
       asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");
       __builtin_amdgcn_s_barrier();
@@ -295,6 +353,8 @@ __global__ __launch_bounds__(512, 1) void gemm256p8_bf16_kernel(
   }
 #undef P8_STAGE_A
 #undef P8_STAGE_B
+#undef P8_READ_A_PH
+#undef P8_MFMA_PH
 
   // epilogue (C/D map: col = lane & 15, row = (lane>>4)*4 + r); bias
   // hoisted to one load per fn column (the round-1 epilogue reloaded it
@@ -343,8 +403,20 @@ void gemm256p8_bf16_launch(const void* A, const void* Bst, const float* bias,
     return e && e[0] == '1';
   }();
   const bool tail = (K % P8_BK) != 0;
+  // phase-pipelined A reads (LITE only; A/B via LCTR_GEMM_P8_APIPE=0/1,
+  // re-read per launch for in-process sweeps)
+  const char* ea = getenv("LCTR_GEMM_P8_APIPE");
+  const bool apipe = ea && ea[0] == '1';
 #define P8_LAUNCH(T, L, X)                                                    hipLaunchKernelGGL((gemm256p8_bf16_kernel<T, L, X>), grid, block, 0,                           stream, (const __bf16*)A, (const __bf16*)Bst, bias, C,                      (__bf16*)Cbf, M, N, K, act)
-  if (!tail && lite && !xcd) P8_LAUNCH(false, true, false);
+#define P8_LAUNCH_AP(T)                                                  \
+  hipLaunchKernelGGL((gemm256p8_bf16_kernel<T, true, false, false,       \
+                                            true>),                     \
+                     grid, block, 0, stream, (const __bf16*)A,           \
+                     (const __bf16*)Bst, bias, C, (__bf16*)Cbf, M, N, K, \
+                     act)
+  if (!tail && lite && !xcd && apipe) P8_LAUNCH_AP(false);
+  else if (tail && lite && !xcd && apipe) P8_LAUNCH_AP(true);
+  else if (!tail && lite && !xcd) P8_LAUNCH(false, true, false);
   else if (!tail && lite && xcd) P8_LAUNCH(false, true, true);
   else if (!tail && !lite && !xcd) P8_LAUNCH(false, false, false);
   else if (!tail && !lite && xcd) P8_LAUNCH(false, false, true);
@@ -353,6 +425,7 @@ void gemm256p8_bf16_launch(const void* A, const void* Bst, const float* bias,
   else if (tail && !lite && !xcd) P8_LAUNCH(true, false, false);
   else P8_LAUNCH(true, false, true);
 #undef P8_LAUNCH
+#undef P8_LAUNCH_AP
 }
 
 }  // namespace lightctr
