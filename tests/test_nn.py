@@ -298,6 +298,36 @@ def test_gemm_p8_race_screen():
 
 
 @pytest.mark.gpu
+def test_gemm_apipe_schedule_variants():
+    """The gemm256 per-phase A-read pipeline (LCTR_GEMM_APIPE, default 2)
+    and its env fallbacks all match the fp32 reference; 5-run screen per
+    variant since the pipelined tile body is a new schedule."""
+    import os
+
+    from lightctr_amd.ops import hip_ops
+
+    M, N, K = 1024, 1024, 1024
+    g = torch.Generator().manual_seed(7)
+    A = (torch.randn(M, K, generator=g) * 0.4).to(torch.bfloat16).cuda()
+    Bst = (torch.randn(N, K, generator=g) * 0.4).to(torch.bfloat16).cuda()
+    ref = A.float() @ Bst.float().t()
+    old = os.environ.get("LCTR_GEMM_APIPE")
+    try:
+        for v in ("0", "1", "2"):
+            os.environ["LCTR_GEMM_APIPE"] = v
+            for it in range(5):
+                C = hip_ops.gemm_bf16(A, Bst, None, M, N, K, 0, 0, 0,
+                                      False)
+                assert torch.allclose(C, ref, atol=2e-2 * K ** 0.5,
+                                      rtol=1e-2), (v, it)
+    finally:
+        if old is None:
+            os.environ.pop("LCTR_GEMM_APIPE", None)
+        else:
+            os.environ["LCTR_GEMM_APIPE"] = old
+
+
+@pytest.mark.gpu
 def test_gemm_degenerate_shapes():
     """The 1-wide-layer fast paths (gemv N=1, weighted rowsum M=1,
     outer product K=1) match the fp32 reference in the exact call
