@@ -703,7 +703,7 @@ struct FfmOptArgs {
   float q0, q1, q2;      // V adagrad: lr, eps, l2
 };
 
-template <bool FUSED>
+template <bool FUSED, bool PREF = true>
 __global__ void ffm_blocks_apply_f16_kernel(
     const int* __restrict__ sorted_fids, const int* __restrict__ perm,
     const _Float16* __restrict__ gblocks, const float* __restrict__ gw,
@@ -731,6 +731,13 @@ __global__ void ffm_blocks_apply_f16_kernel(
   int cur = -1;
   bool head_ok = false;
   float accw = 0.f;
+  // optimizer-state prefetch (FUSED): the V/nV quad lines of the run
+  // being accumulated, issued at run-HEAD detection so their ~600-cycle
+  // scattered-load latency overlaps the run's gblock accumulation
+  // instead of serializing the flush (the in-flush RMW chain was the
+  // diagnosed residual of this latency-bound walk — docs/STATUS.md
+  // open lever #2).
+  float4 vPre[MAXQ], nPre[MAXQ];
   auto flush = [&](int tail_e) {
     if (cur < 0) return;
     const bool tail_ok = tail_e >= nnz || sorted_fids[tail_e] != cur;
@@ -745,8 +752,8 @@ __global__ void ffm_blocks_apply_f16_kernel(
         const int q = t * LCTR_WAVE + lane;
         if (q >= nq) break;
         const float4 ac = acc[t];
-        float4 v = V4[q];
-        float4 a = nV4[q];
+        float4 v = PREF ? vPre[t] : V4[q];
+        float4 a = PREF ? nPre[t] : nV4[q];
         const float gx = ac.x * inv_scale + oa.q2 * v.x;
         const float gy = ac.y * inv_scale + oa.q2 * v.y;
         const float gz = ac.z * inv_scale + oa.q2 * v.z;
@@ -842,6 +849,16 @@ __global__ void ffm_blocks_apply_f16_kernel(
       head_ok = ((e0) == 0 || sorted_fids[(e0) - 1] != fid_);              \
       if (lane == 0 && head_ok && !FUSED)                                  \
         atomicOr(&touched[fid_ >> 6], 1ull << (fid_ & 63));                \
+      if (FUSED && PREF && head_ok) { /* state prefetch: new run */        \
+        const float4* V4_ = (const float4*)&V[(size_t)fid_ * D];           \
+        const float4* nV4_ = (const float4*)&oa.nV[(size_t)fid_ * D];      \
+        _Pragma("unroll") for (int t = 0; t < MAXQ; ++t) {                 \
+          const int q_ = t * LCTR_WAVE + lane;                             \
+          if (q_ >= nq) break;                                             \
+          vPre[t] = V4_[q_];                                               \
+          nPre[t] = nV4_[q_];                                              \
+        }                                                                  \
+      }                                                                    \
     }                                                                      \
     _Pragma("unroll") for (int t = 0; t < MAXQ; ++t) {                     \
       const int q_ = t * LCTR_WAVE + lane;                                 \
@@ -957,11 +974,24 @@ void ffm_blocks_apply_f16_launch(const int* sorted_fids, const int* perm,
   dim3 block(wpb * LCTR_WAVE);
   dim3 grid((nwaves + wpb - 1) / wpb);
   FfmOptArgs oa{W, nW, zW, nV, (__bf16*)Vh, p0, p1, p2, p3, q0, q1, q2};
-  if (opt_mode != 0)
-    hipLaunchKernelGGL((ffm_blocks_apply_f16_kernel<true>), grid, block, 0,
-                       stream, sorted_fids, perm, (const _Float16*)gblocks,
-                       gw, gradW, gradV, touched, D, nnz, chunk, inv_scale,
-                       opt_mode, V, oa);
+  // LCTR_FFM_PREF=1 enables the run-head optimizer-state prefetch —
+  // measured NEGATIVE (1527 vs 1431 us isolated fused apply, 8/8
+  // alternating passes, profiles/r2_11_ffm_pref_negative.txt): the +32
+  // VGPRs (92 -> 124) cost a wave of occupancy (5 -> 4 per SIMD),
+  // which this latency-bound walk needs more than the overlap. Kept
+  // selectable; default off.
+  const char* epf = getenv("LCTR_FFM_PREF");
+  const bool pref = epf && epf[0] == '1';
+  if (opt_mode != 0 && pref)
+    hipLaunchKernelGGL((ffm_blocks_apply_f16_kernel<true, true>), grid,
+                       block, 0, stream, sorted_fids, perm,
+                       (const _Float16*)gblocks, gw, gradW, gradV, touched,
+                       D, nnz, chunk, inv_scale, opt_mode, V, oa);
+  else if (opt_mode != 0)
+    hipLaunchKernelGGL((ffm_blocks_apply_f16_kernel<true, false>), grid,
+                       block, 0, stream, sorted_fids, perm,
+                       (const _Float16*)gblocks, gw, gradW, gradV, touched,
+                       D, nnz, chunk, inv_scale, opt_mode, V, oa);
   else
     hipLaunchKernelGGL((ffm_blocks_apply_f16_kernel<false>), grid, block,
                        0, stream, sorted_fids, perm,
