@@ -703,7 +703,7 @@ struct FfmOptArgs {
   float q0, q1, q2;      // V adagrad: lr, eps, l2
 };
 
-template <bool FUSED, bool PREF = true>
+template <bool FUSED, bool PREF = true, int MAXQ = 4>
 __global__ void ffm_blocks_apply_f16_kernel(
     const int* __restrict__ sorted_fids, const int* __restrict__ perm,
     const _Float16* __restrict__ gblocks, const float* __restrict__ gw,
@@ -716,7 +716,11 @@ __global__ void ffm_blocks_apply_f16_kernel(
   // ds_read_b128 + ds_write_b128 round trip per quad per entry, the
   // dominant dependency chain of this latency-bound walk (PMC: VALUBusy
   // 20%, MemUnitStalled 0.4%).
-  constexpr int MAXQ = 4;  // quads per lane -> D <= 4*64*4 = 1024
+  // MAXQ = quads per lane (template; launcher picks ceil(D/256)): the
+  // register footprint scales with it — acc + the 2-deep entry pipeline
+  // are 3*MAXQ vec-registers, and at the flagship D=312 only 2 of the
+  // round-2 static 4 were ever populated. MAXQ=2 drops the kernel 92 ->
+  // ~68 VGPR, 5 -> 7 waves/SIMD on this latency-bound walk.
   const int lane = threadIdx.x & 63;
   const int wave = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   const int base = wave * chunk;
@@ -982,21 +986,31 @@ void ffm_blocks_apply_f16_launch(const int* sorted_fids, const int* perm,
   // selectable; default off.
   const char* epf = getenv("LCTR_FFM_PREF");
   const bool pref = epf && epf[0] == '1';
-  if (opt_mode != 0 && pref)
-    hipLaunchKernelGGL((ffm_blocks_apply_f16_kernel<true, true>), grid,
-                       block, 0, stream, sorted_fids, perm,
-                       (const _Float16*)gblocks, gw, gradW, gradV, touched,
-                       D, nnz, chunk, inv_scale, opt_mode, V, oa);
-  else if (opt_mode != 0)
-    hipLaunchKernelGGL((ffm_blocks_apply_f16_kernel<true, false>), grid,
-                       block, 0, stream, sorted_fids, perm,
-                       (const _Float16*)gblocks, gw, gradW, gradV, touched,
-                       D, nnz, chunk, inv_scale, opt_mode, V, oa);
-  else
-    hipLaunchKernelGGL((ffm_blocks_apply_f16_kernel<false>), grid, block,
-                       0, stream, sorted_fids, perm,
-                       (const _Float16*)gblocks, gw, gradW, gradV, touched,
-                       D, nnz, chunk, inv_scale, opt_mode, V, oa);
+  // MAXQ sized to D (LCTR_FFM_MAXQ=4 forces the round-2 static size
+  // for A/B); nq = D/4 quads, lanes stride 64
+  const int nq_ = D >> 2;
+  const char* emq = getenv("LCTR_FFM_MAXQ");
+  const int maxq = (emq && emq[0] == '4') ? 4
+                   : (nq_ <= 64 ? 1 : (nq_ <= 128 ? 2 : 4));
+#define FFM_AP_LAUNCH(F_, P_, Q_)                                        \
+  hipLaunchKernelGGL((ffm_blocks_apply_f16_kernel<F_, P_, Q_>), grid,    \
+                     block, 0, stream, sorted_fids, perm,                \
+                     (const _Float16*)gblocks, gw, gradW, gradV,         \
+                     touched, D, nnz, chunk, inv_scale, opt_mode, V, oa)
+  if (opt_mode != 0 && pref) {
+    if (maxq == 1) FFM_AP_LAUNCH(true, true, 1);
+    else if (maxq == 2) FFM_AP_LAUNCH(true, true, 2);
+    else FFM_AP_LAUNCH(true, true, 4);
+  } else if (opt_mode != 0) {
+    if (maxq == 1) FFM_AP_LAUNCH(true, false, 1);
+    else if (maxq == 2) FFM_AP_LAUNCH(true, false, 2);
+    else FFM_AP_LAUNCH(true, false, 4);
+  } else {
+    if (maxq == 1) FFM_AP_LAUNCH(false, true, 1);
+    else if (maxq == 2) FFM_AP_LAUNCH(false, true, 2);
+    else FFM_AP_LAUNCH(false, true, 4);
+  }
+#undef FFM_AP_LAUNCH
 }
 
 void ffm_forward_pp_launch(const int* row_ptr, const int* fields,

@@ -74,16 +74,21 @@ def main():
         return (time.perf_counter() - t0) / reps
 
     bench(40, 10)  # ramp
-    res = {}
-    for p in range(8):
-        for v in ("0", "1"):
-            os.environ["LCTR_FFM_PREF"] = v
-            t = bench()
-            res.setdefault(v, []).append(t)
-            print(f"p{p} PREF={v}: {t*1e6:8.1f} us")
-    for v, ts in sorted(res.items()):
-        ts = sorted(ts)[1:-1]
-        print(f"TRIMMED PREF={v}: {sum(ts)/len(ts)*1e6:.1f} us")
+    for env, vals in (("LCTR_FFM_PREF", ("0", "1")),
+                      ("LCTR_FFM_MAXQ", ("4", "0"))):
+        os.environ["LCTR_FFM_PREF"] = "0"
+        os.environ.pop("LCTR_FFM_MAXQ", None)
+        res = {}
+        for p in range(8):
+            for v in vals:
+                os.environ[env] = v
+                t = bench()
+                res.setdefault(v, []).append(t)
+                print(f"p{p} {env}={v}: {t*1e6:8.1f} us")
+        os.environ.pop(env, None)
+        for v, ts in sorted(res.items()):
+            ts = sorted(ts)[1:-1]
+            print(f"TRIMMED {env}={v}: {sum(ts)/len(ts)*1e6:.1f} us")
 
 
 if __name__ == "__main__":
