@@ -259,8 +259,12 @@ struct FmOptArgs {
   float q0, q1, q2;  // v_lr, v_eps, v_l2
 };
 
-template <int K, bool PP = true>
-__global__ void fm_sorted_apply_kernel(
+// WPE: min waves/SIMD handed to the allocator (HIP __launch_bounds__
+// 2nd arg). The K=16 walk allocates 86 VGPR — one register over the
+// 6-wave boundary (512/6 = 85.3); WPE=6 caps it at 85 for +20%
+// occupancy on this latency-bound kernel (LCTR_FM_APPLY_WPE=6 A/B).
+template <int K, bool PP = true, int WPE = 1>
+__global__ __launch_bounds__(256, WPE) void fm_sorted_apply_kernel(
     const int* __restrict__ sorted_fids, const int* __restrict__ perm,
     const float* __restrict__ gw, const float* __restrict__ gv,
     float* __restrict__ gradW, float* __restrict__ gradV,
@@ -837,6 +841,14 @@ void fm_sorted_apply_launch(const int* sorted_fids, const int* perm,
                                      perm, gw, gv, gradW, gradV, touched,
                                      nnz, chunk, opt_mode, V, oa));
   } else {
+    const char* ew = getenv("LCTR_FM_APPLY_WPE");
+    if (K == 16 && ew && ew[0] == '6') {
+      hipLaunchKernelGGL((fm_sorted_apply_kernel<16, false, 6>), grid,
+                         block, 0, stream, sorted_fids, perm, gw, gv,
+                         gradW, gradV, touched, nnz, chunk, opt_mode, V,
+                         oa);
+      return;
+    }
     DISPATCH_K(K, hipLaunchKernelGGL((fm_sorted_apply_kernel<KC, false>),
                                      grid, block, 0, stream, sorted_fids,
                                      perm, gw, gv, gradW, gradV, touched,
