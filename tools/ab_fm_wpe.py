@@ -22,14 +22,14 @@ def main():
     # equivalence at small F
     Fs = 1 << 20
     gen_s = SyntheticCriteo(num_features=Fs, seed=7, device="cuda")
-    batch_s = gen_s.batch(16384)
+    rp_s, _fs, fid_s, val_s, lab_s = gen_s.batch(16384)
     states = {}
     for v in ("6", "0"):
         os.environ["LCTR_FM_APPLY_WPE"] = v
         m = FMModel(FMHyper(num_features=Fs, k=16, optimizer="ftrl"),
                     device="cuda")
         for _ in range(3):
-            m.train_step(*batch_s)
+            m.train_step(rp_s, fid_s, val_s, lab_s)
         torch.cuda.synchronize()
         states[v] = (m.W.clone(), m.V.clone())
     dW = (states["6"][0] - states["0"][0]).abs().max().item()
