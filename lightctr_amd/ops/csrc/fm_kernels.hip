@@ -841,8 +841,11 @@ void fm_sorted_apply_launch(const int* sorted_fids, const int* perm,
                                      perm, gw, gv, gradW, gradV, touched,
                                      nnz, chunk, opt_mode, V, oa));
   } else {
+    // default for the flagship K=16: the 6-wave cap trims 86 -> 80
+    // VGPR (no spill) and measured 213.4 vs 222.5 us (+4.3%, 8/8
+    // alternating passes, profiles/r2_13_fm_wpe.txt). '0' reverts.
     const char* ew = getenv("LCTR_FM_APPLY_WPE");
-    if (K == 16 && ew && ew[0] == '6') {
+    if (K == 16 && !(ew && ew[0] == '0')) {
       hipLaunchKernelGGL((fm_sorted_apply_kernel<16, false, 6>), grid,
                          block, 0, stream, sorted_fids, perm, gw, gv,
                          gradW, gradV, touched, nnz, chunk, opt_mode, V,
