@@ -14,8 +14,12 @@ namespace lightctr {
 // Gather embeddings concatenated by position-in-row (fixed nnz/row layouts,
 // e.g. Criteo 39 fields): out[row, pos*K+k] = E[fid]*x, emitted bf16 for the
 // MLP GEMM (+optional fp32 copy). Wave per row, (pos-group, k) lanes.
+// 8-wave/SIMD cap: the gather allocated 66 VGPR, two registers over
+// the 8-wave boundary (512/8 = 64); the cap re-allocates to 64 with no
+// spill (same lever as the FM walk's WPE=6 — profiles/r2_13_fm_wpe.txt)
 template <int K>
-__global__ void embed_gather_kernel(const int* __restrict__ row_ptr,
+__global__ __launch_bounds__(256, 8) void embed_gather_kernel(
+    const int* __restrict__ row_ptr,
                                     const int* __restrict__ fids,
                                     const float* __restrict__ vals,
                                     const float* __restrict__ E,
