@@ -703,8 +703,8 @@ struct FfmOptArgs {
   float q0, q1, q2;      // V adagrad: lr, eps, l2
 };
 
-template <bool FUSED, bool PREF = true, int MAXQ = 4>
-__global__ void ffm_blocks_apply_f16_kernel(
+template <bool FUSED, bool PREF = true, int MAXQ = 4, int WPE = 1>
+__global__ __launch_bounds__(256, WPE) void ffm_blocks_apply_f16_kernel(
     const int* __restrict__ sorted_fids, const int* __restrict__ perm,
     const _Float16* __restrict__ gblocks, const float* __restrict__ gw,
     float* __restrict__ gradW, float* __restrict__ gradV,
@@ -997,6 +997,16 @@ void ffm_blocks_apply_f16_launch(const int* sorted_fids, const int* perm,
                      block, 0, stream, sorted_fids, perm,                \
                      (const _Float16*)gblocks, gw, gradW, gradV,         \
                      touched, D, nnz, chunk, inv_scale, opt_mode, V, oa)
+  // LCTR_FFM_WPE=8 caps the fused MAXQ=2 kernel at 8 waves/SIMD
+  // (72 -> <=64 VGPR) for A/B
+  const char* ewp = getenv("LCTR_FFM_WPE");
+  if (opt_mode != 0 && !pref && maxq == 2 && ewp && ewp[0] == '8') {
+    hipLaunchKernelGGL((ffm_blocks_apply_f16_kernel<true, false, 2, 8>),
+                       grid, block, 0, stream, sorted_fids, perm,
+                       (const _Float16*)gblocks, gw, gradW, gradV, touched,
+                       D, nnz, chunk, inv_scale, opt_mode, V, oa);
+    return;
+  }
   if (opt_mode != 0 && pref) {
     if (maxq == 1) FFM_AP_LAUNCH(true, true, 1);
     else if (maxq == 2) FFM_AP_LAUNCH(true, true, 2);

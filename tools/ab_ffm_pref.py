@@ -74,10 +74,8 @@ def main():
         return (time.perf_counter() - t0) / reps
 
     bench(40, 10)  # ramp
-    for env, vals in (("LCTR_FFM_PREF", ("0", "1")),
-                      ("LCTR_FFM_MAXQ", ("4", "0"))):
+    for env, vals in (("LCTR_FFM_WPE", ("0", "8")),):
         os.environ["LCTR_FFM_PREF"] = "0"
-        os.environ.pop("LCTR_FFM_MAXQ", None)
         res = {}
         for p in range(8):
             for v in vals:
