@@ -997,10 +997,12 @@ void ffm_blocks_apply_f16_launch(const int* sorted_fids, const int* perm,
                      block, 0, stream, sorted_fids, perm,                \
                      (const _Float16*)gblocks, gw, gradW, gradV,         \
                      touched, D, nnz, chunk, inv_scale, opt_mode, V, oa)
-  // LCTR_FFM_WPE=8 caps the fused MAXQ=2 kernel at 8 waves/SIMD
-  // (72 -> <=64 VGPR) for A/B
+  // 8-wave cap on the fused MAXQ=2 kernel (72 -> 63 VGPR + 36 B
+  // spill): measured 1316.4 vs 1327.9 us (+0.9%, 8/8 alternating
+  // passes) — the extra wave of latency cover beats the spill on this
+  // gather-bound walk. Default on; LCTR_FFM_WPE=0 reverts.
   const char* ewp = getenv("LCTR_FFM_WPE");
-  if (opt_mode != 0 && !pref && maxq == 2 && ewp && ewp[0] == '8') {
+  if (opt_mode != 0 && !pref && maxq == 2 && !(ewp && ewp[0] == '0')) {
     hipLaunchKernelGGL((ffm_blocks_apply_f16_kernel<true, false, 2, 8>),
                        grid, block, 0, stream, sorted_fids, perm,
                        (const _Float16*)gblocks, gw, gradW, gradV, touched,
