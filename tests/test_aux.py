@@ -465,3 +465,26 @@ def test_tools_importable():
         src = f.read_text()
         tree = ast.parse(src)  # syntax
         assert tree.body, f.name  # non-empty module
+
+
+def test_bench_contract_cpu_smoke():
+    """The driver's bench.py contract: runs on CPU (world 1) and prints
+    exactly one JSON line with the required fields and honest dtype."""
+    import json
+    import subprocess
+    import sys
+
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "2", "--warmup", "1",
+         "--batch", "512", "--features", "16384", "--model", "fm"],
+        capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr[-800:]
+    lines = [ln for ln in out.stdout.splitlines() if ln.startswith("{")]
+    assert len(lines) == 1, out.stdout
+    d = json.loads(lines[0])
+    for k in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+              "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+              "dtype", "data", "config"):
+        assert k in d, k
+    assert d["steps"] == 2 and d["data"] == "synthetic"
+    assert d["config"]["global_batch"] == 512
